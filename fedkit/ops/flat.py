@@ -1,0 +1,73 @@
+"""Flat parameter-vector ops — the federation / optimizer data plane.
+
+The reference flattens trainable parameters with per-tensor copies inside
+Python loops (simple_utils.py:47-77; lbfgsnew.py:81-121).  On MI355X each of
+those is its own tiny kernel launch; here the whole pack/unpack/axpy runs as
+ONE HIP kernel over a device-side descriptor table (csrc/flat_ops.hip), and
+the CPU fallback uses torch._foreach_* multi-tensor ops.
+
+API (all fp32, used by fedkit.utils.paramvec, fedkit.optim.LBFGSNew and the
+aggregation strategies):
+  pack(tensors)            -> new flat fp32 vector [sum numel]
+  pack_into(tensors, flat) -> fills an existing flat vector
+  unpack(flat, tensors)    -> scatters flat back into the tensors
+  add_flat(tensors, flat, alpha) -> t += alpha * flat_slice  per tensor
+"""
+
+from typing import List, Sequence
+
+import torch
+
+
+def _use_native(ts: Sequence[torch.Tensor]) -> bool:
+    from . import native_enabled
+    return len(ts) > 0 and native_enabled(ts[0])
+
+
+def total_numel(tensors: Sequence[torch.Tensor]) -> int:
+    return sum(t.numel() for t in tensors)
+
+
+def _flat_views(flat: torch.Tensor, tensors: Sequence[torch.Tensor]) -> List[torch.Tensor]:
+    views = []
+    off = 0
+    for t in tensors:
+        n = t.numel()
+        views.append(flat.narrow(0, off, n).view_as(t))
+        off += n
+    if off != flat.numel():
+        raise ValueError(f"flat vector has {flat.numel()} elements, tensors need {off}")
+    return views
+
+
+def pack_into(tensors: Sequence[torch.Tensor], flat: torch.Tensor) -> torch.Tensor:
+    if _use_native(tensors):
+        from . import require_ext
+        require_ext().pack_params(list(tensors), flat)
+        return flat
+    torch._foreach_copy_(_flat_views(flat, tensors), list(tensors))
+    return flat
+
+
+def pack(tensors: Sequence[torch.Tensor], device=None) -> torch.Tensor:
+    n = total_numel(tensors)
+    dev = device if device is not None else (tensors[0].device if tensors else "cpu")
+    flat = torch.empty(n, dtype=torch.float32, device=dev)
+    return pack_into(tensors, flat)
+
+
+def unpack(flat: torch.Tensor, tensors: Sequence[torch.Tensor]) -> None:
+    if _use_native(tensors):
+        from . import require_ext
+        require_ext().unpack_params(flat, list(tensors))
+        return
+    torch._foreach_copy_(list(tensors), _flat_views(flat, tensors))
+
+
+def add_flat(tensors: Sequence[torch.Tensor], flat: torch.Tensor, alpha: float) -> None:
+    """t += alpha * flat_slice for each tensor (the LBFGS _add_grad step)."""
+    if _use_native(tensors):
+        from . import require_ext
+        require_ext().add_flat_params(list(tensors), flat, float(alpha))
+        return
+    torch._foreach_add_(list(tensors), _flat_views(flat, tensors), alpha=alpha)

@@ -1,0 +1,62 @@
+"""FedBatchNorm2d — BatchNorm2d whose GPU path is a hand-written NHWC kernel.
+
+Subclasses nn.BatchNorm2d (parameter order / state_dict parity with the
+reference's nn.BatchNorm2d modules, simple_models.py:138-146).
+
+GPU path (csrc/batchnorm.hip): two-kernel NHWC BatchNorm —
+  (1) per-channel mean/var reduction over N*H*W (fp32 accumulation, one pass,
+      channels on the contiguous innermost axis so lanes read coalesced);
+  (2) normalize + scale/shift apply, writing y in the input dtype.
+Training stats + running-stat update happen on device; backward is the
+standard two-pass (reduce dy, dy*xhat; then apply), all fp32 math on bf16 data.
+
+CPU path: stock F.batch_norm (the numerics reference for the GPU tests).
+Stats and running buffers stay fp32 even when activations are bf16
+(SURVEY.md §7 hard part 4).
+"""
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+def _native(t):
+    from . import native_enabled
+    return native_enabled(t)
+
+
+def _ext():
+    from . import require_ext
+    return require_ext()
+
+
+class _BnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, running_mean, running_var,
+                training, momentum, eps):
+        y, save_mean, save_invstd = _ext().bn_fwd(
+            x, weight, bias, running_mean, running_var,
+            bool(training), float(momentum), float(eps))
+        ctx.save_for_backward(x, weight, save_mean, save_invstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, weight, save_mean, save_invstd = ctx.saved_tensors
+        gy = gy.contiguous(memory_format=torch.channels_last)
+        gx, gw, gb = _ext().bn_bwd(gy, x, weight, save_mean, save_invstd)
+        return gx, gw, gb, None, None, None, None, None
+
+
+class FedBatchNorm2d(nn.BatchNorm2d):
+    def forward(self, x):
+        if _native(x):
+            self._check_input_dim(x)
+            if self.training and self.track_running_stats \
+                    and self.num_batches_tracked is not None:
+                self.num_batches_tracked.add_(1)
+            x = x.contiguous(memory_format=torch.channels_last)
+            return _BnFn.apply(x, self.weight, self.bias,
+                               self.running_mean, self.running_var,
+                               self.training, self.momentum, self.eps)
+        return super().forward(x)
