@@ -1,0 +1,268 @@
+#include "hip/hip_runtime.h"
+// NHWC BatchNorm2d fwd/bwd for MI355X (SURVEY.md §2a "BatchNorm2d").
+// channels_last layout puts C innermost, so per-channel reductions read
+// coalesced 16 B/lane vectors.  Structure:
+//   fwd (training): partial-sums kernel (fp32 atomics into a [2][C]
+//     workspace) -> finalize (mean/invstd + running-stat update) -> apply
+//     (normalize + scale/shift, vectorized, output in input dtype).
+//   fwd (eval): apply with running stats.
+//   bwd: partial sums of (dy, dy*xhat) -> finalize (gw, gb, coefficients)
+//     -> apply dx = gamma*invstd*(dy - mean(dy) - xhat*mean(dy*xhat)).
+// All statistics fp32 regardless of activation dtype (bf16 data paths keep
+// fp32 BN stats — SURVEY.md §7 hard part 4).  C must divide 256 or be a
+// multiple of 256 (ResNet: 64/128/256/512) so each thread owns ONE channel
+// across its grid-stride walk; asserted on the host side.
+
+#include "fedkit_common.h"
+
+namespace {
+
+template <typename T, int VEC>
+struct alignas(sizeof(T) * VEC) VecT { T v[VEC]; };
+
+template <typename T>
+__global__ void bn_partials_kernel(const T* __restrict__ x, long long M, int C,
+                                   float* __restrict__ ws /* [2][C] */) {
+  // flat index over [M][C]; stride is a multiple of C so each thread's
+  // channel is fixed -> two register accumulators, one atomic pair at exit
+  long long total = M * (long long)C;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  long long i0 = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  float s = 0.f, sq = 0.f;
+  for (long long i = i0; i < total; i += stride) {
+    float v = to_f32(x[i]);
+    s += v;
+    sq += v * v;
+  }
+  if (i0 < total) {
+    int c = (int)(i0 % C);
+    atomicAdd(&ws[c], s);
+    atomicAdd(&ws[C + c], sq);
+  }
+}
+
+__global__ void bn_finalize_kernel(const float* __restrict__ ws, int C,
+                                   long long count, float eps, float momentum,
+                                   bool training, bool track,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var,
+                                   float* __restrict__ save_mean,
+                                   float* __restrict__ save_invstd) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float mean, var;
+  if (training) {
+    mean = ws[c] / count;
+    var = fmaxf(ws[C + c] / count - mean * mean, 0.f);  // biased
+    if (track) {
+      float unbiased = count > 1 ? var * count / (count - 1) : var;
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    }
+  } else {
+    mean = running_mean[c];
+    var = running_var[c];
+  }
+  save_mean[c] = mean;
+  save_invstd[c] = rsqrtf(var + eps);
+}
+
+template <typename T, int VEC>
+__global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ invstd,
+                                const float* __restrict__ gamma,
+                                const float* __restrict__ beta,
+                                long long nvec, int Cv /* C / VEC */) {
+  using V = VecT<T, VEC>;
+  const V* xv = reinterpret_cast<const V*>(x);
+  V* yv = reinterpret_cast<V*>(y);
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < nvec; i += (long long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % Cv) * VEC;
+    V a = xv[i], r;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      int c = c0 + j;
+      float v = (to_f32(a.v[j]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+      from_f32(v, r.v[j]);
+    }
+    yv[i] = r;
+  }
+}
+
+template <typename T>
+__global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
+                                       const T* __restrict__ gy, long long M,
+                                       int C, const float* __restrict__ mean,
+                                       const float* __restrict__ invstd,
+                                       float* __restrict__ ws /* [2][C] */) {
+  long long total = M * (long long)C;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  long long i0 = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  float sdy = 0.f, sdyx = 0.f;
+  if (i0 < total) {
+    int c = (int)(i0 % C);
+    float m = mean[c], is = invstd[c];
+    for (long long i = i0; i < total; i += stride) {
+      float g = to_f32(gy[i]);
+      sdy += g;
+      sdyx += g * (to_f32(x[i]) - m) * is;
+    }
+    atomicAdd(&ws[c], sdy);
+    atomicAdd(&ws[C + c], sdyx);
+  }
+}
+
+template <typename T, int VEC>
+__global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
+                                    const T* __restrict__ gy,
+                                    T* __restrict__ gx,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    const float* __restrict__ gamma,
+                                    const float* __restrict__ ws,
+                                    long long M, long long nvec, int Cv,
+                                    int VECC) {
+  using V = VecT<T, VEC>;
+  const V* xv = reinterpret_cast<const V*>(x);
+  const V* gv = reinterpret_cast<const V*>(gy);
+  V* ov = reinterpret_cast<V*>(gx);
+  int C = Cv * VEC;
+  float inv_count = 1.f / (float)M;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < nvec; i += (long long)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % Cv) * VEC;
+    V a = xv[i], g = gv[i], r;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      int c = c0 + j;
+      float is = invstd[c];
+      float xhat = (to_f32(a.v[j]) - mean[c]) * is;
+      float dy = to_f32(g.v[j]);
+      float val = gamma[c] * is *
+          (dy - ws[c] * inv_count - xhat * ws[C + c] * inv_count);
+      from_f32(val, r.v[j]);
+    }
+    ov[i] = r;
+  }
+  (void)VECC;
+}
+
+void check_nhwc(const at::Tensor& x) {
+  TORCH_CHECK(x.dim() == 4, "bn expects 4D input");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "bn expects channels_last input");
+  int C = x.size(1);
+  TORCH_CHECK(256 % C == 0 || C % 256 == 0,
+              "bn kernel needs C | 256 or 256 | C, got C=", C);
+}
+
+int bn_grid(long long total, int C) {
+  int g = grid_1d(total, 256, 512);
+  // keep gridDim*256 a multiple of C so each thread's channel is fixed
+  if (C > 256) {
+    int mult = C / 256;
+    g = std::max(mult, (g / mult) * mult);
+  }
+  return g;
+}
+
+}  // namespace
+
+std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& beta,
+                                      at::Tensor running_mean,
+                                      at::Tensor running_var, bool training,
+                                      double momentum, double eps) {
+  check_nhwc(x);
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  long long M = (long long)N * H * W;
+  auto fopts = x.options().dtype(at::kFloat);
+  auto save_mean = at::empty({C}, fopts);
+  auto save_invstd = at::empty({C}, fopts);
+  auto y = at::empty_like(x);
+  auto stream = fedkit_stream();
+  auto gamma_f = gamma.contiguous();
+  auto beta_f = beta.contiguous();
+  TORCH_CHECK(gamma_f.scalar_type() == at::kFloat, "bn gamma must be fp32");
+
+  if (training) {
+    auto ws = at::zeros({2, C}, fopts);
+    DISPATCH_F32_BF16(x, "bn_partials", {
+      hipLaunchKernelGGL((bn_partials_kernel<scalar_t>),
+                         dim3(bn_grid(M * C, C)), dim3(256), 0, stream,
+                         (const scalar_t*)x.data_ptr(), M, C,
+                         ws.data_ptr<float>());
+    });
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                       0, stream, ws.data_ptr<float>(), C, M, (float)eps,
+                       (float)momentum, true, running_mean.defined(),
+                       running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
+                       running_var.defined() ? running_var.data_ptr<float>() : nullptr,
+                       save_mean.data_ptr<float>(),
+                       save_invstd.data_ptr<float>());
+  } else {
+    auto ws = at::zeros({1}, fopts);  // unused
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                       0, stream, ws.data_ptr<float>(), C, M, (float)eps,
+                       (float)momentum, false, false,
+                       running_mean.data_ptr<float>(),
+                       running_var.data_ptr<float>(),
+                       save_mean.data_ptr<float>(),
+                       save_invstd.data_ptr<float>());
+  }
+  DISPATCH_F32_BF16(x, "bn_apply", {
+    constexpr int VEC = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % VEC == 0, "bn needs C % ", VEC, " == 0");
+    long long nvec = M * C / VEC;
+    hipLaunchKernelGGL((bn_apply_kernel<scalar_t, VEC>),
+                       dim3(grid_1d(nvec, 256)), dim3(256), 0, stream,
+                       (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(),
+                       save_mean.data_ptr<float>(),
+                       save_invstd.data_ptr<float>(),
+                       gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
+                       nvec, C / VEC);
+  });
+  return {y, save_mean, save_invstd};
+}
+
+std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& save_mean,
+                                      const at::Tensor& save_invstd) {
+  check_nhwc(x);
+  check_nhwc(gy);
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  long long M = (long long)N * H * W;
+  auto fopts = x.options().dtype(at::kFloat);
+  auto ws = at::zeros({2, C}, fopts);
+  auto gx = at::empty_like(x);
+  auto stream = fedkit_stream();
+  auto gamma_f = gamma.contiguous();
+  DISPATCH_F32_BF16(x, "bn_bwd_partials", {
+    hipLaunchKernelGGL((bn_bwd_partials_kernel<scalar_t>),
+                       dim3(bn_grid(M * C, C)), dim3(256), 0, stream,
+                       (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)gy.data_ptr(), M, C,
+                       save_mean.data_ptr<float>(),
+                       save_invstd.data_ptr<float>(), ws.data_ptr<float>());
+  });
+  DISPATCH_F32_BF16(x, "bn_bwd_apply", {
+    constexpr int VEC = 16 / sizeof(scalar_t);
+    long long nvec = M * C / VEC;
+    hipLaunchKernelGGL((bn_bwd_apply_kernel<scalar_t, VEC>),
+                       dim3(grid_1d(nvec, 256)), dim3(256), 0, stream,
+                       (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)gy.data_ptr(), (scalar_t*)gx.data_ptr(),
+                       save_mean.data_ptr<float>(),
+                       save_invstd.data_ptr<float>(),
+                       gamma_f.data_ptr<float>(), ws.data_ptr<float>(), M,
+                       nvec, C / VEC, VEC);
+  });
+  // gw = sum(dy * xhat), gb = sum(dy)
+  auto gw = ws[1].clone();
+  auto gb = ws[0].clone();
+  return {gx, gw, gb};
+}

@@ -1,0 +1,419 @@
+#include "hip/hip_runtime.h"
+// NHWC implicit-GEMM convolution on MFMA matrix cores (gfx950 / CDNA4).
+// The single hottest kernel family of the framework (SURVEY.md §2a rows 1-2:
+// all ResNet conv3x3 s1/s2 and 1x1 shortcut convs, fwd + bwd-data).
+//
+// GEMM view:  y[M=N*P*Q, Kout] = A[M, Kg=R*S*C] * B[Kg, Kout]
+//   A(m, k)   = xpad[n, p*stride + r, q*stride + s, c]   (im2col, implicit;
+//               input is PRE-PADDED so there are no boundary branches and
+//               global_load_lds (direct HBM->LDS DMA, 16 B/lane) can stage A)
+//   B(k, kout)= w[kout, r, s, c]  — [Kout][R*S*C] rows are k-contiguous, so
+//               B stages as a plain 2D tile copy.
+//
+// Structure (guide §5 anatomy, 2-phase double-buffered):
+//   tile BM x BN=64 x BK=64, 256 threads = 4 waves (2x2), wave tile
+//   (BM/2) x 32, mfma_f32_16x16x32_bf16 accumulating fp32;
+//   LDS: double-buffered A[BM][64] + B[64][64] bf16 in ONE __shared__ block;
+//   XOR swizzle on the 16-B k-chunk (chunk ^= row & 7) applied on the glds
+//   SOURCE address and the ds_read address (guide T2 / rule 21) to kill the
+//   16-way ds_read_b128 bank conflict of 128-B rows;
+//   K-loop: STAGE(next) || ds_read+MFMA(cur) || vmcnt(0)+barrier.
+//
+// bwd-data reuses this kernel: dx = conv_s1(dilate_pad(dy), rot180(w)^T)
+// (host-side transform kernels below).  bwd-weight = im2col + rocBLAS GEMM
+// for now (dy^T @ xcol is a plain library GEMM; a hand-written tr_b16
+// MFMA bwd-weight kernel is the planned replacement).
+//
+// Constraints (host-checked): C % 8 == 0, (R*S*C) % 64 == 0, Kout % 64 == 0,
+// bf16 tensors.  conv1 of ResNet (C=3) uses the dedicated direct kernel in
+// conv_small.hip.
+
+#include "fedkit_common.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __hip_bfloat16 bf16;
+
+#define GLDS16(gptr, lptr) \
+  __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)(gptr), \
+                                   (__attribute__((address_space(3))) void*)(lptr), 16, 0, 0)
+
+constexpr int BN = 64;   // out-channel tile
+constexpr int BK = 64;   // im2col-k tile
+
+// LDS byte offset of element (row, k) of a [rows][64] bf16 tile with the
+// chunk-XOR swizzle (16-B chunk index ^ (row & 7)).
+__device__ __forceinline__ int lds_off(int row, int k) {
+  int chunk = (k >> 3) ^ (row & 7);
+  return row * 128 + chunk * 16 + (k & 7) * 2;
+}
+
+template <int BM, int STRIDE>
+__global__ __launch_bounds__(256)
+void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
+                     const bf16* __restrict__ w,   // [Kout][R*S*C]
+                     bf16* __restrict__ y,         // [M][Kout_total]
+                     int N, int Hp, int Wp, int C, int Kout,
+                     int R, int S, int P, int Q, int Kg) {
+  constexpr int AB = BM * BK * 2;          // A tile bytes
+  constexpr int BB = BN * BK * 2;          // B tile bytes
+  __shared__ char smem[2 * (AB + BB)];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const long long M = (long long)N * P * Q;
+
+  const int bm = blockIdx.x;               // M tile
+  const int bn = blockIdx.y;               // Kout tile
+
+  // ---- staging geometry: each thread owns (BM*BK)/(256*8) A-slots and
+  //      (BN*BK)/(256*8) B-slots of 8 bf16 (16 B) each, fixed across the
+  //      K loop; only the (r,s,c) decomposition changes per K-tile.
+  constexpr int A_SLOTS = (BM * BK) / (256 * 8);   // 4 (BM=128) or 2 (BM=64)
+  constexpr int B_SLOTS = (BN * BK) / (256 * 8);   // 2
+
+  // A slot d = pass*256 + tid -> LDS row = d/8, stored chunk = d%8,
+  // source element (row, k8 = (d%8) ^ (row&7)).
+  long long a_rowbase[A_SLOTS];   // &xp[n][p*STRIDE][q*STRIDE][0] offset
+  int a_k8[A_SLOTS];
+#pragma unroll
+  for (int i = 0; i < A_SLOTS; ++i) {
+    int d = i * 256 + tid;
+    int row = d >> 3;
+    long long m = (long long)bm * BM + row;
+    if (m >= M) m = M - 1;                 // clamp (store side is guarded)
+    int q = (int)(m % Q);
+    int p = (int)((m / Q) % P);
+    int n = (int)(m / ((long long)P * Q));
+    a_rowbase[i] = (((long long)n * Hp + p * STRIDE) * Wp + q * STRIDE) * C;
+    a_k8[i] = (d & 7) ^ (row & 7);
+  }
+  long long b_rowbase[B_SLOTS];
+  int b_k8[B_SLOTS];
+#pragma unroll
+  for (int i = 0; i < B_SLOTS; ++i) {
+    int d = i * 256 + tid;
+    int row = d >> 3;                      // out channel within tile
+    b_rowbase[i] = ((long long)bn * BN + row) * Kg;
+    b_k8[i] = (d & 7) ^ (row & 7);
+  }
+
+  auto bufA = [&](int b) -> char* { return smem + b * (AB + BB); };
+  auto bufB = [&](int b) -> char* { return smem + b * (AB + BB) + AB; };
+
+  auto stage = [&](int buf, int kt) {
+    // A tile: per-slot source (r,s,c) from k_global; 16-B LDS-DMA.
+    // LDS dest for a glds is wave-uniform base + lane*16: slot d = pass*256
+    // + wave*64 + lane matches d = pass*256 + tid exactly.
+#pragma unroll
+    for (int i = 0; i < A_SLOTS; ++i) {
+      int kg = kt * BK + a_k8[i] * 8;
+      int c = kg % C;
+      int rs = kg / C;
+      int s = rs % S;
+      int r = rs / S;
+      const bf16* src = xp + a_rowbase[i] + ((long long)r * Wp + s) * C + c;
+      char* dst = bufA(buf) + (i * 4 + wave) * 1024;
+      GLDS16(src, dst);
+    }
+#pragma unroll
+    for (int i = 0; i < B_SLOTS; ++i) {
+      const bf16* src = w + b_rowbase[i] + kt * BK + b_k8[i] * 8;
+      char* dst = bufB(buf) + (i * 4 + wave) * 1024;
+      GLDS16(src, dst);
+    }
+  };
+
+  // ---- wave -> output sub-tile: 2x2 waves, wave tile (BM/2) x 32
+  constexpr int MR = BM / 2 / 16;          // A frags per wave (4 or 2)
+  constexpr int NR = 2;                    // B frags per wave
+  const int wm = (wave >> 1) * (BM / 2);   // wave row offset
+  const int wn = (wave & 1) * 32;          // wave col offset
+
+  f32x4 acc[MR][NR] = {};
+
+  const int frag_row = lane & 15;          // fragment row/col within 16
+  const int frag_k = (lane >> 4) * 8;      // 8 contiguous k per lane
+
+  const int nkt = Kg / BK;
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  int cur = 0;
+  for (int kt = 0; kt < nkt; ++kt) {
+    if (kt + 1 < nkt) stage(cur ^ 1, kt + 1);
+    const char* A = bufA(cur);
+    const char* B = bufB(cur);
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 32) {
+      bf16x8 a[MR], b[NR];
+#pragma unroll
+      for (int mfrag = 0; mfrag < MR; ++mfrag)
+        a[mfrag] = *(const bf16x8*)(A + lds_off(wm + mfrag * 16 + frag_row,
+                                                kk + frag_k));
+#pragma unroll
+      for (int nfrag = 0; nfrag < NR; ++nfrag)
+        b[nfrag] = *(const bf16x8*)(B + lds_off(wn + nfrag * 16 + frag_row,
+                                                kk + frag_k));
+#pragma unroll
+      for (int mfrag = 0; mfrag < MR; ++mfrag)
+#pragma unroll
+        for (int nfrag = 0; nfrag < NR; ++nfrag)
+          acc[mfrag][nfrag] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[mfrag], b[nfrag], acc[mfrag][nfrag], 0, 0, 0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- epilogue: D fragment lane l holds col = l&15, rows (l>>4)*4 + v
+  const int col = bn * BN + wn + frag_row;
+#pragma unroll
+  for (int mfrag = 0; mfrag < MR; ++mfrag) {
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      long long m = (long long)bm * BM + wm + mfrag * 16 + (lane >> 4) * 4 + v;
+      if (m < M) {
+#pragma unroll
+        for (int nfrag = 0; nfrag < NR; ++nfrag)
+          y[m * Kout + col + nfrag * 16] =
+              __float2bfloat16(acc[mfrag][nfrag][v]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------- transforms
+
+// zero-pad NHWC: out[n][h+pt][w+pl][c] = in[n][h][w][c]
+template <typename T>
+__global__ void pad_nhwc_kernel(const T* __restrict__ in, T* __restrict__ out,
+                                int N, int H, int W, int C, int Hp, int Wp,
+                                int pt, int pl) {
+  long long total = (long long)N * H * W * C;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long long t = i / C;
+    int w = (int)(t % W);
+    t /= W;
+    int h = (int)(t % H);
+    int n = (int)(t / H);
+    out[(((long long)n * Hp + h + pt) * Wp + (w + pl)) * C + c] = in[i];
+  }
+}
+
+// zero-dilate + pad NHWC (for bwd-data): out[n][h*str+pt][w*str+pl][c] = in
+template <typename T>
+__global__ void dilate_pad_nhwc_kernel(const T* __restrict__ in,
+                                       T* __restrict__ out, int N, int H,
+                                       int W, int C, int Hp, int Wp, int pt,
+                                       int pl, int str) {
+  long long total = (long long)N * H * W * C;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long long t = i / C;
+    int w = (int)(t % W);
+    t /= W;
+    int h = (int)(t % H);
+    int n = (int)(t / H);
+    out[(((long long)n * Hp + h * str + pt) * Wp + (w * str + pl)) * C + c] =
+        in[i];
+  }
+}
+
+// weight rotation for bwd-data: wrot[c][r][s][k] = w[k][R-1-r][S-1-s][c]
+template <typename T>
+__global__ void rot_weight_kernel(const T* __restrict__ w, T* __restrict__ wr,
+                                  int K, int R, int S, int C) {
+  long long total = (long long)K * R * S * C;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long long t = i / C;
+    int s = (int)(t % S);
+    t /= S;
+    int r = (int)(t % R);
+    int k = (int)(t / R);
+    wr[((((long long)c * R + (R - 1 - r)) * S + (S - 1 - s)) * K) + k] = w[i];
+  }
+}
+
+// im2col from the PADDED input (for bwd-weight GEMM):
+// col[m][r*S*C + s*C + c] = xp[n][p*stride+r][q*stride+s][c], m=(n,p,q)
+template <typename T>
+__global__ void im2col_kernel(const T* __restrict__ xp, T* __restrict__ col,
+                              int N, int Hp, int Wp, int C, int R, int S,
+                              int P, int Q, int stride) {
+  long long M = (long long)N * P * Q;
+  long long Kg = (long long)R * S * C;
+  long long total = M * Kg;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long m = i / Kg;
+    int k = (int)(i % Kg);
+    int c = k % C;
+    int rs = k / C;
+    int s = rs % S;
+    int r = rs / S;
+    int q = (int)(m % Q);
+    int p = (int)((m / Q) % P);
+    int n = (int)(m / ((long long)P * Q));
+    col[i] = xp[(((long long)n * Hp + p * stride + r) * Wp +
+                 (q * stride + s)) * C + c];
+  }
+}
+
+at::Tensor pad_nhwc(const at::Tensor& x, int pt, int pb, int pl, int pr) {
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int Hp = H + pt + pb, Wp = W + pl + pr;
+  auto xp = at::zeros({N, C, Hp, Wp},
+                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto stream = fedkit_stream();
+  long long total = (long long)N * H * W * C;
+  hipLaunchKernelGGL((pad_nhwc_kernel<bf16>), dim3(grid_1d(total, 256)),
+                     dim3(256), 0, stream, (const bf16*)x.data_ptr(),
+                     (bf16*)xp.data_ptr(), N, H, W, C, Hp, Wp, pt, pl);
+  return xp;
+}
+
+at::Tensor dilate_pad_nhwc(const at::Tensor& x, int pt, int pb, int pl, int pr,
+                           int str) {
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int Hp = (H - 1) * str + 1 + pt + pb, Wp = (W - 1) * str + 1 + pl + pr;
+  auto xp = at::zeros({N, C, Hp, Wp},
+                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto stream = fedkit_stream();
+  long long total = (long long)N * H * W * C;
+  hipLaunchKernelGGL((dilate_pad_nhwc_kernel<bf16>), dim3(grid_1d(total, 256)),
+                     dim3(256), 0, stream, (const bf16*)x.data_ptr(),
+                     (bf16*)xp.data_ptr(), N, H, W, C, Hp, Wp, pt, pl, str);
+  return xp;
+}
+
+void check_conv_inputs(const at::Tensor& x, const at::Tensor& w) {
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              w.scalar_type() == at::kBFloat16,
+              "conv MFMA kernel is bf16-only (the MI355X fast path)");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv input must be channels_last");
+  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv weight must be channels_last");
+}
+
+// launch on the PRE-PADDED input; pad already folded into Hp/Wp
+at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
+                     int stride, int P, int Q) {
+  int N = xp.size(0), C = xp.size(1), Hp = xp.size(2), Wp = xp.size(3);
+  int Kout = w_krs_c.size(0), R = w_krs_c.size(2), S = w_krs_c.size(3);
+  int Kg = R * S * C;
+  TORCH_CHECK(C % 8 == 0, "conv kernel needs C % 8 == 0, got ", C);
+  TORCH_CHECK(Kg % BK == 0, "conv kernel needs R*S*C % 64 == 0, got ", Kg);
+  TORCH_CHECK(Kout % BN == 0, "conv kernel needs Kout % 64 == 0, got ", Kout);
+  long long M = (long long)N * P * Q;
+  auto y = at::empty({N, Kout, P, Q},
+                     xp.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto stream = fedkit_stream();
+  // pick BM so the grid fills the 256 CUs when possible
+  bool bm64 = ((M + 127) / 128) * (Kout / BN) < 256;
+  int BM = bm64 ? 64 : 128;
+  dim3 grid((unsigned)((M + BM - 1) / BM), Kout / BN);
+  if (stride == 1) {
+    if (bm64)
+      hipLaunchKernelGGL((conv_fwd_kernel<64, 1>), grid, dim3(256), 0, stream,
+                         (const bf16*)xp.data_ptr(), (const bf16*)w_krs_c.data_ptr(),
+                         (bf16*)y.data_ptr(), N, Hp, Wp, C, Kout, R, S, P, Q, Kg);
+    else
+      hipLaunchKernelGGL((conv_fwd_kernel<128, 1>), grid, dim3(256), 0, stream,
+                         (const bf16*)xp.data_ptr(), (const bf16*)w_krs_c.data_ptr(),
+                         (bf16*)y.data_ptr(), N, Hp, Wp, C, Kout, R, S, P, Q, Kg);
+  } else {
+    TORCH_CHECK(stride == 2, "conv kernel supports stride 1/2");
+    if (bm64)
+      hipLaunchKernelGGL((conv_fwd_kernel<64, 2>), grid, dim3(256), 0, stream,
+                         (const bf16*)xp.data_ptr(), (const bf16*)w_krs_c.data_ptr(),
+                         (bf16*)y.data_ptr(), N, Hp, Wp, C, Kout, R, S, P, Q, Kg);
+    else
+      hipLaunchKernelGGL((conv_fwd_kernel<128, 2>), grid, dim3(256), 0, stream,
+                         (const bf16*)xp.data_ptr(), (const bf16*)w_krs_c.data_ptr(),
+                         (bf16*)y.data_ptr(), N, Hp, Wp, C, Kout, R, S, P, Q, Kg);
+  }
+  return y;
+}
+
+}  // namespace
+
+at::Tensor fedkit_conv_small_fwd(const at::Tensor& x, const at::Tensor& w,
+                                 long stride, long padding);  // conv_small.hip
+
+at::Tensor fedkit_conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
+                             long stride, long padding) {
+  check_conv_inputs(x, w);
+  int C = x.size(1), H = x.size(2), W = x.size(3);
+  int R = w.size(2);
+  if (C % 8 != 0)   // e.g. ResNet conv1 (C=3): direct small-C kernel
+    return fedkit_conv_small_fwd(x, w, stride, padding);
+  int P = (H + 2 * (int)padding - R) / (int)stride + 1;
+  int Q = (W + 2 * (int)padding - R) / (int)stride + 1;
+  at::Tensor xp = padding > 0 ? pad_nhwc(x, padding, padding, padding, padding)
+                              : x;
+  return conv_core(xp, w, (int)stride, P, Q);
+}
+
+at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
+                                  long stride, long padding, long H, long W) {
+  check_conv_inputs(gy, w);
+  int K = w.size(0), C = w.size(1), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(K % 8 == 0, "bwd-data needs Kout % 8 == 0");
+  // rotate weights: wrot[c][r][s][k]
+  auto wrot = at::empty({C, K, R, S},
+                        w.options().memory_format(at::MemoryFormat::ChannelsLast));
+  {
+    auto stream = fedkit_stream();
+    long long total = (long long)K * R * S * C;
+    hipLaunchKernelGGL((rot_weight_kernel<bf16>), dim3(grid_1d(total, 256)),
+                       dim3(256), 0, stream, (const bf16*)w.data_ptr(),
+                       (bf16*)wrot.data_ptr(), K, R, S, C);
+  }
+  // dilate+pad gy: pl = R-1-pad, pr = pl + a with a = (H + 2p - R) % stride
+  int pl = R - 1 - (int)padding;
+  int a = (int)((H + 2 * padding - R) % stride);
+  at::Tensor gyp = dilate_pad_nhwc(gy, pl, pl + a, pl, pl + a, (int)stride);
+  return conv_core(gyp, wrot, 1, (int)H, (int)W);
+}
+
+at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
+                                    long stride, long padding, long R_in,
+                                    long S_in) {
+  check_conv_inputs(gy, x);
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int K = gy.size(1), P = gy.size(2), Q = gy.size(3);
+  int R = (int)R_in, S = (int)S_in;
+  long long M = (long long)N * P * Q;
+  at::Tensor xp = padding > 0 ? pad_nhwc(x, padding, padding, padding, padding)
+                              : x;
+  at::Tensor col;
+  if (R == 1 && S == 1 && stride == 1) {
+    col = xp.permute({0, 2, 3, 1}).reshape({M, C});  // NHWC view, no copy
+  } else {
+    col = at::empty({M, (long long)R * S * C},
+                    x.options().memory_format(at::MemoryFormat::Contiguous));
+    auto stream = fedkit_stream();
+    long long total = M * R * S * C;
+    hipLaunchKernelGGL((im2col_kernel<bf16>), dim3(grid_1d(total, 512)),
+                       dim3(256), 0, stream, (const bf16*)xp.data_ptr(),
+                       (bf16*)col.data_ptr(), N, xp.size(2), xp.size(3), C,
+                       R, S, P, Q, (int)stride);
+  }
+  // dy^T @ col: plain library GEMM (rocBLAS/hipBLASLt) — fp32 accumulate
+  auto dy2d = gy.permute({0, 2, 3, 1}).reshape({M, K});
+  auto dw = at::matmul(dy2d.t(), col);             // [K, R*S*C] bf16
+  return dw.reshape({K, R, S, C}).permute({0, 3, 1, 2})
+      .contiguous(at::MemoryFormat::ChannelsLast);
+}

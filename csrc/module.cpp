@@ -1,0 +1,60 @@
+// fedkit._C — pybind bindings for the CDNA4 (gfx950) kernel library.
+// Build: __graft_entry__.build() -> setup.py build_ext --inplace with
+// PYTORCH_ROCM_ARCH=gfx950 (in-tree fedkit/_C.so; no JIT cache).
+
+#include <torch/extension.h>
+#include <vector>
+
+at::Tensor fedkit_elu_fwd(const at::Tensor& x);
+at::Tensor fedkit_elu_bwd(const at::Tensor& gy, const at::Tensor& y);
+
+void fedkit_pack_params(std::vector<at::Tensor> tensors, at::Tensor flat);
+void fedkit_unpack_params(at::Tensor flat, std::vector<at::Tensor> tensors);
+void fedkit_add_flat_params(std::vector<at::Tensor> tensors, at::Tensor flat,
+                            double alpha);
+
+std::vector<at::Tensor> fedkit_cross_entropy_fwd(const at::Tensor& logits,
+                                                 const at::Tensor& labels);
+at::Tensor fedkit_cross_entropy_bwd(const at::Tensor& logits,
+                                    const at::Tensor& labels,
+                                    const at::Tensor& lse,
+                                    const at::Tensor& gloss);
+
+std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& beta,
+                                      at::Tensor running_mean,
+                                      at::Tensor running_var, bool training,
+                                      double momentum, double eps);
+std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& save_mean,
+                                      const at::Tensor& save_invstd);
+
+at::Tensor fedkit_conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
+                             long stride, long padding);
+at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
+                                  long stride, long padding, long H, long W);
+at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
+                                    long stride, long padding, long R, long S);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "fedkit hand-written CDNA4 (gfx950 / MI355X) kernels";
+  m.def("elu_fwd", &fedkit_elu_fwd, "ELU forward (vectorized)");
+  m.def("elu_bwd", &fedkit_elu_bwd, "ELU backward from saved output");
+  m.def("pack_params", &fedkit_pack_params, "multi-tensor -> flat fp32");
+  m.def("unpack_params", &fedkit_unpack_params, "flat fp32 -> multi-tensor");
+  m.def("add_flat_params", &fedkit_add_flat_params,
+        "t += alpha * flat slice (multi-tensor axpy)");
+  m.def("cross_entropy_fwd", &fedkit_cross_entropy_fwd,
+        "fused log-softmax + NLL (mean): returns (loss, lse)");
+  m.def("cross_entropy_bwd", &fedkit_cross_entropy_bwd, "CE backward");
+  m.def("bn_fwd", &fedkit_bn_fwd,
+        "NHWC BatchNorm fwd: returns (y, save_mean, save_invstd)");
+  m.def("bn_bwd", &fedkit_bn_bwd,
+        "NHWC BatchNorm bwd: returns (gx, gw, gb)");
+  m.def("conv2d_fwd", &fedkit_conv2d_fwd,
+        "NHWC implicit-GEMM conv fwd on MFMA (3x3/1x1, stride 1/2)");
+  m.def("conv2d_bwd_data", &fedkit_conv2d_bwd_data, "conv bwd-data");
+  m.def("conv2d_bwd_weight", &fedkit_conv2d_bwd_weight, "conv bwd-weight");
+}

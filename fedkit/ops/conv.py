@@ -48,7 +48,8 @@ class _ConvFn(torch.autograd.Function):
             gx = _ext().conv2d_bwd_data(gy, w, ctx.stride, ctx.padding,
                                         x.shape[2], x.shape[3])
         if ctx.needs_input_grad[1]:
-            gw = _ext().conv2d_bwd_weight(gy, x, ctx.stride, ctx.padding)
+            gw = _ext().conv2d_bwd_weight(gy, x, ctx.stride, ctx.padding,
+                                          w.shape[2], w.shape[3])
         return gx, gw, None, None
 
 
@@ -56,12 +57,16 @@ class FedConv2d(nn.Conv2d):
     def forward(self, x):
         if _native(x) and self.bias is None and self.groups == 1 \
                 and self.dilation == (1, 1) and self.kernel_size[0] in (1, 3):
-            x = x.contiguous(memory_format=torch.channels_last)
-            w = self.weight
-            if x.dtype == torch.bfloat16 and w.dtype != torch.bfloat16:
-                w = w.to(torch.bfloat16)
-            w = w.contiguous(memory_format=torch.channels_last)
-            return _ConvFn.apply(x, w, self.stride[0], self.padding[0])
+            if x.dtype != torch.bfloat16 and torch.is_autocast_enabled():
+                x = x.to(torch.bfloat16)
+            if x.dtype == torch.bfloat16:
+                # the MI355X fast path: bf16 NHWC through the MFMA kernels;
+                # the fp32->bf16 weight cast is autograd-tracked, so bwd-weight
+                # gradients land on the fp32 master copy
+                x = x.contiguous(memory_format=torch.channels_last)
+                w = self.weight.to(torch.bfloat16).contiguous(
+                    memory_format=torch.channels_last)
+                return _ConvFn.apply(x, w, self.stride[0], self.padding[0])
         return F.conv2d(x, self.weight.to(x.dtype),
                         self.bias.to(x.dtype) if self.bias is not None else None,
                         self.stride, self.padding, self.dilation, self.groups)

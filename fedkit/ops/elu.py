@@ -23,14 +23,14 @@ def _native(t):
 class _EluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):
-        y = _ext().elu_fwd(x.contiguous())
+        y = _ext().elu_fwd(x)
         ctx.save_for_backward(y)
         return y
 
     @staticmethod
     def backward(ctx, gy):
         (y,) = ctx.saved_tensors
-        return _ext().elu_bwd(gy.contiguous(), y)
+        return _ext().elu_bwd(gy, y)
 
 
 def elu(x: torch.Tensor) -> torch.Tensor:

@@ -1,0 +1,293 @@
+"""GPU numerics tests: every HIP kernel vs a plain torch fp32 reference of
+the same op (SURVEY.md §4 test plan).  All marked @pytest.mark.gpu."""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+
+def _ext():
+    import fedkit.ops
+    assert fedkit.ops.has_ext(), "fedkit._C must be built on a GPU box"
+    return fedkit.ops.ext()
+
+
+def rel_err(got, want):
+    got = got.float()
+    want = want.float()
+    denom = want.abs().max().clamp_min(1e-6)
+    return (got - want).abs().max().item() / denom.item()
+
+
+def frob_err(got, want):
+    got = got.float()
+    want = want.float()
+    return ((got - want).norm() / want.norm().clamp_min(1e-12)).item()
+
+
+# ----------------------------------------------------------------- elementwise
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_elu_fwd_bwd(dtype):
+    ext = _ext()
+    torch.manual_seed(0)
+    x = torch.randn(3, 64, 17, 31, device="cuda", dtype=dtype)
+    y = ext.elu_fwd(x)
+    want = F.elu(x.float())
+    tol = 1e-6 if dtype == torch.float32 else 1e-2
+    assert rel_err(y, want) < tol
+    gy = torch.randn_like(x)
+    gx = ext.elu_bwd(gy, y)
+    xref = x.float().detach().requires_grad_(True)
+    F.elu(xref).backward(gy.float())
+    assert rel_err(gx, xref.grad) < (1e-5 if dtype == torch.float32 else 2e-2)
+
+
+def test_elu_channels_last_no_copy():
+    ext = _ext()
+    x = torch.randn(2, 64, 8, 8, device="cuda").contiguous(
+        memory_format=torch.channels_last)
+    y = ext.elu_fwd(x)
+    assert y.is_contiguous(memory_format=torch.channels_last)
+    assert rel_err(y, F.elu(x)) < 1e-6
+
+
+# -------------------------------------------------------------------- flat ops
+
+def test_pack_unpack_axpy_gpu():
+    ext = _ext()
+    torch.manual_seed(1)
+    ts = [torch.randn(s, device="cuda") for s in
+          [(64, 3, 3, 3), (64,), (11,), (512, 256), (1,)]]
+    n = sum(t.numel() for t in ts)
+    flat = torch.empty(n, device="cuda")
+    ext.pack_params(ts, flat)
+    want = torch.cat([t.reshape(-1) for t in ts])
+    assert torch.equal(flat, want)
+    outs = [torch.zeros_like(t) for t in ts]
+    ext.unpack_params(flat, outs)
+    for a, b in zip(ts, outs):
+        assert torch.equal(a, b)
+    upd = torch.randn(n, device="cuda")
+    ref = [t.clone() for t in ts]
+    ext.add_flat_params(ts, upd, 0.3)
+    off = 0
+    for t, r in zip(ts, ref):
+        k = r.numel()
+        assert torch.allclose(t, r + 0.3 * upd[off:off + k].view_as(r),
+                              atol=1e-6)
+        off += k
+
+
+def test_pack_many_tensors_chunking():
+    """More than one 48-tensor descriptor chunk (ResNet18 has 62 tensors)."""
+    ext = _ext()
+    ts = [torch.randn(7, device="cuda") for _ in range(130)]
+    flat = torch.empty(7 * 130, device="cuda")
+    ext.pack_params(ts, flat)
+    assert torch.equal(flat, torch.cat(ts))
+
+
+# ----------------------------------------------------------------------- loss
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_cross_entropy(dtype):
+    ext = _ext()
+    torch.manual_seed(2)
+    logits = torch.randn(128, 10, device="cuda", dtype=dtype)
+    labels = torch.randint(0, 10, (128,), device="cuda")
+    loss, lse = ext.cross_entropy_fwd(logits, labels)
+    want = F.cross_entropy(logits.float(), labels)
+    assert abs(loss.item() - want.item()) < (1e-5 if dtype == torch.float32 else 5e-3)
+    g = torch.ones((), device="cuda")
+    gx = ext.cross_entropy_bwd(logits, labels, lse, g)
+    ref = logits.float().detach().requires_grad_(True)
+    F.cross_entropy(ref, labels).backward()
+    assert rel_err(gx, ref.grad) < (1e-4 if dtype == torch.float32 else 2e-2)
+
+
+# ------------------------------------------------------------------ batchnorm
+
+@pytest.mark.parametrize("C", [64, 128, 256, 512])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_bn_fwd_bwd_vs_torch(C, dtype):
+    ext = _ext()
+    torch.manual_seed(3)
+    N, H, W = 8, 9, 9
+    x = (torch.randn(N, C, H, W, device="cuda") * 2 + 0.5).to(dtype)
+    x = x.contiguous(memory_format=torch.channels_last)
+    gamma = torch.rand(C, device="cuda") + 0.5
+    beta = torch.randn(C, device="cuda")
+    rm = torch.zeros(C, device="cuda")
+    rv = torch.ones(C, device="cuda")
+    y, sm, siv = ext.bn_fwd(x, gamma, beta, rm, rv, True, 0.1, 1e-5)
+
+    xf = x.float().detach().requires_grad_(True)
+    rm_ref = torch.zeros(C, device="cuda")
+    rv_ref = torch.ones(C, device="cuda")
+    gref = gamma.detach().requires_grad_(True)
+    bref = beta.detach().requires_grad_(True)
+    yref = F.batch_norm(xf, rm_ref, rv_ref, gref, bref, True, 0.1, 1e-5)
+    tol = 1e-4 if dtype == torch.float32 else 3e-2
+    assert rel_err(y, yref) < tol
+    assert rel_err(rm, rm_ref) < 1e-3
+    assert rel_err(rv, rv_ref) < 1e-3
+
+    gy = torch.randn_like(x)
+    gx, gw, gb = ext.bn_bwd(gy, x, gamma, sm, siv)
+    yref.backward(gy.float())
+    assert rel_err(gb, bref.grad) < (1e-3 if dtype == torch.float32 else 3e-2)
+    assert rel_err(gw, gref.grad) < (1e-3 if dtype == torch.float32 else 3e-2)
+    assert rel_err(gx, xf.grad) < (1e-3 if dtype == torch.float32 else 5e-2)
+
+
+def test_bn_eval_mode():
+    ext = _ext()
+    C = 64
+    x = torch.randn(4, C, 8, 8, device="cuda").contiguous(
+        memory_format=torch.channels_last)
+    gamma = torch.rand(C, device="cuda") + 0.5
+    beta = torch.randn(C, device="cuda")
+    rm = torch.randn(C, device="cuda")
+    rv = torch.rand(C, device="cuda") + 0.5
+    y, _, _ = ext.bn_fwd(x, gamma, beta, rm.clone(), rv.clone(), False, 0.1, 1e-5)
+    yref = F.batch_norm(x, rm, rv, gamma, beta, False, 0.1, 1e-5)
+    assert rel_err(y, yref) < 1e-4
+
+
+# ----------------------------------------------------------------------- conv
+
+RESNET_SHAPES = [
+    # (N, C, H, K, R, stride) — the ResNet18 CIFAR conv menu
+    (16, 64, 32, 64, 3, 1),
+    (16, 64, 32, 128, 3, 2),
+    (16, 128, 16, 128, 3, 1),
+    (16, 128, 16, 256, 3, 2),
+    (16, 256, 8, 256, 3, 1),
+    (16, 256, 8, 512, 3, 2),
+    (16, 512, 4, 512, 3, 1),
+    (16, 64, 32, 128, 1, 2),   # shortcut projections
+    (16, 128, 16, 256, 1, 2),
+    (16, 256, 8, 512, 1, 2),
+]
+
+
+@pytest.mark.parametrize("N,C,H,K,R,stride", RESNET_SHAPES)
+def test_conv_fwd_vs_torch(N, C, H, K, R, stride):
+    ext = _ext()
+    torch.manual_seed(4)
+    pad = 1 if R == 3 else 0
+    x = torch.randn(N, C, H, H, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(K, C, R, R, device="cuda", dtype=torch.bfloat16) * 0.05
+    xc = x.contiguous(memory_format=torch.channels_last)
+    wc = w.contiguous(memory_format=torch.channels_last)
+    y = ext.conv2d_fwd(xc, wc, stride, pad)
+    want = F.conv2d(x.float(), w.float(), None, stride, pad)
+    assert y.shape == want.shape
+    assert frob_err(y, want) < 1.5e-2, f"fwd mismatch {frob_err(y, want)}"
+
+
+@pytest.mark.parametrize("N,C,H,K,R,stride", RESNET_SHAPES)
+def test_conv_bwd_data_vs_torch(N, C, H, K, R, stride):
+    ext = _ext()
+    torch.manual_seed(5)
+    pad = 1 if R == 3 else 0
+    P = (H + 2 * pad - R) // stride + 1
+    gy = torch.randn(N, K, P, P, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(K, C, R, R, device="cuda", dtype=torch.bfloat16) * 0.05
+    gyc = gy.contiguous(memory_format=torch.channels_last)
+    wc = w.contiguous(memory_format=torch.channels_last)
+    gx = ext.conv2d_bwd_data(gyc, wc, stride, pad, H, H)
+    xref = torch.zeros(N, C, H, H, device="cuda", requires_grad=True)
+    F.conv2d(xref, w.float(), None, stride, pad).backward(gy.float())
+    assert gx.shape == xref.grad.shape
+    assert frob_err(gx, xref.grad) < 1.5e-2
+
+
+@pytest.mark.parametrize("N,C,H,K,R,stride", RESNET_SHAPES)
+def test_conv_bwd_weight_vs_torch(N, C, H, K, R, stride):
+    ext = _ext()
+    torch.manual_seed(6)
+    pad = 1 if R == 3 else 0
+    P = (H + 2 * pad - R) // stride + 1
+    x = torch.randn(N, C, H, H, device="cuda", dtype=torch.bfloat16)
+    gy = torch.randn(N, K, P, P, device="cuda", dtype=torch.bfloat16)
+    xc = x.contiguous(memory_format=torch.channels_last)
+    gyc = gy.contiguous(memory_format=torch.channels_last)
+    gw = ext.conv2d_bwd_weight(gyc, xc, stride, pad, R, R)
+    wref = torch.zeros(K, C, R, R, device="cuda", requires_grad=True)
+    F.conv2d(x.float(), wref, None, stride, pad).backward(gy.float())
+    assert gw.shape == wref.grad.shape
+    assert frob_err(gw, wref.grad) < 1.5e-2
+
+
+def test_conv_small_c3():
+    """ResNet conv1: C=3 direct kernel."""
+    ext = _ext()
+    torch.manual_seed(7)
+    x = torch.randn(8, 3, 32, 32, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(64, 3, 3, 3, device="cuda", dtype=torch.bfloat16) * 0.2
+    y = ext.conv2d_fwd(x.contiguous(memory_format=torch.channels_last),
+                       w.contiguous(memory_format=torch.channels_last), 1, 1)
+    want = F.conv2d(x.float(), w.float(), None, 1, 1)
+    assert frob_err(y, want) < 1.5e-2
+
+
+def test_conv_identity_transpose_detecting():
+    """Asymmetric-B check (guide G9): catches swapped MFMA operand layouts."""
+    ext = _ext()
+    C, K = 64, 64
+    x = torch.zeros(2, C, 8, 8, device="cuda", dtype=torch.bfloat16)
+    # delta input at a single (pixel, channel)
+    x[0, 5, 3, 4] = 1.0
+    w = torch.arange(K * C, device="cuda", dtype=torch.float32).reshape(K, C, 1, 1)
+    w = (w / (K * C)).to(torch.bfloat16)
+    y = ext.conv2d_fwd(x.contiguous(memory_format=torch.channels_last),
+                       w.contiguous(memory_format=torch.channels_last), 1, 0)
+    want = F.conv2d(x.float(), w.float(), None, 1, 0)
+    assert frob_err(y, want) < 1e-2
+
+
+# ----------------------------------------------------------- module-level path
+
+def test_fedconv_module_autograd():
+    from fedkit.ops import FedConv2d
+    torch.manual_seed(8)
+    m = FedConv2d(64, 128, 3, stride=2, padding=1, bias=False).cuda()
+    m = m.to(memory_format=torch.channels_last)
+    x = torch.randn(4, 64, 16, 16, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True).contiguous(memory_format=torch.channels_last)
+    y = m(x)
+    loss = y.float().square().mean()
+    loss.backward()
+    assert m.weight.grad is not None
+    assert x.grad is not None
+    # fp32 reference
+    xr = x.detach().float().requires_grad_(True)
+    yr = F.conv2d(xr, m.weight.detach().float(), None, 2, 1)
+    yr.square().mean().backward()
+    assert frob_err(y, yr) < 2e-2
+    assert frob_err(x.grad, xr.grad) < 5e-2
+
+
+def test_resnet18_training_step_bf16():
+    """Whole-model smoke: one fwd+bwd+Adam step through the HIP kernels."""
+    from fedkit.models import ResNet18
+    from fedkit.ops import losses as L
+    torch.manual_seed(9)
+    net = ResNet18().cuda().to(memory_format=torch.channels_last)
+    opt = torch.optim.Adam(net.parameters(), lr=1e-3)
+    x = torch.randn(32, 3, 32, 32, device="cuda").contiguous(
+        memory_format=torch.channels_last)
+    y = torch.randint(0, 10, (32,), device="cuda")
+    with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+        loss = L.cross_entropy(net(x), y)
+    opt.zero_grad()
+    loss.backward()
+    opt.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
+    for p in net.parameters():
+        assert torch.isfinite(p).all()
