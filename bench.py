@@ -79,9 +79,9 @@ def main():
     # synthetic CIFAR-shaped pool resident on device (no network; BASELINE
     # config: random-init weights, synthetic data of the CIFAR10 shape)
     pool = 16
-    xs = torch.randn(pool, args.batch, 3, 32, 32, device=device)
+    xs = [torch.randn(args.batch, 3, 32, 32, device=device) for _ in range(pool)]
     if use_cuda:
-        xs = xs.contiguous(memory_format=torch.channels_last)
+        xs = [x.contiguous(memory_format=torch.channels_last) for x in xs]
     ys = torch.randint(0, 10, (pool, args.batch), device=device)
 
     def autocast():

@@ -257,8 +257,8 @@ def test_fedconv_module_autograd():
     torch.manual_seed(8)
     m = FedConv2d(64, 128, 3, stride=2, padding=1, bias=False).cuda()
     m = m.to(memory_format=torch.channels_last)
-    x = torch.randn(4, 64, 16, 16, device="cuda", dtype=torch.bfloat16,
-                    requires_grad=True).contiguous(memory_format=torch.channels_last)
+    x = torch.randn(4, 64, 16, 16, device="cuda", dtype=torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last).requires_grad_(True)
     y = m(x)
     loss = y.float().square().mean()
     loss.backward()
