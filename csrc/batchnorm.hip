@@ -19,24 +19,51 @@ namespace {
 template <typename T, int VEC>
 struct alignas(sizeof(T) * VEC) VecT { T v[VEC]; };
 
-template <typename T>
-__global__ void bn_partials_kernel(const T* __restrict__ x, long long M, int C,
+// Vectorized (16 B/lane) per-channel reduction: each thread owns VEC
+// consecutive channels fixed across its grid-stride walk (C/VEC divides the
+// 256-thread block), accumulates in registers, reduces across the block via
+// LDS, then one atomicAdd per (block, channel) into the [2][C] workspace.
+template <typename T, int VEC>
+__global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
+                                   int Cv /* C/VEC */,
                                    float* __restrict__ ws /* [2][C] */) {
-  // flat index over [M][C]; stride is a multiple of C so each thread's
-  // channel is fixed -> two register accumulators, one atomic pair at exit
-  long long total = M * (long long)C;
+  using V = VecT<T, VEC>;
+  const int C = Cv * VEC;
+  const V* xv = reinterpret_cast<const V*>(x);
+  __shared__ float red[256 * VEC];
+  long long total = M * Cv;
   long long stride = (long long)gridDim.x * blockDim.x;
   long long i0 = blockIdx.x * (long long)blockDim.x + threadIdx.x;
-  float s = 0.f, sq = 0.f;
+  int tid = threadIdx.x;
+  int c0 = (int)(i0 % Cv) * VEC;
+  float s[VEC] = {}, sq[VEC] = {};
   for (long long i = i0; i < total; i += stride) {
-    float v = to_f32(x[i]);
-    s += v;
-    sq += v * v;
+    V a = xv[i];
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float v = to_f32(a.v[j]);
+      s[j] += v;
+      sq[j] += v * v;
+    }
   }
-  if (i0 < total) {
-    int c = (int)(i0 % C);
-    atomicAdd(&ws[c], s);
-    atomicAdd(&ws[C + c], sq);
+  const int members = 256 / Cv;
+  const int g = tid % Cv;
+#pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    float* src = pass == 0 ? s : sq;
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) red[tid * VEC + j] = src[j];
+    __syncthreads();
+    if (tid < Cv) {
+      float acc[VEC] = {};
+      for (int k = 0; k < members; ++k)
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) acc[j] += red[(g + k * Cv) * VEC + j];
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        atomicAdd(&ws[pass * C + c0 + j], acc[j]);
+    }
   }
 }
 
@@ -90,26 +117,56 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
   }
 }
 
-template <typename T>
+template <typename T, int VEC>
 __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
                                        const T* __restrict__ gy, long long M,
-                                       int C, const float* __restrict__ mean,
+                                       int Cv, const float* __restrict__ mean,
                                        const float* __restrict__ invstd,
                                        float* __restrict__ ws /* [2][C] */) {
-  long long total = M * (long long)C;
+  using V = VecT<T, VEC>;
+  const int C = Cv * VEC;
+  const V* xv = reinterpret_cast<const V*>(x);
+  const V* gv = reinterpret_cast<const V*>(gy);
+  __shared__ float red[256 * VEC];
+  long long total = M * Cv;
   long long stride = (long long)gridDim.x * blockDim.x;
   long long i0 = blockIdx.x * (long long)blockDim.x + threadIdx.x;
-  float sdy = 0.f, sdyx = 0.f;
-  if (i0 < total) {
-    int c = (int)(i0 % C);
-    float m = mean[c], is = invstd[c];
-    for (long long i = i0; i < total; i += stride) {
-      float g = to_f32(gy[i]);
-      sdy += g;
-      sdyx += g * (to_f32(x[i]) - m) * is;
+  int tid = threadIdx.x;
+  int c0 = (int)(i0 % Cv) * VEC;
+  float m[VEC], is[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    m[j] = mean[c0 + j];
+    is[j] = invstd[c0 + j];
+  }
+  float sdy[VEC] = {}, sdyx[VEC] = {};
+  for (long long i = i0; i < total; i += stride) {
+    V a = xv[i], g = gv[i];
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float gg = to_f32(g.v[j]);
+      sdy[j] += gg;
+      sdyx[j] += gg * (to_f32(a.v[j]) - m[j]) * is[j];
     }
-    atomicAdd(&ws[c], sdy);
-    atomicAdd(&ws[C + c], sdyx);
+  }
+  const int members = 256 / Cv;
+  const int g8 = tid % Cv;
+#pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    float* src = pass == 0 ? sdy : sdyx;
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) red[tid * VEC + j] = src[j];
+    __syncthreads();
+    if (tid < Cv) {
+      float acc[VEC] = {};
+      for (int k = 0; k < members; ++k)
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) acc[j] += red[(g8 + k * Cv) * VEC + j];
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        atomicAdd(&ws[pass * C + c0 + j], acc[j]);
+    }
   }
 }
 
@@ -152,19 +209,6 @@ void check_nhwc(const at::Tensor& x) {
   TORCH_CHECK(x.dim() == 4, "bn expects 4D input");
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
               "bn expects channels_last input");
-  int C = x.size(1);
-  TORCH_CHECK(256 % C == 0 || C % 256 == 0,
-              "bn kernel needs C | 256 or 256 | C, got C=", C);
-}
-
-int bn_grid(long long total, int C) {
-  int g = grid_1d(total, 256, 512);
-  // keep gridDim*256 a multiple of C so each thread's channel is fixed
-  if (C > 256) {
-    int mult = C / 256;
-    g = std::max(mult, (g / mult) * mult);
-  }
-  return g;
 }
 
 }  // namespace
@@ -190,9 +234,13 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
   if (training) {
     auto ws = at::zeros({2, C}, fopts);
     DISPATCH_F32_BF16(x, "bn_partials", {
-      hipLaunchKernelGGL((bn_partials_kernel<scalar_t>),
-                         dim3(bn_grid(M * C, C)), dim3(256), 0, stream,
-                         (const scalar_t*)x.data_ptr(), M, C,
+      constexpr int VEC = 16 / sizeof(scalar_t);
+      TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
+                  "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
+                  ") | 256, got C=", C);
+      hipLaunchKernelGGL((bn_partials_kernel<scalar_t, VEC>),
+                         dim3(grid_1d(M * C / VEC, 256, 304)), dim3(256), 0,
+                         stream, (const scalar_t*)x.data_ptr(), M, C / VEC,
                          ws.data_ptr<float>());
     });
     hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
@@ -241,10 +289,14 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
   auto stream = fedkit_stream();
   auto gamma_f = gamma.contiguous();
   DISPATCH_F32_BF16(x, "bn_bwd_partials", {
-    hipLaunchKernelGGL((bn_bwd_partials_kernel<scalar_t>),
-                       dim3(bn_grid(M * C, C)), dim3(256), 0, stream,
-                       (const scalar_t*)x.data_ptr(),
-                       (const scalar_t*)gy.data_ptr(), M, C,
+    constexpr int VEC = 16 / sizeof(scalar_t);
+    TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
+                "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
+                ") | 256, got C=", C);
+    hipLaunchKernelGGL((bn_bwd_partials_kernel<scalar_t, VEC>),
+                       dim3(grid_1d(M * C / VEC, 256, 304)), dim3(256), 0,
+                       stream, (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)gy.data_ptr(), M, C / VEC,
                        save_mean.data_ptr<float>(),
                        save_invstd.data_ptr<float>(), ws.data_ptr<float>());
   });

@@ -66,8 +66,20 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   const int lane = tid & 63;
   const long long M = (long long)N * P * Q;
 
-  const int bm = blockIdx.x;               // M tile
-  const int bn = blockIdx.y;               // Kout tile
+  // XCD-aware block swizzle (guide T1, bijective form): the dispatcher
+  // places block b on XCD b%8; remap so each XCD owns a CONTIGUOUS chunk of
+  // the (bm, bn) space and neighbor tiles share its private L2.
+  int bm, bn;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    const int orig = blockIdx.y * gridDim.x + blockIdx.x;
+    const int q8 = nwg / 8, r8 = nwg % 8;
+    const int xcd = orig % 8, idx = orig / 8;
+    const int swz = (xcd < r8 ? xcd * (q8 + 1)
+                              : r8 * (q8 + 1) + (xcd - r8) * q8) + idx;
+    bm = swz % gridDim.x;                  // M tile
+    bn = swz / gridDim.x;                  // Kout tile
+  }
 
   // ---- staging geometry: each thread owns (BM*BK)/(256*8) A-slots and
   //      (BN*BK)/(256*8) B-slots of 8 bf16 (16 B) each, fixed across the
@@ -190,41 +202,34 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
 
 // ---------------------------------------------------------------- transforms
 
-// zero-pad NHWC: out[n][h+pt][w+pl][c] = in[n][h][w][c]
-template <typename T>
-__global__ void pad_nhwc_kernel(const T* __restrict__ in, T* __restrict__ out,
-                                int N, int H, int W, int C, int Hp, int Wp,
-                                int pt, int pl) {
-  long long total = (long long)N * H * W * C;
-  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
-       i < total; i += (long long)gridDim.x * blockDim.x) {
-    int c = (int)(i % C);
-    long long t = i / C;
-    int w = (int)(t % W);
-    t /= W;
-    int h = (int)(t % H);
-    int n = (int)(t / H);
-    out[(((long long)n * Hp + h + pt) * Wp + (w + pl)) * C + c] = in[i];
-  }
-}
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16v8;
 
-// zero-dilate + pad NHWC (for bwd-data): out[n][h*str+pt][w*str+pl][c] = in
-template <typename T>
-__global__ void dilate_pad_nhwc_kernel(const T* __restrict__ in,
-                                       T* __restrict__ out, int N, int H,
-                                       int W, int C, int Hp, int Wp, int pt,
-                                       int pl, int str) {
-  long long total = (long long)N * H * W * C;
+// OUTPUT-indexed zero-pad/dilate NHWC, 16 B/lane (8 bf16 = one C-chunk):
+// out[n][h*str+pt][w*str+pl][c] = in[n][h][w][c], everything else 0.
+// str=1 is plain padding.  Writes the whole output (no separate zero fill).
+__global__ void dilate_pad_out_kernel(const bf16* __restrict__ in,
+                                      bf16* __restrict__ out, int N, int H,
+                                      int W, int Cv /* C/8 */, int Hp, int Wp,
+                                      int pt, int pl, int str) {
+  const bf16v8* inv = reinterpret_cast<const bf16v8*>(in);
+  bf16v8* outv = reinterpret_cast<bf16v8*>(out);
+  long long total = (long long)N * Hp * Wp * Cv;
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
        i < total; i += (long long)gridDim.x * blockDim.x) {
-    int c = (int)(i % C);
-    long long t = i / C;
-    int w = (int)(t % W);
-    t /= W;
-    int h = (int)(t % H);
-    int n = (int)(t / H);
-    out[(((long long)n * Hp + h * str + pt) * Wp + (w * str + pl)) * C + c] =
-        in[i];
+    int cv = (int)(i % Cv);
+    long long t = i / Cv;
+    int wp = (int)(t % Wp);
+    t /= Wp;
+    int hp = (int)(t % Hp);
+    int n = (int)(t / Hp);
+    int hs = hp - pt, ws = wp - pl;
+    bf16v8 v = {};
+    if (hs >= 0 && ws >= 0 && hs % str == 0 && ws % str == 0) {
+      int h = hs / str, w = ws / str;
+      if (h < H && w < W)
+        v = inv[(((long long)n * H + h) * W + w) * Cv + cv];
+    }
+    outv[i] = v;
   }
 }
 
@@ -245,56 +250,55 @@ __global__ void rot_weight_kernel(const T* __restrict__ w, T* __restrict__ wr,
   }
 }
 
-// im2col from the PADDED input (for bwd-weight GEMM):
+// im2col from the PADDED input (for bwd-weight GEMM), 16 B/lane vectors
+// (the 8-chunk never crosses a (r,s) boundary since C % 8 == 0):
 // col[m][r*S*C + s*C + c] = xp[n][p*stride+r][q*stride+s][c], m=(n,p,q)
-template <typename T>
-__global__ void im2col_kernel(const T* __restrict__ xp, T* __restrict__ col,
-                              int N, int Hp, int Wp, int C, int R, int S,
-                              int P, int Q, int stride) {
+__global__ void im2col_kernel(const bf16* __restrict__ xp,
+                              bf16* __restrict__ col, int N, int Hp, int Wp,
+                              int C, int R, int S, int P, int Q, int stride) {
+  const bf16v8* xv = reinterpret_cast<const bf16v8*>(xp);
+  bf16v8* cv = reinterpret_cast<bf16v8*>(col);
   long long M = (long long)N * P * Q;
-  long long Kg = (long long)R * S * C;
-  long long total = M * Kg;
+  long long Kv = (long long)R * S * C / 8;
+  long long total = M * Kv;
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
        i < total; i += (long long)gridDim.x * blockDim.x) {
-    long long m = i / Kg;
-    int k = (int)(i % Kg);
-    int c = k % C;
-    int rs = k / C;
+    long long m = i / Kv;
+    int k8 = (int)(i % Kv);
+    int c8 = k8 % (C / 8);
+    int rs = k8 / (C / 8);
     int s = rs % S;
     int r = rs / S;
     int q = (int)(m % Q);
     int p = (int)((m / Q) % P);
     int n = (int)(m / ((long long)P * Q));
-    col[i] = xp[(((long long)n * Hp + p * stride + r) * Wp +
-                 (q * stride + s)) * C + c];
+    cv[i] = xv[((((long long)n * Hp + p * stride + r) * Wp +
+                 (q * stride + s)) * C) / 8 + c8];
   }
 }
 
-at::Tensor pad_nhwc(const at::Tensor& x, int pt, int pb, int pl, int pr) {
+at::Tensor dilate_pad_core(const at::Tensor& x, int pt, int pb, int pl,
+                           int pr, int str) {
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  int Hp = H + pt + pb, Wp = W + pl + pr;
-  auto xp = at::zeros({N, C, Hp, Wp},
+  TORCH_CHECK(C % 8 == 0, "pad kernel needs C % 8 == 0");
+  int Hp = (H - 1) * str + 1 + pt + pb, Wp = (W - 1) * str + 1 + pl + pr;
+  auto xp = at::empty({N, C, Hp, Wp},
                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto stream = fedkit_stream();
-  long long total = (long long)N * H * W * C;
-  hipLaunchKernelGGL((pad_nhwc_kernel<bf16>), dim3(grid_1d(total, 256)),
+  long long total = (long long)N * Hp * Wp * C / 8;
+  hipLaunchKernelGGL(dilate_pad_out_kernel, dim3(grid_1d(total, 256)),
                      dim3(256), 0, stream, (const bf16*)x.data_ptr(),
-                     (bf16*)xp.data_ptr(), N, H, W, C, Hp, Wp, pt, pl);
+                     (bf16*)xp.data_ptr(), N, H, W, C / 8, Hp, Wp, pt, pl, str);
   return xp;
+}
+
+at::Tensor pad_nhwc(const at::Tensor& x, int pt, int pb, int pl, int pr) {
+  return dilate_pad_core(x, pt, pb, pl, pr, 1);
 }
 
 at::Tensor dilate_pad_nhwc(const at::Tensor& x, int pt, int pb, int pl, int pr,
                            int str) {
-  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  int Hp = (H - 1) * str + 1 + pt + pb, Wp = (W - 1) * str + 1 + pl + pr;
-  auto xp = at::zeros({N, C, Hp, Wp},
-                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
-  auto stream = fedkit_stream();
-  long long total = (long long)N * H * W * C;
-  hipLaunchKernelGGL((dilate_pad_nhwc_kernel<bf16>), dim3(grid_1d(total, 256)),
-                     dim3(256), 0, stream, (const bf16*)x.data_ptr(),
-                     (bf16*)xp.data_ptr(), N, H, W, C, Hp, Wp, pt, pl, str);
-  return xp;
+  return dilate_pad_core(x, pt, pb, pl, pr, str);
 }
 
 void check_conv_inputs(const at::Tensor& x, const at::Tensor& w) {
@@ -405,15 +409,30 @@ at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
     col = at::empty({M, (long long)R * S * C},
                     x.options().memory_format(at::MemoryFormat::Contiguous));
     auto stream = fedkit_stream();
-    long long total = M * R * S * C;
-    hipLaunchKernelGGL((im2col_kernel<bf16>), dim3(grid_1d(total, 512)),
+    long long total = M * R * S * C / 8;
+    hipLaunchKernelGGL(im2col_kernel, dim3(grid_1d(total, 512)),
                        dim3(256), 0, stream, (const bf16*)xp.data_ptr(),
                        (bf16*)col.data_ptr(), N, xp.size(2), xp.size(3), C,
                        R, S, P, Q, (int)stride);
   }
-  // dy^T @ col: plain library GEMM (rocBLAS/hipBLASLt) — fp32 accumulate
+  // dw = dy^T @ col is a tall-skinny reduction GEMM ([K<=512, RSC<=4608]
+  // output, reduction dim up to 131072).  Tensile picks no split-K for that
+  // shape, so split it ourselves: batch the reduction dim into chunks, bmm
+  // into fp32 partials, sum — a plain library GEMM per chunk.
   auto dy2d = gy.permute({0, 2, 3, 1}).reshape({M, K});
-  auto dw = at::matmul(dy2d.t(), col);             // [K, R*S*C] bf16
+  long long Kg2 = (long long)R * S * C;
+  int chunks = 1;
+  while (chunks < 64 && (M / (chunks * 2)) >= 2048 && M % (chunks * 2) == 0)
+    chunks *= 2;
+  at::Tensor dw;
+  if (chunks > 1) {
+    auto dyb = dy2d.reshape({chunks, M / chunks, K});
+    auto colb = col.reshape({chunks, M / chunks, Kg2});
+    auto partial = at::bmm(dyb.transpose(1, 2), colb);  // [chunks, K, Kg2]
+    dw = partial.sum(0, false, at::kFloat).to(at::kBFloat16);
+  } else {
+    dw = at::matmul(dy2d.t(), col);
+  }
   return dw.reshape({K, R, S, C}).permute({0, 3, 1, 2})
       .contiguous(at::MemoryFormat::ChannelsLast);
 }
