@@ -204,15 +204,18 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16v8;
 
-// OUTPUT-indexed zero-pad/dilate NHWC, 16 B/lane (8 bf16 = one C-chunk):
+// OUTPUT-indexed zero-pad/dilate NHWC, 16 B/lane (8 bf16 = one C-chunk;
+// VEC=1 scalar fallback for C % 8 != 0, e.g. conv1's C=3):
 // out[n][h*str+pt][w*str+pl][c] = in[n][h][w][c], everything else 0.
 // str=1 is plain padding.  Writes the whole output (no separate zero fill).
+template <int VEC>
 __global__ void dilate_pad_out_kernel(const bf16* __restrict__ in,
                                       bf16* __restrict__ out, int N, int H,
-                                      int W, int Cv /* C/8 */, int Hp, int Wp,
-                                      int pt, int pl, int str) {
-  const bf16v8* inv = reinterpret_cast<const bf16v8*>(in);
-  bf16v8* outv = reinterpret_cast<bf16v8*>(out);
+                                      int W, int Cv /* C/VEC */, int Hp,
+                                      int Wp, int pt, int pl, int str) {
+  using V = __attribute__((ext_vector_type(VEC))) __bf16;
+  const V* inv = reinterpret_cast<const V*>(in);
+  V* outv = reinterpret_cast<V*>(out);
   long long total = (long long)N * Hp * Wp * Cv;
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
        i < total; i += (long long)gridDim.x * blockDim.x) {
@@ -223,7 +226,7 @@ __global__ void dilate_pad_out_kernel(const bf16* __restrict__ in,
     int hp = (int)(t % Hp);
     int n = (int)(t / Hp);
     int hs = hp - pt, ws = wp - pl;
-    bf16v8 v = {};
+    V v = {};
     if (hs >= 0 && ws >= 0 && hs % str == 0 && ws % str == 0) {
       int h = hs / str, w = ws / str;
       if (h < H && w < W)
@@ -251,44 +254,54 @@ __global__ void rot_weight_kernel(const T* __restrict__ w, T* __restrict__ wr,
 }
 
 // im2col from the PADDED input (for bwd-weight GEMM), 16 B/lane vectors
-// (the 8-chunk never crosses a (r,s) boundary since C % 8 == 0):
+// (the 8-chunk never crosses a (r,s) boundary since C % 8 == 0; VEC=1
+// scalar fallback for small C):
 // col[m][r*S*C + s*C + c] = xp[n][p*stride+r][q*stride+s][c], m=(n,p,q)
+template <int VEC>
 __global__ void im2col_kernel(const bf16* __restrict__ xp,
                               bf16* __restrict__ col, int N, int Hp, int Wp,
                               int C, int R, int S, int P, int Q, int stride) {
-  const bf16v8* xv = reinterpret_cast<const bf16v8*>(xp);
-  bf16v8* cv = reinterpret_cast<bf16v8*>(col);
+  using V = __attribute__((ext_vector_type(VEC))) __bf16;
+  const V* xv = reinterpret_cast<const V*>(xp);
+  V* cv = reinterpret_cast<V*>(col);
   long long M = (long long)N * P * Q;
-  long long Kv = (long long)R * S * C / 8;
+  long long Kv = (long long)R * S * C / VEC;
   long long total = M * Kv;
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
        i < total; i += (long long)gridDim.x * blockDim.x) {
     long long m = i / Kv;
     int k8 = (int)(i % Kv);
-    int c8 = k8 % (C / 8);
-    int rs = k8 / (C / 8);
+    int c8 = k8 % (C / VEC);
+    int rs = k8 / (C / VEC);
     int s = rs % S;
     int r = rs / S;
     int q = (int)(m % Q);
     int p = (int)((m / Q) % P);
     int n = (int)(m / ((long long)P * Q));
     cv[i] = xv[((((long long)n * Hp + p * stride + r) * Wp +
-                 (q * stride + s)) * C) / 8 + c8];
+                 (q * stride + s)) * C) / VEC + c8];
   }
 }
 
 at::Tensor dilate_pad_core(const at::Tensor& x, int pt, int pb, int pl,
                            int pr, int str) {
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  TORCH_CHECK(C % 8 == 0, "pad kernel needs C % 8 == 0");
   int Hp = (H - 1) * str + 1 + pt + pb, Wp = (W - 1) * str + 1 + pl + pr;
   auto xp = at::empty({N, C, Hp, Wp},
                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto stream = fedkit_stream();
-  long long total = (long long)N * Hp * Wp * C / 8;
-  hipLaunchKernelGGL(dilate_pad_out_kernel, dim3(grid_1d(total, 256)),
-                     dim3(256), 0, stream, (const bf16*)x.data_ptr(),
-                     (bf16*)xp.data_ptr(), N, H, W, C / 8, Hp, Wp, pt, pl, str);
+  if (C % 8 == 0) {
+    long long total = (long long)N * Hp * Wp * C / 8;
+    hipLaunchKernelGGL(dilate_pad_out_kernel<8>, dim3(grid_1d(total, 256)),
+                       dim3(256), 0, stream, (const bf16*)x.data_ptr(),
+                       (bf16*)xp.data_ptr(), N, H, W, C / 8, Hp, Wp, pt, pl,
+                       str);
+  } else {
+    long long total = (long long)N * Hp * Wp * C;
+    hipLaunchKernelGGL(dilate_pad_out_kernel<1>, dim3(grid_1d(total, 256)),
+                       dim3(256), 0, stream, (const bf16*)x.data_ptr(),
+                       (bf16*)xp.data_ptr(), N, H, W, C, Hp, Wp, pt, pl, str);
+  }
   return xp;
 }
 
@@ -409,11 +422,19 @@ at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
     col = at::empty({M, (long long)R * S * C},
                     x.options().memory_format(at::MemoryFormat::Contiguous));
     auto stream = fedkit_stream();
-    long long total = M * R * S * C / 8;
-    hipLaunchKernelGGL(im2col_kernel, dim3(grid_1d(total, 512)),
-                       dim3(256), 0, stream, (const bf16*)xp.data_ptr(),
-                       (bf16*)col.data_ptr(), N, xp.size(2), xp.size(3), C,
-                       R, S, P, Q, (int)stride);
+    if (C % 8 == 0) {
+      long long total = M * R * S * C / 8;
+      hipLaunchKernelGGL(im2col_kernel<8>, dim3(grid_1d(total, 512)),
+                         dim3(256), 0, stream, (const bf16*)xp.data_ptr(),
+                         (bf16*)col.data_ptr(), N, xp.size(2), xp.size(3), C,
+                         R, S, P, Q, (int)stride);
+    } else {
+      long long total = M * R * S * C;
+      hipLaunchKernelGGL(im2col_kernel<1>, dim3(grid_1d(total, 512)),
+                         dim3(256), 0, stream, (const bf16*)xp.data_ptr(),
+                         (bf16*)col.data_ptr(), N, xp.size(2), xp.size(3), C,
+                         R, S, P, Q, (int)stride);
+    }
   }
   // dw = dy^T @ col is a tall-skinny reduction GEMM ([K<=512, RSC<=4608]
   // output, reduction dim up to 131072).  Tensile picks no split-K for that
