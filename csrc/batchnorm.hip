@@ -19,14 +19,16 @@ namespace {
 template <typename T, int VEC>
 struct alignas(sizeof(T) * VEC) VecT { T v[VEC]; };
 
-// Vectorized (16 B/lane) per-channel reduction: each thread owns VEC
-// consecutive channels fixed across its grid-stride walk (C/VEC divides the
-// 256-thread block), accumulates in registers, reduces across the block via
-// LDS, then one atomicAdd per (block, channel) into the [2][C] workspace.
+// Vectorized (16 B/lane) per-channel reduction, stage 1: each thread owns
+// VEC consecutive channels fixed across its grid-stride walk (C/VEC divides
+// the 256-thread block), accumulates in registers (4x unrolled so four 16-B
+// loads are in flight per lane), reduces across the block via LDS, and each
+// block writes its private partial row [2][C] — NO atomics; stage 2 (the
+// finalize / bwd-reduce kernels) sums the <=1024 partial rows.
 template <typename T, int VEC>
 __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
                                    int Cv /* C/VEC */,
-                                   float* __restrict__ ws /* [2][C] */) {
+                                   float* __restrict__ part /* [nb][2][C] */) {
   using V = VecT<T, VEC>;
   const int C = Cv * VEC;
   const V* xv = reinterpret_cast<const V*>(x);
@@ -37,7 +39,19 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
   int tid = threadIdx.x;
   int c0 = (int)(i0 % Cv) * VEC;
   float s[VEC] = {}, sq[VEC] = {};
-  for (long long i = i0; i < total; i += stride) {
+  long long i = i0;
+  for (; i + 3 * stride < total; i += 4 * stride) {
+    V a0 = xv[i], a1 = xv[i + stride], a2 = xv[i + 2 * stride],
+      a3 = xv[i + 3 * stride];
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float v0 = to_f32(a0.v[j]), v1 = to_f32(a1.v[j]);
+      float v2 = to_f32(a2.v[j]), v3 = to_f32(a3.v[j]);
+      s[j] += (v0 + v1) + (v2 + v3);
+      sq[j] += (v0 * v0 + v1 * v1) + (v2 * v2 + v3 * v3);
+    }
+  }
+  for (; i < total; i += stride) {
     V a = xv[i];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
@@ -48,6 +62,7 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
   }
   const int members = 256 / Cv;
   const int g = tid % Cv;
+  float* out = part + (long long)blockIdx.x * 2 * C;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     float* src = pass == 0 ? s : sq;
@@ -61,15 +76,15 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
 #pragma unroll
         for (int j = 0; j < VEC; ++j) acc[j] += red[(g + k * Cv) * VEC + j];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j)
-        atomicAdd(&ws[pass * C + c0 + j], acc[j]);
+      for (int j = 0; j < VEC; ++j) out[pass * C + c0 + j] = acc[j];
     }
   }
 }
 
-__global__ void bn_finalize_kernel(const float* __restrict__ ws, int C,
-                                   long long count, float eps, float momentum,
-                                   bool training, bool track,
+// stage 2: sum the per-block partial rows, then mean/invstd + running update
+__global__ void bn_finalize_kernel(const float* __restrict__ part, int nb,
+                                   int C, long long count, float eps,
+                                   float momentum, bool training, bool track,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var,
                                    float* __restrict__ save_mean,
@@ -78,8 +93,13 @@ __global__ void bn_finalize_kernel(const float* __restrict__ ws, int C,
   if (c >= C) return;
   float mean, var;
   if (training) {
-    mean = ws[c] / count;
-    var = fmaxf(ws[C + c] / count - mean * mean, 0.f);  // biased
+    float s = 0.f, sq = 0.f;
+    for (int b = 0; b < nb; ++b) {
+      s += part[(long long)b * 2 * C + c];
+      sq += part[(long long)b * 2 * C + C + c];
+    }
+    mean = s / count;
+    var = fmaxf(sq / count - mean * mean, 0.f);  // biased
     if (track) {
       float unbiased = count > 1 ? var * count / (count - 1) : var;
       running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
@@ -91,6 +111,16 @@ __global__ void bn_finalize_kernel(const float* __restrict__ ws, int C,
   }
   save_mean[c] = mean;
   save_invstd[c] = rsqrtf(var + eps);
+}
+
+// stage 2 for backward: ws[2][C] = sum of partial rows
+__global__ void bn_reduce_kernel(const float* __restrict__ part, int nb,
+                                 int C2 /* 2*C */, float* __restrict__ ws) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C2) return;
+  float s = 0.f;
+  for (int b = 0; b < nb; ++b) s += part[(long long)b * C2 + c];
+  ws[c] = s;
 }
 
 template <typename T, int VEC>
@@ -122,7 +152,7 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
                                        const T* __restrict__ gy, long long M,
                                        int Cv, const float* __restrict__ mean,
                                        const float* __restrict__ invstd,
-                                       float* __restrict__ ws /* [2][C] */) {
+                                       float* __restrict__ part /* [nb][2][C] */) {
   using V = VecT<T, VEC>;
   const int C = Cv * VEC;
   const V* xv = reinterpret_cast<const V*>(x);
@@ -140,7 +170,18 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
     is[j] = invstd[c0 + j];
   }
   float sdy[VEC] = {}, sdyx[VEC] = {};
-  for (long long i = i0; i < total; i += stride) {
+  long long i = i0;
+  for (; i + stride < total; i += 2 * stride) {
+    V a0 = xv[i], g0 = gv[i], a1 = xv[i + stride], g1 = gv[i + stride];
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float gg0 = to_f32(g0.v[j]), gg1 = to_f32(g1.v[j]);
+      sdy[j] += gg0 + gg1;
+      sdyx[j] += gg0 * (to_f32(a0.v[j]) - m[j]) * is[j]
+               + gg1 * (to_f32(a1.v[j]) - m[j]) * is[j];
+    }
+  }
+  for (; i < total; i += stride) {
     V a = xv[i], g = gv[i];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
@@ -151,6 +192,7 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
   }
   const int members = 256 / Cv;
   const int g8 = tid % Cv;
+  float* out = part + (long long)blockIdx.x * 2 * C;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     float* src = pass == 0 ? sdy : sdyx;
@@ -164,8 +206,7 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
 #pragma unroll
         for (int j = 0; j < VEC; ++j) acc[j] += red[(g8 + k * Cv) * VEC + j];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j)
-        atomicAdd(&ws[pass * C + c0 + j], acc[j]);
+      for (int j = 0; j < VEC; ++j) out[pass * C + c0 + j] = acc[j];
     }
   }
 }
@@ -232,28 +273,29 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
   TORCH_CHECK(gamma_f.scalar_type() == at::kFloat, "bn gamma must be fp32");
 
   if (training) {
-    auto ws = at::zeros({2, C}, fopts);
     DISPATCH_F32_BF16(x, "bn_partials", {
       constexpr int VEC = 16 / sizeof(scalar_t);
       TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
                   "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                   ") | 256, got C=", C);
+      int nb = grid_1d(M * C / VEC, 256, 1024);
+      auto part = at::empty({nb, 2, C}, fopts);
       hipLaunchKernelGGL((bn_partials_kernel<scalar_t, VEC>),
-                         dim3(grid_1d(M * C / VEC, 256, 304)), dim3(256), 0,
+                         dim3(nb), dim3(256), 0,
                          stream, (const scalar_t*)x.data_ptr(), M, C / VEC,
-                         ws.data_ptr<float>());
+                         part.data_ptr<float>());
+      hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                         0, stream, part.data_ptr<float>(), nb, C, M,
+                         (float)eps, (float)momentum, true,
+                         running_mean.defined(),
+                         running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
+                         running_var.defined() ? running_var.data_ptr<float>() : nullptr,
+                         save_mean.data_ptr<float>(),
+                         save_invstd.data_ptr<float>());
     });
-    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
-                       0, stream, ws.data_ptr<float>(), C, M, (float)eps,
-                       (float)momentum, true, running_mean.defined(),
-                       running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
-                       running_var.defined() ? running_var.data_ptr<float>() : nullptr,
-                       save_mean.data_ptr<float>(),
-                       save_invstd.data_ptr<float>());
   } else {
-    auto ws = at::zeros({1}, fopts);  // unused
     hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
-                       0, stream, ws.data_ptr<float>(), C, M, (float)eps,
+                       0, stream, (const float*)nullptr, 0, C, M, (float)eps,
                        (float)momentum, false, false,
                        running_mean.data_ptr<float>(),
                        running_var.data_ptr<float>(),
@@ -284,7 +326,7 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   long long M = (long long)N * H * W;
   auto fopts = x.options().dtype(at::kFloat);
-  auto ws = at::zeros({2, C}, fopts);
+  auto ws = at::empty({2, C}, fopts);
   auto gx = at::empty_like(x);
   auto stream = fedkit_stream();
   auto gamma_f = gamma.contiguous();
@@ -293,12 +335,17 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
     TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
                 "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                 ") | 256, got C=", C);
+    int nb = grid_1d(M * C / VEC, 256, 1024);
+    auto part = at::empty({nb, 2, C}, fopts);
     hipLaunchKernelGGL((bn_bwd_partials_kernel<scalar_t, VEC>),
-                       dim3(grid_1d(M * C / VEC, 256, 304)), dim3(256), 0,
+                       dim3(nb), dim3(256), 0,
                        stream, (const scalar_t*)x.data_ptr(),
                        (const scalar_t*)gy.data_ptr(), M, C / VEC,
                        save_mean.data_ptr<float>(),
-                       save_invstd.data_ptr<float>(), ws.data_ptr<float>());
+                       save_invstd.data_ptr<float>(), part.data_ptr<float>());
+    hipLaunchKernelGGL(bn_reduce_kernel, dim3((2 * C + 255) / 256), dim3(256),
+                       0, stream, part.data_ptr<float>(), nb, 2 * C,
+                       ws.data_ptr<float>());
   });
   DISPATCH_F32_BF16(x, "bn_bwd_apply", {
     constexpr int VEC = 16 / sizeof(scalar_t);

@@ -368,6 +368,25 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
 at::Tensor fedkit_conv_small_fwd(const at::Tensor& x, const at::Tensor& w,
                                  long stride, long padding);  // conv_small.hip
 
+at::Tensor fedkit_conv2d_pad_input(const at::Tensor& x, long padding) {
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "pad_input expects channels_last");
+  if (padding == 0) return x;
+  return pad_nhwc(x, padding, padding, padding, padding);
+}
+
+// forward on a PRE-PADDED input (the autograd wrapper saves xp so backward
+// never re-pads): P = (Hp - R)/stride + 1
+at::Tensor fedkit_conv2d_fwd_prepadded(const at::Tensor& xp,
+                                       const at::Tensor& w, long stride) {
+  check_conv_inputs(xp, w);
+  int Hp = xp.size(2), Wp = xp.size(3);
+  int R = w.size(2), S = w.size(3);
+  int P = (Hp - R) / (int)stride + 1;
+  int Q = (Wp - S) / (int)stride + 1;
+  return conv_core(xp, w, (int)stride, P, Q);
+}
+
 at::Tensor fedkit_conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                              long stride, long padding) {
   check_conv_inputs(x, w);
@@ -404,16 +423,16 @@ at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
   return conv_core(gyp, wrot, 1, (int)H, (int)W);
 }
 
-at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
-                                    long stride, long padding, long R_in,
-                                    long S_in) {
-  check_conv_inputs(gy, x);
-  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+// bwd-weight from the PRE-PADDED input saved by the forward
+at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
+                                              const at::Tensor& xp,
+                                              long stride, long R_in,
+                                              long S_in) {
+  check_conv_inputs(gy, xp);
+  int N = xp.size(0), C = xp.size(1);
   int K = gy.size(1), P = gy.size(2), Q = gy.size(3);
   int R = (int)R_in, S = (int)S_in;
   long long M = (long long)N * P * Q;
-  at::Tensor xp = padding > 0 ? pad_nhwc(x, padding, padding, padding, padding)
-                              : x;
   at::Tensor col;
   if (R == 1 && S == 1 && stride == 1) {
     col = xp.permute({0, 2, 3, 1}).reshape({M, C});  // NHWC view, no copy
@@ -455,4 +474,12 @@ at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
   }
   return dw.reshape({K, R, S, C}).permute({0, 3, 1, 2})
       .contiguous(at::MemoryFormat::ChannelsLast);
+}
+
+at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
+                                    long stride, long padding, long R_in,
+                                    long S_in) {
+  at::Tensor xp = padding > 0 ? pad_nhwc(x, padding, padding, padding, padding)
+                              : x;
+  return fedkit_conv2d_bwd_weight_prepadded(gy, xp, stride, R_in, S_in);
 }

@@ -33,6 +33,12 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
 
 at::Tensor fedkit_conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                              long stride, long padding);
+at::Tensor fedkit_conv2d_pad_input(const at::Tensor& x, long padding);
+at::Tensor fedkit_conv2d_fwd_prepadded(const at::Tensor& xp,
+                                       const at::Tensor& w, long stride);
+at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
+                                              const at::Tensor& xp,
+                                              long stride, long R, long S);
 at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
                                   long stride, long padding, long H, long W);
 at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
@@ -55,6 +61,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "NHWC BatchNorm bwd: returns (gx, gw, gb)");
   m.def("conv2d_fwd", &fedkit_conv2d_fwd,
         "NHWC implicit-GEMM conv fwd on MFMA (3x3/1x1, stride 1/2)");
+  m.def("conv2d_pad_input", &fedkit_conv2d_pad_input, "zero-pad NHWC input");
+  m.def("conv2d_fwd_prepadded", &fedkit_conv2d_fwd_prepadded,
+        "conv fwd on a pre-padded input");
+  m.def("conv2d_bwd_weight_prepadded", &fedkit_conv2d_bwd_weight_prepadded,
+        "conv bwd-weight from the saved padded input");
   m.def("conv2d_bwd_data", &fedkit_conv2d_bwd_data, "conv bwd-data");
   m.def("conv2d_bwd_weight", &fedkit_conv2d_bwd_weight, "conv bwd-weight");
 }

@@ -32,24 +32,39 @@ def _ext():
 class _ConvFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, stride, padding):
-        # x: [N,C,H,W] logical, carried NHWC-contiguous; w: [K,C,R,S]
-        y = _ext().conv2d_fwd(x, w, stride, padding)
-        ctx.save_for_backward(x, w)
+        # x: [N,C,H,W] logical, carried NHWC-contiguous; w: [K,C,R,S].
+        # The padded input is computed ONCE and saved, so bwd-weight reuses
+        # it instead of re-padding (and x itself need not be kept).
+        ext = _ext()
+        small_c = x.shape[1] % 8 != 0
+        if small_c:
+            y = ext.conv2d_fwd(x, w, stride, padding)
+            ctx.save_for_backward(x, w)
+        else:
+            xp = ext.conv2d_pad_input(x, padding)
+            y = ext.conv2d_fwd_prepadded(xp, w, stride)
+            ctx.save_for_backward(xp, w)
+        ctx.small_c = small_c
         ctx.stride = stride
         ctx.padding = padding
+        ctx.hw = (x.shape[2], x.shape[3])
         return y
 
     @staticmethod
     def backward(ctx, gy):
-        x, w = ctx.saved_tensors
+        xsaved, w = ctx.saved_tensors
         gy = gy.contiguous(memory_format=torch.channels_last)
         gx = gw = None
         if ctx.needs_input_grad[0]:
             gx = _ext().conv2d_bwd_data(gy, w, ctx.stride, ctx.padding,
-                                        x.shape[2], x.shape[3])
+                                        ctx.hw[0], ctx.hw[1])
         if ctx.needs_input_grad[1]:
-            gw = _ext().conv2d_bwd_weight(gy, x, ctx.stride, ctx.padding,
-                                          w.shape[2], w.shape[3])
+            if ctx.small_c:
+                gw = _ext().conv2d_bwd_weight(gy, xsaved, ctx.stride,
+                                              ctx.padding, w.shape[2], w.shape[3])
+            else:
+                gw = _ext().conv2d_bwd_weight_prepadded(
+                    gy, xsaved, ctx.stride, w.shape[2], w.shape[3])
         return gx, gw, None, None
 
 
