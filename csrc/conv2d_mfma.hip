@@ -438,7 +438,7 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
     col = xp.permute({0, 2, 3, 1}).reshape({M, C});  // NHWC view, no copy
   } else {
     col = at::empty({M, (long long)R * S * C},
-                    x.options().memory_format(at::MemoryFormat::Contiguous));
+                    xp.options().memory_format(at::MemoryFormat::Contiguous));
     auto stream = fedkit_stream();
     if (C % 8 == 0) {
       long long total = M * R * S * C / 8;
