@@ -1,0 +1,182 @@
+#!/usr/bin/env python3
+"""Per-kernel microbenchmark: fedkit CDNA4 kernels vs torch-ROCm (MIOpen/
+rocBLAS) on the flagship ResNet18/CIFAR10 shapes.
+
+Run on a GPU box:
+    python bench_kernels.py [--iters 50] [--op conv|bn|all]
+
+Prints one line per (op, shape): fedkit time, eager time, ratio, and the
+achieved TFLOP/s (conv) or GB/s (bn) against the MI355X roofline (bf16 MFMA
+dense peak ~2.5 PFLOP/s, HBM3E ~8 TB/s).
+"""
+
+import argparse
+import sys
+import os
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+# (C_in, H, K_out, stride) for every distinct conv in ResNet18/CIFAR10 b=128,
+# with multiplicity per fwd pass
+RESNET18_CONVS = [
+    # Cin, H,  Kout, stride, 3x3? count
+    (64, 32, 64, 1, 3, 4),     # layer1 convs
+    (64, 32, 128, 2, 3, 1),    # layer2.0 conv1
+    (64, 32, 128, 2, 1, 1),    # layer2.0 shortcut
+    (128, 16, 128, 1, 3, 3),
+    (128, 16, 256, 2, 3, 1),
+    (128, 16, 256, 2, 1, 1),
+    (256, 8, 256, 1, 3, 3),
+    (256, 8, 512, 2, 3, 1),
+    (256, 8, 512, 2, 1, 1),
+    (512, 4, 512, 1, 3, 3),
+]
+BN_SHAPES = [(64, 32), (128, 16), (256, 8), (512, 4)]
+BATCH = 128
+
+
+def timeit(fn, iters, warmup=10):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    start = torch.cuda.Event(enable_timing=True)
+    end = torch.cuda.Event(enable_timing=True)
+    start.record()
+    for _ in range(iters):
+        fn()
+    end.record()
+    torch.cuda.synchronize()
+    return start.elapsed_time(end) / iters * 1e3  # us
+
+
+def bench_conv(iters):
+    import fedkit.ops  # noqa
+    ext = fedkit.ops.require_ext()
+    print(f"{'conv shape':38s} {'fed_us':>8s} {'eag_us':>8s} {'ratio':>6s} "
+          f"{'fedTF/s':>8s} {'SOL_us':>7s}")
+    tot_f = tot_e = 0.0
+    for Cin, H, Kout, stride, ks, count in RESNET18_CONVS:
+        pad = 1 if ks == 3 else 0
+        x = torch.randn(BATCH, Cin, H, H, device="cuda", dtype=torch.bfloat16
+                        ).contiguous(memory_format=torch.channels_last)
+        w = torch.randn(Kout, Cin, ks, ks, device="cuda", dtype=torch.bfloat16
+                        ).contiguous(memory_format=torch.channels_last)
+        P = (H + 2 * pad - ks) // stride + 1
+        M = BATCH * P * P
+        flops = 2.0 * M * Kout * ks * ks * Cin
+
+        def fed():
+            return ext.conv2d_fwd(x, w, stride, pad)
+
+        def eag():
+            return torch.nn.functional.conv2d(x, w, None, stride, pad)
+
+        y1, y2 = fed().float(), eag().float()
+        err = (y1 - y2).abs().max().item()
+        t_f = timeit(fed, iters)
+        t_e = timeit(eag, iters)
+        tot_f += t_f * count
+        tot_e += t_e * count
+        name = f"{ks}x{ks} C{Cin} H{H} K{Kout} s{stride} x{count}"
+        print(f"{name:38s} {t_f:8.1f} {t_e:8.1f} {t_e/t_f:6.2f} "
+              f"{flops/t_f/1e6:8.1f} {flops/2.5e15*1e6:7.2f}"
+              + (f"  MAXERR {err:.3f}" if err > 0.5 else ""))
+    print(f"{'TOTAL fwd (weighted)':38s} {tot_f:8.1f} {tot_e:8.1f} "
+          f"{tot_e/tot_f:6.2f}")
+
+
+def bench_conv_bwd(iters):
+    import fedkit.ops  # noqa
+    ext = fedkit.ops.require_ext()
+    print(f"{'conv bwd shape':38s} {'dx_us':>8s} {'dw_us':>8s} "
+          f"{'eagdx':>8s} {'eagdw':>8s}")
+    for Cin, H, Kout, stride, ks, count in RESNET18_CONVS:
+        pad = 1 if ks == 3 else 0
+        x = torch.randn(BATCH, Cin, H, H, device="cuda", dtype=torch.bfloat16
+                        ).contiguous(memory_format=torch.channels_last)
+        w = torch.randn(Kout, Cin, ks, ks, device="cuda", dtype=torch.bfloat16
+                        ).contiguous(memory_format=torch.channels_last)
+        P = (H + 2 * pad - ks) // stride + 1
+        gy = torch.randn(BATCH, Kout, P, P, device="cuda", dtype=torch.bfloat16
+                         ).contiguous(memory_format=torch.channels_last)
+        xp = ext.conv2d_pad_input(x, pad) if Cin % 8 == 0 else x
+
+        def fed_dx():
+            return ext.conv2d_bwd_data(gy, w, stride, pad, H, H)
+
+        def fed_dw():
+            if Cin % 8 == 0:
+                return ext.conv2d_bwd_weight_prepadded(gy, xp, stride, ks, ks)
+            return ext.conv2d_bwd_weight(gy, x, stride, pad, ks, ks)
+
+        xf = x.float().requires_grad_(True)
+        wf = w.float().requires_grad_(True)
+
+        def eager_both():
+            y = torch.nn.functional.conv2d(xf, wf, None, stride, pad)
+            gx, gw = torch.autograd.grad(y, [xf, wf], gy.float())
+            return gx, gw
+
+        t_dx = timeit(fed_dx, iters)
+        t_dw = timeit(fed_dw, iters)
+        t_e = timeit(eager_both, iters)
+        name = f"{ks}x{ks} C{Cin} H{H} K{Kout} s{stride} x{count}"
+        print(f"{name:38s} {t_dx:8.1f} {t_dw:8.1f} {t_e:8.1f} (dx+dw fp32)")
+
+
+def bench_bn(iters):
+    import fedkit.ops  # noqa
+    ext = fedkit.ops.require_ext()
+    print(f"{'bn shape':28s} {'fwd_us':>8s} {'bwd_us':>8s} {'eagf_us':>8s} "
+          f"{'GB/s fwd':>9s}")
+    for C, H in BN_SHAPES:
+        x = torch.randn(BATCH, C, H, H, device="cuda", dtype=torch.bfloat16
+                        ).contiguous(memory_format=torch.channels_last)
+        gy = torch.randn_like(x)
+        gamma = torch.ones(C, device="cuda")
+        beta = torch.zeros(C, device="cuda")
+        rm = torch.zeros(C, device="cuda")
+        rv = torch.ones(C, device="cuda")
+
+        def fed_fwd():
+            return ext.bn_fwd(x, gamma, beta, rm, rv, True, 0.1, 1e-5)
+
+        y, sm, si = fed_fwd()
+
+        def fed_bwd():
+            return ext.bn_bwd(gy, x, gamma, sm, si)
+
+        xf = x.float()
+
+        def eag_fwd():
+            return torch.nn.functional.batch_norm(
+                xf, rm, rv, gamma, beta, True, 0.1, 1e-5)
+
+        t_f = timeit(fed_fwd, iters)
+        t_b = timeit(fed_bwd, iters)
+        t_e = timeit(eag_fwd, iters)
+        nbytes = x.numel() * 2 * 3  # read x twice + write y
+        print(f"C{C} H{H} b{BATCH:28d}"[:28] +
+              f" {t_f:8.1f} {t_b:8.1f} {t_e:8.1f} {nbytes/t_f/1e3:9.1f}")
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--iters", type=int, default=50)
+    ap.add_argument("--op", default="all",
+                    choices=["conv", "convbwd", "bn", "all"])
+    args = ap.parse_args()
+    assert torch.cuda.is_available(), "GPU microbench needs a ROCm GPU"
+    torch.manual_seed(0)
+    if args.op in ("conv", "all"):
+        bench_conv(args.iters)
+    if args.op in ("convbwd", "all"):
+        bench_conv_bwd(args.iters)
+    if args.op in ("bn", "all"):
+        bench_bn(args.iters)
+
+
+if __name__ == "__main__":
+    main()
