@@ -130,8 +130,13 @@ __global__ void bn_colsum_kernel(const float* __restrict__ part, int nb,
   if (lane == 0) ws[col] = s;
 }
 
-template <typename T, int VEC>
-__global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
+// normalize + scale/shift, with optional fused residual add (RES) and ELU
+// epilogue (the reference's `elu(bn(conv) [+ shortcut])` patterns,
+// simple_models.py:150-153) — one pass instead of bn/add/elu separate
+// kernels (each an extra HBM read+write of the activation).
+template <typename T, int VEC, bool ELU, bool RES>
+__global__ void bn_apply_kernel(const T* __restrict__ x,
+                                const T* __restrict__ res, T* __restrict__ y,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
                                 const float* __restrict__ gamma,
@@ -139,15 +144,20 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
                                 long long nvec, int Cv /* C / VEC */) {
   using V = VecT<T, VEC>;
   const V* xv = reinterpret_cast<const V*>(x);
+  const V* rv = reinterpret_cast<const V*>(res);
   V* yv = reinterpret_cast<V*>(y);
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
        i < nvec; i += (long long)gridDim.x * blockDim.x) {
     int c0 = (int)(i % Cv) * VEC;
     V a = xv[i], r;
+    V rr;
+    if (RES) rr = rv[i];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       int c = c0 + j;
       float v = (to_f32(a.v[j]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+      if (RES) v += to_f32(rr.v[j]);
+      if (ELU) v = fedkit_elu_f(v);
       from_f32(v, r.v[j]);
     }
     yv[i] = r;
@@ -266,7 +276,9 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
                                       const at::Tensor& beta,
                                       at::Tensor running_mean,
                                       at::Tensor running_var, bool training,
-                                      double momentum, double eps) {
+                                      double momentum, double eps,
+                                      c10::optional<at::Tensor> residual,
+                                      bool elu) {
   check_nhwc(x);
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   long long M = (long long)N * H * W;
@@ -313,17 +325,35 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
                        save_mean.data_ptr<float>(),
                        save_invstd.data_ptr<float>());
   }
+  const bool has_res = residual.has_value();
+  const void* res_ptr = nullptr;
+  if (has_res) {
+    check_nhwc(*residual);
+    TORCH_CHECK(residual->scalar_type() == x.scalar_type() &&
+                residual->sizes() == x.sizes(),
+                "bn residual must match input shape/dtype");
+    res_ptr = residual->data_ptr();
+  }
   DISPATCH_F32_BF16(x, "bn_apply", {
     constexpr int VEC = 16 / sizeof(scalar_t);
     TORCH_CHECK(C % VEC == 0, "bn needs C % ", VEC, " == 0");
     long long nvec = M * C / VEC;
-    hipLaunchKernelGGL((bn_apply_kernel<scalar_t, VEC>),
-                       dim3(grid_1d(nvec, 256)), dim3(256), 0, stream,
-                       (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(),
-                       save_mean.data_ptr<float>(),
-                       save_invstd.data_ptr<float>(),
-                       gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(),
-                       nvec, C / VEC);
+    auto launch = [&](auto ekind, auto rkind) {
+      hipLaunchKernelGGL(
+          (bn_apply_kernel<scalar_t, VEC, decltype(ekind)::value,
+                           decltype(rkind)::value>),
+          dim3(grid_1d(nvec, 256)), dim3(256), 0, stream,
+          (const scalar_t*)x.data_ptr(), (const scalar_t*)res_ptr,
+          (scalar_t*)y.data_ptr(), save_mean.data_ptr<float>(),
+          save_invstd.data_ptr<float>(), gamma_f.data_ptr<float>(),
+          beta_f.data_ptr<float>(), nvec, C / VEC);
+    };
+    using T0 = std::integral_constant<bool, false>;
+    using T1 = std::integral_constant<bool, true>;
+    if (elu && has_res)       launch(T1{}, T1{});
+    else if (elu)             launch(T1{}, T0{});
+    else if (has_res)         launch(T0{}, T1{});
+    else                      launch(T0{}, T0{});
   });
   return {y, save_mean, save_invstd};
 }

@@ -32,7 +32,11 @@ __device__ __forceinline__ void from_f32(float v, __hip_bfloat16& out) {
   out = __float2bfloat16(v);
 }
 
-#define DISPATCH_F32_BF16(TENSOR, NAME, ...)                                   \
+__device__ __forceinline__ float fedkit_elu_f(float x) {
+  return x > 0.f ? x : __expf(x) - 1.f;
+}
+
+#define DISPATCH_F32_BF16(TENSOR, NAME, ...)                                 \
   do {                                                                         \
     if ((TENSOR).scalar_type() == at::kFloat) {                                \
       using scalar_t = float;                                                  \

@@ -25,7 +25,9 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
                                       const at::Tensor& beta,
                                       at::Tensor running_mean,
                                       at::Tensor running_var, bool training,
-                                      double momentum, double eps);
+                                      double momentum, double eps,
+                                      c10::optional<at::Tensor> residual,
+                                      bool elu);
 std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& gamma,
                                       const at::Tensor& save_mean,
@@ -56,7 +58,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused log-softmax + NLL (mean): returns (loss, lse)");
   m.def("cross_entropy_bwd", &fedkit_cross_entropy_bwd, "CE backward");
   m.def("bn_fwd", &fedkit_bn_fwd,
-        "NHWC BatchNorm fwd: returns (y, save_mean, save_invstd)");
+        "NHWC BatchNorm fwd (optional fused residual add + ELU epilogue): "
+        "returns (y, save_mean, save_invstd)",
+        py::arg("x"), py::arg("gamma"), py::arg("beta"),
+        py::arg("running_mean"), py::arg("running_var"), py::arg("training"),
+        py::arg("momentum"), py::arg("eps"),
+        py::arg("residual") = c10::nullopt, py::arg("elu") = false);
   m.def("bn_bwd", &fedkit_bn_bwd,
         "NHWC BatchNorm bwd: returns (gx, gw, gb)");
   m.def("conv2d_fwd", &fedkit_conv2d_fwd,

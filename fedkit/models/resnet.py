@@ -17,9 +17,8 @@ partition) is identical to the reference's.
 
 import torch.nn as nn
 
-from ..ops.elu import elu
 from ..ops.conv import FedConv2d
-from ..ops.norm import FedBatchNorm2d
+from ..ops.norm import FedBatchNorm2d, bn_elu
 
 
 def _conv3x3(cin, cout, stride=1):
@@ -47,10 +46,8 @@ class BasicBlock(nn.Module):
             )
 
     def forward(self, x):
-        out = elu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
-        out = out + self.shortcut(x)
-        return elu(out)
+        out = bn_elu(self.bn1, self.conv1(x))
+        return bn_elu(self.bn2, self.conv2(out), residual=self.shortcut(x))
 
 
 class Bottleneck(nn.Module):
@@ -74,11 +71,9 @@ class Bottleneck(nn.Module):
             )
 
     def forward(self, x):
-        out = elu(self.bn1(self.conv1(x)))
-        out = elu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
-        out = out + self.shortcut(x)
-        return elu(out)
+        out = bn_elu(self.bn1, self.conv1(x))
+        out = bn_elu(self.bn2, self.conv2(out))
+        return bn_elu(self.bn3, self.conv3(out), residual=self.shortcut(x))
 
 
 class ResNet(nn.Module):
@@ -102,7 +97,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        out = elu(self.bn1(self.conv1(x)))
+        out = bn_elu(self.bn1, self.conv1(x))
         out = self.layer1(out)
         out = self.layer2(out)
         out = self.layer3(out)

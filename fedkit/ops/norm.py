@@ -48,15 +48,67 @@ class _BnFn(torch.autograd.Function):
         return gx, gw, gb, None, None, None, None, None
 
 
+class _BnActFn(torch.autograd.Function):
+    """BatchNorm with fused residual-add + ELU epilogue (one HBM pass).
+
+    Replaces the reference's `elu(bn(conv(x)) [+ shortcut])` patterns
+    (simple_models.py:150-153) with a single apply kernel.  Backward:
+    g = elu'(y) * gy (from the saved post-activation output), which is both
+    the residual grad and the BN-output grad.
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, running_mean, running_var,
+                training, momentum, eps, residual):
+        y, save_mean, save_invstd = _ext().bn_fwd(
+            x, weight, bias, running_mean, running_var,
+            bool(training), float(momentum), float(eps),
+            residual=residual, elu=True)
+        ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
+        ctx.has_res = residual is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, weight, save_mean, save_invstd, y = ctx.saved_tensors
+        gy = gy.contiguous(memory_format=torch.channels_last)
+        g = _ext().elu_bwd(gy, y)
+        gx, gw, gb = _ext().bn_bwd(g, x, weight, save_mean, save_invstd)
+        gres = g if ctx.has_res else None
+        return gx, gw, gb, None, None, None, None, None, gres
+
+
 class FedBatchNorm2d(nn.BatchNorm2d):
     def forward(self, x):
         if _native(x):
-            self._check_input_dim(x)
-            if self.training and self.track_running_stats \
-                    and self.num_batches_tracked is not None:
-                self.num_batches_tracked.add_(1)
+            self._prep(x)
             x = x.contiguous(memory_format=torch.channels_last)
             return _BnFn.apply(x, self.weight, self.bias,
                                self.running_mean, self.running_var,
                                self.training, self.momentum, self.eps)
         return super().forward(x)
+
+    def _prep(self, x):
+        self._check_input_dim(x)
+        if self.training and self.track_running_stats \
+                and self.num_batches_tracked is not None:
+            self.num_batches_tracked.add_(1)
+
+
+def bn_elu(bn: FedBatchNorm2d, x: torch.Tensor,
+           residual: torch.Tensor = None) -> torch.Tensor:
+    """elu(bn(x) [+ residual]) — fused on GPU, composed on CPU."""
+    if _native(x):
+        bn._prep(x)
+        x = x.contiguous(memory_format=torch.channels_last)
+        if residual is not None:
+            residual = residual.contiguous(memory_format=torch.channels_last)
+            if residual.dtype != x.dtype:
+                residual = residual.to(x.dtype)
+        return _BnActFn.apply(x, bn.weight, bn.bias,
+                              bn.running_mean, bn.running_var,
+                              bn.training, bn.momentum, bn.eps, residual)
+    y = nn.BatchNorm2d.forward(bn, x)
+    if residual is not None:
+        y = y + residual
+    return F.elu(y)

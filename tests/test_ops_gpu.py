@@ -143,6 +143,60 @@ def test_bn_fwd_bwd_vs_torch(C, dtype):
     assert rel_err(gx, xf.grad) < (1e-3 if dtype == torch.float32 else 5e-2)
 
 
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("with_res", [False, True])
+def test_bn_fused_elu_residual(dtype, with_res):
+    """bn_fwd(residual=..., elu=True) == elu(batch_norm(x) + residual)."""
+    ext = _ext()
+    torch.manual_seed(5)
+    C = 128
+    x = (torch.randn(8, C, 9, 9, device="cuda")).to(dtype)
+    x = x.contiguous(memory_format=torch.channels_last)
+    res = torch.randn_like(x) if with_res else None
+    gamma = torch.rand(C, device="cuda") + 0.5
+    beta = torch.randn(C, device="cuda")
+    rm = torch.zeros(C, device="cuda")
+    rv = torch.ones(C, device="cuda")
+    y, _, _ = ext.bn_fwd(x, gamma, beta, rm, rv, True, 0.1, 1e-5,
+                         residual=res, elu=True)
+    yref = F.batch_norm(x.float(), torch.zeros(C, device="cuda"),
+                        torch.ones(C, device="cuda"), gamma, beta,
+                        True, 0.1, 1e-5)
+    if with_res:
+        yref = yref + res.float()
+    yref = F.elu(yref)
+    assert rel_err(y, yref) < (1e-4 if dtype == torch.float32 else 3e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_bn_elu_autograd_vs_composed(dtype):
+    """fedkit.ops.norm.bn_elu fwd+bwd == composed BN/add/ELU autograd."""
+    from fedkit.ops.norm import FedBatchNorm2d, bn_elu
+    torch.manual_seed(6)
+    C = 64
+    bn = FedBatchNorm2d(C).cuda().train()
+    x = torch.randn(8, C, 9, 9, device="cuda", dtype=dtype,
+                    requires_grad=True)
+    res = torch.randn_like(x, requires_grad=True)
+    xcl = x.detach().clone().requires_grad_(True)
+    rescl = res.detach().clone().requires_grad_(True)
+    y = bn_elu(bn, x.contiguous(memory_format=torch.channels_last),
+               residual=res)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+
+    bn2 = FedBatchNorm2d(C).cuda().train()
+    yref = F.elu(F.batch_norm(xcl.float(), bn2.running_mean, bn2.running_var,
+                              bn2.weight, bn2.bias, True, 0.1, 1e-5)
+                 + rescl.float())
+    yref.backward(gy.float())
+    tol = 1e-3 if dtype == torch.float32 else 5e-2
+    assert rel_err(y, yref) < tol
+    assert rel_err(x.grad, xcl.grad) < tol
+    assert rel_err(res.grad, rescl.grad) < tol
+    assert rel_err(bn.weight.grad, bn2.weight.grad) < tol
+
+
 def test_bn_eval_mode():
     ext = _ext()
     C = 64
