@@ -58,7 +58,7 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      int R, int S, int P, int Q, int Kg) {
   constexpr int AB = BM * BK * 2;          // A tile bytes
   constexpr int BB = BN * BK * 2;          // B tile bytes
-  __shared__ char smem[2 * (AB + BB)];
+  __shared__ char smem[3 * (AB + BB)];     // 3-stage pipeline
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -150,15 +150,26 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   const int frag_k = (lane >> 4) * 8;      // 8 contiguous k per lane
 
   const int nkt = Kg / BK;
+  // 3-stage software pipeline: tiles kt and kt+1 are in flight on entry to
+  // iteration kt; kt+2 is issued right after the barrier.  The wait is a
+  // MANUAL vmcnt(LPS) (LPS = this thread's GLDS ops per stage) so only tile
+  // kt's DMA is drained — __syncthreads() would emit vmcnt(0) and kill the
+  // overlap (guide: LDS-DMA is a pending LDS write on the VM counter).  The
+  // single raw s_barrier does double duty: every wave's tile-kt loads have
+  // landed, and every wave is done reading buffer (kt+2)%3 (used by kt-1).
+  constexpr int LPS = A_SLOTS + B_SLOTS;
   stage(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
+  if (nkt > 1) stage(1, 1);
 
-  int cur = 0;
   for (int kt = 0; kt < nkt; ++kt) {
-    if (kt + 1 < nkt) stage(cur ^ 1, kt + 1);
-    const char* A = bufA(cur);
-    const char* B = bufB(cur);
+    if (kt + 1 < nkt)
+      asm volatile("s_waitcnt vmcnt(%0)" ::"n"(LPS) : "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (kt + 2 < nkt) stage((kt + 2) % 3, kt + 2);
+    const char* A = bufA(kt % 3);
+    const char* B = bufB(kt % 3);
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 32) {
       bf16x8 a[MR], b[NR];
@@ -177,9 +188,6 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
           acc[mfrag][nfrag] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[mfrag], b[nfrag], acc[mfrag][nfrag], 0, 0, 0);
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    cur ^= 1;
   }
 
   // ---- epilogue: D fragment lane l holds col = l&15, rows (l>>4)*4 + v
