@@ -130,6 +130,41 @@ __global__ void bn_colsum_kernel(const float* __restrict__ part, int nb,
   if (lane == 0) ws[col] = s;
 }
 
+// fused stage 2 + statistics for the TRAINING forward: one wave per channel
+// sums both partial columns (sum, sumsq) and lane 0 derives mean/invstd and
+// updates the running stats — one launch instead of colsum + finalize.
+__global__ void bn_colsum_finalize_kernel(
+    const float* __restrict__ part, int nb, int C, long long count, float eps,
+    float momentum, bool track, float* __restrict__ running_mean,
+    float* __restrict__ running_var, float* __restrict__ save_mean,
+    float* __restrict__ save_invstd) {
+  int c = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  int lane = threadIdx.x & 63;
+  if (c >= C) return;
+  float s = 0.f, sq = 0.f;
+  for (int b = lane; b < nb; b += 64) {
+    const float* row = part + (long long)b * 2 * C;
+    s += row[c];
+    sq += row[C + c];
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) {
+    s += __shfl_down(s, off, 64);
+    sq += __shfl_down(sq, off, 64);
+  }
+  if (lane == 0) {
+    float mean = s / count;
+    float var = fmaxf(sq / count - mean * mean, 0.f);  // biased
+    if (track) {
+      float unbiased = count > 1 ? var * count / (count - 1) : var;
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    }
+    save_mean[c] = mean;
+    save_invstd[c] = rsqrtf(var + eps);
+  }
+}
+
 // normalize + scale/shift, with optional fused residual add (RES) and ELU
 // epilogue (the reference's `elu(bn(conv) [+ shortcut])` patterns,
 // simple_models.py:150-153) — one pass instead of bn/add/elu separate
@@ -299,17 +334,13 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
                   ") | 256, got C=", C);
       int nb = grid_1d(M * C / VEC, 256, 1024);
       auto part = at::empty({nb, 2, C}, fopts);
-      auto sums = at::empty({2, C}, fopts);
       hipLaunchKernelGGL((bn_partials_kernel<scalar_t, VEC>),
                          dim3(nb), dim3(256), 0,
                          stream, (const scalar_t*)x.data_ptr(), M, C / VEC,
                          part.data_ptr<float>());
-      hipLaunchKernelGGL(bn_colsum_kernel, dim3((2 * C + 3) / 4), dim3(256),
-                         0, stream, part.data_ptr<float>(), nb, 2 * C,
-                         sums.data_ptr<float>());
-      hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
-                         0, stream, sums.data_ptr<float>(), 1, C, M,
-                         (float)eps, (float)momentum, true,
+      hipLaunchKernelGGL(bn_colsum_finalize_kernel, dim3((C + 3) / 4),
+                         dim3(256), 0, stream, part.data_ptr<float>(), nb, C,
+                         M, (float)eps, (float)momentum,
                          running_mean.defined(),
                          running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
                          running_var.defined() ? running_var.data_ptr<float>() : nullptr,
@@ -400,8 +431,6 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                        gamma_f.data_ptr<float>(), ws.data_ptr<float>(), M,
                        nvec, C / VEC, VEC);
   });
-  // gw = sum(dy * xhat), gb = sum(dy)
-  auto gw = ws[1].clone();
-  auto gb = ws[0].clone();
-  return {gx, gw, gb};
+  // gw = sum(dy * xhat), gb = sum(dy): zero-copy views into ws
+  return {gx, ws.select(0, 1), ws.select(0, 0)};
 }
