@@ -207,6 +207,165 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   }
 }
 
+// ------------------------------------------------------- fused bwd-weight
+// dw[kout][rsc] = sum_m dy[m][kout] * im2col(xp)[m][rsc]  (TN GEMM, reduction
+// over M = N*P*Q, im2col implicit).  Replaces the materialized im2col +
+// split-K library GEMM: the col buffer alone was ~RSC/C times the input
+// (151 MB for ResNet layer1) written and re-read through HBM.
+//
+// Both operands need m-contiguous MFMA fragments but are m-major in memory,
+// so each m-tile is staged TRANSPOSED into LDS: every thread gathers 8
+// consecutive-m bf16 with scalar d16 loads (the tile's cache lines are
+// L1/L2-resident after the first touch) and stores them as ONE 16-B LDS
+// write; the compute then reuses the forward kernel's ds_read_b128 + XOR
+// swizzle + mfma structure verbatim.  Output: fp32 partials [SPLITS][K][RSC]
+// (one M-slice per split), column-summed to bf16 by colsum_to_bf16_kernel.
+//
+// P and Q are powers of two for every supported shape (CIFAR 32/16/8/4), so
+// m -> (n, p, q) uses shifts/masks passed from the host.
+
+template <int STRIDE>
+__global__ __launch_bounds__(256)
+void conv_bwd_weight_kernel(const bf16* __restrict__ dy,  // [M][K] (NHWC view)
+                            const bf16* __restrict__ xp,  // [N][Hp][Wp][C]
+                            float* __restrict__ part,     // [SPLITS][K][RSC]
+                            int K, int C, int Hp, int Wp, int S,
+                            long long M, int RSC, int mtiles_per_split,
+                            int qshift /* log2 Q */, int qmask,
+                            int pshift /* log2 (P*Q) */, int pmask) {
+  constexpr int BKM = 64;                  // m-tile (GEMM reduction dim)
+  constexpr int BMK = 64;                  // kout tile rows
+  constexpr int BNR = 64;                  // rsc tile rows
+  __shared__ char smem[(BMK + BNR) * BKM * 2];
+  char* smA = smem;                        // [64 kout][64 m] bf16, XOR swizzle
+  char* smB = smem + BMK * BKM * 2;        // [64 rsc ][64 m]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int krow0 = blockIdx.y * BMK;      // kout tile base
+  const int rrow0 = blockIdx.x * BNR;      // rsc tile base
+  const int split = blockIdx.z;
+
+  // this thread's fixed staging slots: slot d covers LDS row d>>3 (kout or
+  // rsc), swizzled m-chunk (d&7)^(row&7), i.e. 8 consecutive m
+  // (same swizzle the ds_read side expects — identical to the fwd kernel)
+  long long m0 = (long long)split * mtiles_per_split * BKM;
+  long long mend = M;  // tail tiles are masked per element
+
+  // decompose rsc rows once (fixed per thread across the m loop)
+  int a_row[2], a_mc[2];                   // two A slots: d = i*256 + tid
+  int b_row[2], b_mc[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int d = i * 256 + tid;
+    a_row[i] = d >> 3;
+    a_mc[i] = ((d & 7) ^ (a_row[i] & 7)) * 8;
+    b_row[i] = d >> 3;
+    b_mc[i] = ((d & 7) ^ (b_row[i] & 7)) * 8;
+  }
+  int b_c[2], b_r[2], b_s[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int rsc = rrow0 + b_row[i];
+    b_c[i] = rsc % C;
+    int rs = rsc / C;
+    b_s[i] = rs % S;
+    b_r[i] = rs / S;
+  }
+
+  const int frag_row = lane & 15;
+  const int frag_k = (lane >> 4) * 8;
+  f32x4 acc[2][2] = {};                    // wave tile 32(k) x 32(rsc)
+  const int wk = (wave >> 1) * 32;
+  const int wr = (wave & 1) * 32;
+
+  const int ntile = mtiles_per_split;
+  for (int t = 0; t < ntile; ++t) {
+    long long mt = m0 + (long long)t * BKM;
+    if (mt >= mend) break;
+    __syncthreads();                       // previous tile fully consumed
+    // ---- stage A^T: smA[k][m] from dy[m][kout]
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int k = krow0 + a_row[i];
+      const __bf16* dyv = reinterpret_cast<const __bf16*>(dy);
+      bf16x8 v = {};
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        long long m = mt + a_mc[i] + j;
+        if (m < mend) v[j] = dyv[m * K + k];
+      }
+      *(bf16x8*)(smA + lds_off(a_row[i], a_mc[i])) = v;
+    }
+    // ---- stage B^T: smB[rsc][m] from xp (implicit im2col)
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const __bf16* xv = reinterpret_cast<const __bf16*>(xp);
+      bf16x8 v = {};
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        long long m = mt + b_mc[i] + j;
+        if (m < mend) {
+          int q = (int)(m & qmask);
+          int p = (int)((m >> qshift) & pmask);
+          int n = (int)(m >> pshift);
+          v[j] = xv[(((long long)n * Hp + p * STRIDE + b_r[i]) * Wp +
+                     q * STRIDE + b_s[i]) * C + b_c[i]];
+        }
+      }
+      *(bf16x8*)(smB + lds_off(b_row[i], b_mc[i])) = v;
+    }
+    __syncthreads();
+    // ---- mfma on the transposed tiles (same fragment reads as forward)
+#pragma unroll
+    for (int kk = 0; kk < BKM; kk += 32) {
+      bf16x8 a[2], b[2];
+#pragma unroll
+      for (int f = 0; f < 2; ++f)
+        a[f] = *(const bf16x8*)(smA + lds_off(wk + f * 16 + frag_row,
+                                              kk + frag_k));
+#pragma unroll
+      for (int f = 0; f < 2; ++f)
+        b[f] = *(const bf16x8*)(smB + lds_off(wr + f * 16 + frag_row,
+                                              kk + frag_k));
+#pragma unroll
+      for (int fa = 0; fa < 2; ++fa)
+#pragma unroll
+        for (int fb = 0; fb < 2; ++fb)
+          acc[fa][fb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[fa], b[fb], acc[fa][fb], 0, 0, 0);
+    }
+  }
+
+  // ---- epilogue: D[k][rsc] fragments -> fp32 partial slab
+  float* out = part + (long long)split * K * RSC;
+  const int col = rrow0 + wr + frag_row;
+#pragma unroll
+  for (int fa = 0; fa < 2; ++fa) {
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      int k = krow0 + wk + fa * 16 + (lane >> 4) * 4 + v;
+#pragma unroll
+      for (int fb = 0; fb < 2; ++fb)
+        out[(long long)k * RSC + col + fb * 16] = acc[fa][fb][v];
+    }
+  }
+}
+
+// column-sum the [SPLITS][L] fp32 partials into bf16 (thread per column,
+// rows are contiguous so warps read coalesced)
+__global__ void colsum_to_bf16_kernel(const float* __restrict__ part,
+                                      int splits, long long L,
+                                      bf16* __restrict__ out) {
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < L; i += (long long)gridDim.x * blockDim.x) {
+    float s = 0.f;
+    for (int b = 0; b < splits; ++b) s += part[(long long)b * L + i];
+    out[i] = __float2bfloat16(s);
+  }
+}
+
 // ---------------------------------------------------------------- transforms
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16v8;
@@ -441,6 +600,45 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
   int K = gy.size(1), P = gy.size(2), Q = gy.size(3);
   int R = (int)R_in, S = (int)S_in;
   long long M = (long long)N * P * Q;
+  long long RSC = (long long)R * S * C;
+
+  // fused implicit-im2col MFMA path (no col buffer, no library GEMM)
+  bool pow2 = P > 0 && Q > 0 && (P & (P - 1)) == 0 && (Q & (Q - 1)) == 0;
+  if (K % 64 == 0 && RSC % 64 == 0 && pow2 && (stride == 1 || stride == 2)) {
+    int qshift = __builtin_ctz((unsigned)Q);
+    int pshift = qshift + __builtin_ctz((unsigned)P);
+    long long mtiles = (M + 63) / 64;
+    long long tiles_xy = (RSC / 64) * (K / 64);
+    int splits = 1;
+    while (splits < 64 && tiles_xy * splits < 512 &&
+           (long long)splits * 2 <= mtiles)
+      splits *= 2;
+    int mps = (int)((mtiles + splits - 1) / splits);
+    auto fopts = xp.options().dtype(at::kFloat);
+    auto part = at::empty({splits, (long long)K, RSC}, fopts);
+    auto dw = at::empty({K, C, R, S},
+                        xp.options().memory_format(at::MemoryFormat::ChannelsLast));
+    auto stream = fedkit_stream();
+    dim3 grid((unsigned)(RSC / 64), K / 64, splits);
+    if (stride == 1)
+      hipLaunchKernelGGL((conv_bwd_weight_kernel<1>), grid, dim3(256), 0,
+                         stream, (const bf16*)gy.data_ptr(),
+                         (const bf16*)xp.data_ptr(), part.data_ptr<float>(),
+                         K, C, xp.size(2), xp.size(3), S, M, (int)RSC, mps,
+                         qshift, Q - 1, pshift, P - 1);
+    else
+      hipLaunchKernelGGL((conv_bwd_weight_kernel<2>), grid, dim3(256), 0,
+                         stream, (const bf16*)gy.data_ptr(),
+                         (const bf16*)xp.data_ptr(), part.data_ptr<float>(),
+                         K, C, xp.size(2), xp.size(3), S, M, (int)RSC, mps,
+                         qshift, Q - 1, pshift, P - 1);
+    long long L = (long long)K * RSC;
+    hipLaunchKernelGGL(colsum_to_bf16_kernel, dim3(grid_1d(L, 256)), dim3(256),
+                       0, stream, part.data_ptr<float>(), splits, L,
+                       (bf16*)dw.data_ptr());
+    return dw;
+  }
+
   at::Tensor col;
   if (R == 1 && S == 1 && stride == 1) {
     col = xp.permute({0, 2, 3, 1}).reshape({M, C});  // NHWC view, no copy
