@@ -74,7 +74,9 @@ def bench_conv(iters):
             return torch.nn.functional.conv2d(x, w, None, stride, pad)
 
         y1, y2 = fed().float(), eag().float()
-        err = (y1 - y2).abs().max().item()
+        # relative error: with reduction length R*S*C up to 4608 the outputs
+        # are ~N(0, 4608), so bf16 rounding alone gives abs err ~1.0
+        err = ((y1 - y2).abs().max() / y2.abs().max().clamp_min(1e-6)).item()
         t_f = timeit(fed, iters)
         t_e = timeit(eag, iters)
         tot_f += t_f * count

@@ -207,139 +207,190 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   }
 }
 
-// ------------------------------------------------------- fused bwd-weight
-// dw[kout][rsc] = sum_m dy[m][kout] * im2col(xp)[m][rsc]  (TN GEMM, reduction
-// over M = N*P*Q, im2col implicit).  Replaces the materialized im2col +
-// split-K library GEMM: the col buffer alone was ~RSC/C times the input
-// (151 MB for ResNet layer1) written and re-read through HBM.
+// ------------------------------------------------------- bwd-weight (dw)
+// dw[kout][rsc] = sum_m dy[m][kout] * im2col(xp)[m][rsc] — a GEMM whose
+// reduction dim (m = N*P*Q) is the MAJOR axis of both operands, so MFMA
+// fragments (8 reduction-contiguous bf16 per lane) cannot be loaded from the
+// natural layouts.  Instead of gathering scalars (latency-bound) or
+// materializing im2col + a library split-K GEMM (extra HBM round-trips),
+// both operands are TRANSPOSED ONCE with an LDS-tiled transpose:
 //
-// Both operands need m-contiguous MFMA fragments but are m-major in memory,
-// so each m-tile is staged TRANSPOSED into LDS: every thread gathers 8
-// consecutive-m bf16 with scalar d16 loads (the tile's cache lines are
-// L1/L2-resident after the first touch) and stores them as ONE 16-B LDS
-// write; the compute then reuses the forward kernel's ds_read_b128 + XOR
-// swizzle + mfma structure verbatim.  Output: fp32 partials [SPLITS][K][RSC]
-// (one M-slice per split), column-summed to bf16 by colsum_to_bf16_kernel.
+//   dyT [K][M]          <- dy  [M][K]      (transpose_mk_kernel)
+//   xpT [C][N][Hp][Wp]  <- xp  [N][Hp][Wp][C]   (same kernel, M' = N*Hp*Wp)
 //
-// P and Q are powers of two for every supported shape (CIFAR 32/16/8/4), so
-// m -> (n, p, q) uses shifts/masks passed from the host.
+// and the GEMM reads its B operand IMPLICITLY from xpT: for a fixed
+// (r, s, c) row, 8 consecutive m are 8 consecutive q, which in xpT is 16
+// contiguous bytes (stride 1, Q % 8 == 0) — so both tiles stage with the
+// same glds 16 B/lane + XOR-swizzle + 3-stage pipeline as the forward
+// kernel.  Output: fp32 partials [SPLITS][K][RSC] (one m-slice per split),
+// column-summed to bf16 by colsum_to_bf16_kernel.
+//
+// Qualifying shapes (host-checked): stride 1, Q % 8 == 0, P/Q powers of two,
+// K % 64 == 0, RSC % 64 == 0, M % 64 == 0 — i.e. 13 of the 17 ResNet18
+// convs, carrying ~80 % of dw cost.  Everything else takes the im2col +
+// split-K library-GEMM path below.
 
-template <int STRIDE>
+// LDS-tiled 64x64 bf16 transpose: out[k][m] = in[m][k].  Both global phases
+// are 16 B/lane coalesced; the store-phase LDS gather reads 8 scalars whose
+// rows step by 8, so the swizzle folds in (row>>3)&7 as well as row&7 —
+// plain chunk^(row&7) would leave all 64 lanes on one 4-bank group.
+__device__ __forceinline__ int tr_off(int row, int k) {
+  int chunk = (k >> 3) ^ (row & 7) ^ ((row >> 3) & 7);
+  return row * 128 + chunk * 16 + (k & 7) * 2;
+}
+
 __global__ __launch_bounds__(256)
-void conv_bwd_weight_kernel(const bf16* __restrict__ dy,  // [M][K] (NHWC view)
-                            const bf16* __restrict__ xp,  // [N][Hp][Wp][C]
-                            float* __restrict__ part,     // [SPLITS][K][RSC]
-                            int K, int C, int Hp, int Wp, int S,
-                            long long M, int RSC, int mtiles_per_split,
-                            int qshift /* log2 Q */, int qmask,
-                            int pshift /* log2 (P*Q) */, int pmask) {
-  constexpr int BKM = 64;                  // m-tile (GEMM reduction dim)
-  constexpr int BMK = 64;                  // kout tile rows
-  constexpr int BNR = 64;                  // rsc tile rows
-  __shared__ char smem[(BMK + BNR) * BKM * 2];
-  char* smA = smem;                        // [64 kout][64 m] bf16, XOR swizzle
-  char* smB = smem + BMK * BKM * 2;        // [64 rsc ][64 m]
+void transpose_mk_kernel(const bf16* __restrict__ in,  // [M][K]
+                         bf16* __restrict__ out,       // [K][M]
+                         long long M, int K) {
+  __shared__ char smem[64 * 128];
+  const int tid = threadIdx.x;
+  const long long mt = (long long)blockIdx.x * 64;
+  const int kt = blockIdx.y * 64;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {            // load: row = m, 8 k per lane
+    int d = i * 256 + tid;
+    int row = d >> 3, c = d & 7;
+    *(bf16x8*)(smem + tr_off(row, c * 8)) =
+        *(const bf16x8*)(in + (mt + row) * K + kt + c * 8);
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {            // store: row = k, 8 m per lane
+    int d = i * 256 + tid;
+    int kr = d >> 3, c = d & 7;
+    bf16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      v[j] = *(const bf16*)(smem + tr_off(c * 8 + j, kr));
+    *(bf16x8*)(out + (long long)(kt + kr) * M + mt + c * 8) = v;
+  }
+}
+
+// dw GEMM on the transposed operands: D[kout][rsc] += A[kout][m] B[rsc][m]
+// with A = dyT (plain 2-D) and B = implicit transposed im2col of xpT.
+// Same 256-thread / 4-wave / 64x64x64-tile / 3-stage-glds structure as
+// conv_fwd_kernel; grid.z splits the m range into independent fp32 partials.
+__global__ __launch_bounds__(256)
+void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
+                    const bf16* __restrict__ xpT,  // [C][N][Hp][Wp]
+                    float* __restrict__ part,      // [SPLITS][K][RSC]
+                    int K, int C, int N, int Hp, int Wp, int S,
+                    long long M, int RSC, int mtiles_per_split,
+                    int qshift /* log2 Q */, int qmask,
+                    int pshift /* log2 (P*Q) */, int pmask) {
+  constexpr int AB = 64 * 64 * 2;
+  __shared__ char smem[3 * 2 * AB];        // 3-stage A+B double tiles
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
-  const int krow0 = blockIdx.y * BMK;      // kout tile base
-  const int rrow0 = blockIdx.x * BNR;      // rsc tile base
-  const int split = blockIdx.z;
+  const int rrow0 = blockIdx.x * BN;       // rsc tile base
+  const int krow0 = blockIdx.y * 64;       // kout tile base
+  const long long m0 =
+      (long long)blockIdx.z * mtiles_per_split * BK;
+  const int nkt = (int)std::min<long long>(mtiles_per_split,
+                                           (M - m0 + BK - 1) / BK);
 
-  // this thread's fixed staging slots: slot d covers LDS row d>>3 (kout or
-  // rsc), swizzled m-chunk (d&7)^(row&7), i.e. 8 consecutive m
-  // (same swizzle the ds_read side expects — identical to the fwd kernel)
-  long long m0 = (long long)split * mtiles_per_split * BKM;
-  long long mend = M;  // tail tiles are masked per element
-
-  // decompose rsc rows once (fixed per thread across the m loop)
-  int a_row[2], a_mc[2];                   // two A slots: d = i*256 + tid
-  int b_row[2], b_mc[2];
+  // fixed staging slots (2 A + 2 B per thread), fwd-kernel scheme: slot
+  // d = i*256 + tid -> LDS row d>>3, swizzled m-chunk (d&7)^(row&7)
+  int s_row[2], s_k8[2];
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
     int d = i * 256 + tid;
-    a_row[i] = d >> 3;
-    a_mc[i] = ((d & 7) ^ (a_row[i] & 7)) * 8;
-    b_row[i] = d >> 3;
-    b_mc[i] = ((d & 7) ^ (b_row[i] & 7)) * 8;
+    s_row[i] = d >> 3;
+    s_k8[i] = (d & 7) ^ (s_row[i] & 7);
   }
-  int b_c[2], b_r[2], b_s[2];
+  long long a_base[2];                     // dyT row offset
+  long long b_base[2];                     // xpT (c, r-row, s-col) offset
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
-    int rsc = rrow0 + b_row[i];
-    b_c[i] = rsc % C;
+    a_base[i] = (long long)(krow0 + s_row[i]) * M;
+    int rsc = rrow0 + s_row[i];
+    int c = rsc % C;
     int rs = rsc / C;
-    b_s[i] = rs % S;
-    b_r[i] = rs / S;
+    int s = rs % S;
+    int r = rs / S;
+    b_base[i] = ((long long)c * N * Hp + r) * Wp + s;  // + n*Hp*Wp + p*Wp + q
   }
+
+  auto bufA = [&](int b) -> char* { return smem + b * 2 * AB; };
+  auto bufB = [&](int b) -> char* { return smem + b * 2 * AB + AB; };
+
+  // A (dyT rows, always 16-B aligned) stages via glds; B (xpT rows shifted by
+  // the filter column s, so only 2-B aligned) REGISTER-stages: an
+  // unaligned-capable global vector load into VGPRs, committed to LDS with
+  // ds_write_b128 right before the barrier (guide T14 — ties glds within a
+  // few %, and sidesteps the LDS-DMA alignment question entirely).
+  bf16x8 breg[2][2];
+  auto stage = [&](int buf, int kt, bf16x8* br) {
+    const long long mt = m0 + (long long)kt * BK;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const bf16* src = dyT + a_base[i] + mt + s_k8[i] * 8;
+      GLDS16(src, bufA(buf) + (i * 4 + wave) * 1024);
+    }
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      long long mm = mt + s_k8[i] * 8;     // 8 consecutive m = 8 consecutive q
+      int q = (int)(mm & qmask);
+      int p = (int)((mm >> qshift) & pmask);
+      int n = (int)(mm >> pshift);
+      __builtin_memcpy(&br[i], xpT + b_base[i] + ((long long)n * Hp + p) * Wp + q,
+                       sizeof(bf16x8));
+    }
+  };
+  auto commitB = [&](int buf, const bf16x8* br) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+      *(bf16x8*)(bufB(buf) + (i * 4 + wave) * 1024 + lane * 16) = br[i];
+  };
 
   const int frag_row = lane & 15;
   const int frag_k = (lane >> 4) * 8;
-  f32x4 acc[2][2] = {};                    // wave tile 32(k) x 32(rsc)
+  f32x4 acc[2][2] = {};                    // wave tile 32(kout) x 32(rsc)
   const int wk = (wave >> 1) * 32;
   const int wr = (wave & 1) * 32;
 
-  const int ntile = mtiles_per_split;
-  for (int t = 0; t < ntile; ++t) {
-    long long mt = m0 + (long long)t * BKM;
-    if (mt >= mend) break;
-    __syncthreads();                       // previous tile fully consumed
-    // ---- stage A^T: smA[k][m] from dy[m][kout]
+  if (nkt > 0) {
+    stage(0, 0, breg[0]);
+    if (nkt > 1) stage(1, 1, breg[1]);
+    for (int kt = 0; kt < nkt; ++kt) {
+      // drain stage kt's 4 VMEM ops (2 glds + 2 B reg loads); stage kt+1's
+      // stay in flight
+      if (kt + 1 < nkt)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(4) : "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      commitB(kt % 3, breg[kt % 2]);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      if (kt + 2 < nkt) stage((kt + 2) % 3, kt + 2, breg[kt % 2]);
+      const char* A = bufA(kt % 3);
+      const char* B = bufB(kt % 3);
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      int k = krow0 + a_row[i];
-      const __bf16* dyv = reinterpret_cast<const __bf16*>(dy);
-      bf16x8 v = {};
+      for (int kk = 0; kk < BK; kk += 32) {
+        bf16x8 a[2], b[2];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        long long m = mt + a_mc[i] + j;
-        if (m < mend) v[j] = dyv[m * K + k];
-      }
-      *(bf16x8*)(smA + lds_off(a_row[i], a_mc[i])) = v;
-    }
-    // ---- stage B^T: smB[rsc][m] from xp (implicit im2col)
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      const __bf16* xv = reinterpret_cast<const __bf16*>(xp);
-      bf16x8 v = {};
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        long long m = mt + b_mc[i] + j;
-        if (m < mend) {
-          int q = (int)(m & qmask);
-          int p = (int)((m >> qshift) & pmask);
-          int n = (int)(m >> pshift);
-          v[j] = xv[(((long long)n * Hp + p * STRIDE + b_r[i]) * Wp +
-                     q * STRIDE + b_s[i]) * C + b_c[i]];
-        }
-      }
-      *(bf16x8*)(smB + lds_off(b_row[i], b_mc[i])) = v;
-    }
-    __syncthreads();
-    // ---- mfma on the transposed tiles (same fragment reads as forward)
-#pragma unroll
-    for (int kk = 0; kk < BKM; kk += 32) {
-      bf16x8 a[2], b[2];
-#pragma unroll
-      for (int f = 0; f < 2; ++f)
-        a[f] = *(const bf16x8*)(smA + lds_off(wk + f * 16 + frag_row,
+        for (int f = 0; f < 2; ++f)
+          a[f] = *(const bf16x8*)(A + lds_off(wk + f * 16 + frag_row,
                                               kk + frag_k));
 #pragma unroll
-      for (int f = 0; f < 2; ++f)
-        b[f] = *(const bf16x8*)(smB + lds_off(wr + f * 16 + frag_row,
+        for (int f = 0; f < 2; ++f)
+          b[f] = *(const bf16x8*)(B + lds_off(wr + f * 16 + frag_row,
                                               kk + frag_k));
 #pragma unroll
-      for (int fa = 0; fa < 2; ++fa)
+        for (int fa = 0; fa < 2; ++fa)
 #pragma unroll
-        for (int fb = 0; fb < 2; ++fb)
-          acc[fa][fb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[fa], b[fb], acc[fa][fb], 0, 0, 0);
+          for (int fb = 0; fb < 2; ++fb)
+            acc[fa][fb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a[fa], b[fb], acc[fa][fb], 0, 0, 0);
+      }
     }
   }
 
-  // ---- epilogue: D[k][rsc] fragments -> fp32 partial slab
-  float* out = part + (long long)split * K * RSC;
+  // ---- epilogue: D[kout][rsc] fragments -> this split's fp32 partial slab
+  float* out = part + (long long)blockIdx.z * K * RSC;
   const int col = rrow0 + wr + frag_row;
 #pragma unroll
   for (int fa = 0; fa < 2; ++fa) {
@@ -602,36 +653,44 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
   long long M = (long long)N * P * Q;
   long long RSC = (long long)R * S * C;
 
-  // fused implicit-im2col MFMA path (no col buffer, no library GEMM)
+  // transpose-then-implicit-GEMM MFMA path (no col buffer, no library GEMM):
+  // see the dw block comment above conv kernels.  stride-1 only (stride-2 m
+  // chunks are not q-contiguous in xpT), Q >= 8 so an m-chunk stays within
+  // one q row — covers 13/17 ResNet18 convs, ~80 % of dw time.
+  int Hp = xp.size(2), Wp = xp.size(3);
+  long long NHW = (long long)N * Hp * Wp;
   bool pow2 = P > 0 && Q > 0 && (P & (P - 1)) == 0 && (Q & (Q - 1)) == 0;
-  if (K % 64 == 0 && RSC % 64 == 0 && pow2 && (stride == 1 || stride == 2)) {
+  if (stride == 1 && pow2 && Q % 8 == 0 && K % 64 == 0 && RSC % 64 == 0 &&
+      C % 64 == 0 && M % 64 == 0 && NHW % 64 == 0) {
+    auto stream = fedkit_stream();
+    auto dyT = at::empty({(long long)K, M}, gy.options());
+    auto xpT = at::empty({(long long)C, NHW}, xp.options());
+    hipLaunchKernelGGL(transpose_mk_kernel, dim3((unsigned)(M / 64), K / 64),
+                       dim3(256), 0, stream, (const bf16*)gy.data_ptr(),
+                       (bf16*)dyT.data_ptr(), M, K);
+    hipLaunchKernelGGL(transpose_mk_kernel,
+                       dim3((unsigned)(NHW / 64), C / 64), dim3(256), 0,
+                       stream, (const bf16*)xp.data_ptr(),
+                       (bf16*)xpT.data_ptr(), NHW, C);
     int qshift = __builtin_ctz((unsigned)Q);
     int pshift = qshift + __builtin_ctz((unsigned)P);
-    long long mtiles = (M + 63) / 64;
+    long long mtiles = M / 64;
     long long tiles_xy = (RSC / 64) * (K / 64);
     int splits = 1;
     while (splits < 64 && tiles_xy * splits < 512 &&
            (long long)splits * 2 <= mtiles)
       splits *= 2;
     int mps = (int)((mtiles + splits - 1) / splits);
-    auto fopts = xp.options().dtype(at::kFloat);
-    auto part = at::empty({splits, (long long)K, RSC}, fopts);
+    auto part = at::empty({splits, (long long)K, RSC},
+                          xp.options().dtype(at::kFloat));
     auto dw = at::empty({K, C, R, S},
                         xp.options().memory_format(at::MemoryFormat::ChannelsLast));
-    auto stream = fedkit_stream();
     dim3 grid((unsigned)(RSC / 64), K / 64, splits);
-    if (stride == 1)
-      hipLaunchKernelGGL((conv_bwd_weight_kernel<1>), grid, dim3(256), 0,
-                         stream, (const bf16*)gy.data_ptr(),
-                         (const bf16*)xp.data_ptr(), part.data_ptr<float>(),
-                         K, C, xp.size(2), xp.size(3), S, M, (int)RSC, mps,
-                         qshift, Q - 1, pshift, P - 1);
-    else
-      hipLaunchKernelGGL((conv_bwd_weight_kernel<2>), grid, dim3(256), 0,
-                         stream, (const bf16*)gy.data_ptr(),
-                         (const bf16*)xp.data_ptr(), part.data_ptr<float>(),
-                         K, C, xp.size(2), xp.size(3), S, M, (int)RSC, mps,
-                         qshift, Q - 1, pshift, P - 1);
+    hipLaunchKernelGGL(dw_gemm_kernel, grid, dim3(256), 0, stream,
+                       (const bf16*)dyT.data_ptr(),
+                       (const bf16*)xpT.data_ptr(), part.data_ptr<float>(),
+                       K, C, N, Hp, Wp, S, M, (int)RSC, mps,
+                       qshift, Q - 1, pshift, P - 1);
     long long L = (long long)K * RSC;
     hipLaunchKernelGGL(colsum_to_bf16_kernel, dim3(grid_1d(L, 256)), dim3(256),
                        0, stream, part.data_ptr<float>(), splits, L,
