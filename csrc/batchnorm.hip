@@ -27,8 +27,8 @@ struct alignas(sizeof(T) * VEC) VecT { T v[VEC]; };
 // finalize / bwd-reduce kernels) sums the <=1024 partial rows.
 template <typename T, int VEC>
 __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
-                                   int Cv /* C/VEC */,
-                                   float* __restrict__ part /* [nb][2][C] */) {
+                                   int Cv /* C/VEC */, int nb,
+                                   float* __restrict__ part /* [2][C][nb] */) {
   using V = VecT<T, VEC>;
   const int C = Cv * VEC;
   const V* xv = reinterpret_cast<const V*>(x);
@@ -62,7 +62,8 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
   }
   const int members = 256 / Cv;
   const int g = tid % Cv;
-  float* out = part + (long long)blockIdx.x * 2 * C;
+  // partials land TRANSPOSED ([2][C][nb], nb innermost) so the stage-2
+  // column sums read each channel's rows as one contiguous run
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     float* src = pass == 0 ? s : sq;
@@ -76,7 +77,8 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
 #pragma unroll
         for (int j = 0; j < VEC; ++j) acc[j] += red[(g + k * Cv) * VEC + j];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j) out[pass * C + c0 + j] = acc[j];
+      for (int j = 0; j < VEC; ++j)
+        part[(long long)(pass * C + c0 + j) * nb + blockIdx.x] = acc[j];
     }
   }
 }
@@ -95,8 +97,8 @@ __global__ void bn_finalize_kernel(const float* __restrict__ part, int nb,
   if (training) {
     float s = 0.f, sq = 0.f;
     for (int b = 0; b < nb; ++b) {
-      s += part[(long long)b * 2 * C + c];
-      sq += part[(long long)b * 2 * C + C + c];
+      s += part[(long long)c * nb + b];
+      sq += part[(long long)(C + c) * nb + b];
     }
     mean = s / count;
     var = fmaxf(sq / count - mean * mean, 0.f);  // biased
@@ -124,7 +126,7 @@ __global__ void bn_colsum_kernel(const float* __restrict__ part, int nb,
   int lane = threadIdx.x & 63;
   if (col >= C2) return;
   float s = 0.f;
-  for (int b = lane; b < nb; b += 64) s += part[(long long)b * C2 + col];
+  for (int b = lane; b < nb; b += 64) s += part[(long long)col * nb + b];
 #pragma unroll
   for (int off = 32; off; off >>= 1) s += __shfl_down(s, off, 64);
   if (lane == 0) ws[col] = s;
@@ -142,10 +144,11 @@ __global__ void bn_colsum_finalize_kernel(
   int lane = threadIdx.x & 63;
   if (c >= C) return;
   float s = 0.f, sq = 0.f;
+  const float* col_s = part + (long long)c * nb;
+  const float* col_sq = part + (long long)(C + c) * nb;
   for (int b = lane; b < nb; b += 64) {
-    const float* row = part + (long long)b * 2 * C;
-    s += row[c];
-    sq += row[C + c];
+    s += col_s[b];
+    sq += col_sq[b];
   }
 #pragma unroll
   for (int off = 32; off; off >>= 1) {
@@ -202,9 +205,10 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
 template <typename T, int VEC>
 __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
                                        const T* __restrict__ gy, long long M,
-                                       int Cv, const float* __restrict__ mean,
+                                       int Cv, int nb,
+                                       const float* __restrict__ mean,
                                        const float* __restrict__ invstd,
-                                       float* __restrict__ part /* [nb][2][C] */) {
+                                       float* __restrict__ part /* [2][C][nb] */) {
   using V = VecT<T, VEC>;
   const int C = Cv * VEC;
   const V* xv = reinterpret_cast<const V*>(x);
@@ -244,7 +248,6 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
   }
   const int members = 256 / Cv;
   const int g8 = tid % Cv;
-  float* out = part + (long long)blockIdx.x * 2 * C;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     float* src = pass == 0 ? sdy : sdyx;
@@ -258,7 +261,8 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
 #pragma unroll
         for (int j = 0; j < VEC; ++j) acc[j] += red[(g8 + k * Cv) * VEC + j];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j) out[pass * C + c0 + j] = acc[j];
+      for (int j = 0; j < VEC; ++j)
+        part[(long long)(pass * C + c0 + j) * nb + blockIdx.x] = acc[j];
     }
   }
 }
@@ -333,11 +337,11 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
                   "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                   ") | 256, got C=", C);
       int nb = grid_1d(M * C / VEC, 256, 1024);
-      auto part = at::empty({nb, 2, C}, fopts);
+      auto part = at::empty({2 * C, nb}, fopts);
       hipLaunchKernelGGL((bn_partials_kernel<scalar_t, VEC>),
                          dim3(nb), dim3(256), 0,
                          stream, (const scalar_t*)x.data_ptr(), M, C / VEC,
-                         part.data_ptr<float>());
+                         nb, part.data_ptr<float>());
       hipLaunchKernelGGL(bn_colsum_finalize_kernel, dim3((C + 3) / 4),
                          dim3(256), 0, stream, part.data_ptr<float>(), nb, C,
                          M, (float)eps, (float)momentum,
@@ -408,11 +412,11 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                 "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                 ") | 256, got C=", C);
     int nb = grid_1d(M * C / VEC, 256, 1024);
-    auto part = at::empty({nb, 2, C}, fopts);
+    auto part = at::empty({2 * C, nb}, fopts);
     hipLaunchKernelGGL((bn_bwd_partials_kernel<scalar_t, VEC>),
                        dim3(nb), dim3(256), 0,
                        stream, (const scalar_t*)x.data_ptr(),
-                       (const scalar_t*)gy.data_ptr(), M, C / VEC,
+                       (const scalar_t*)gy.data_ptr(), M, C / VEC, nb,
                        save_mean.data_ptr<float>(),
                        save_invstd.data_ptr<float>(), part.data_ptr<float>());
     hipLaunchKernelGGL(bn_colsum_kernel, dim3((2 * C + 3) / 4), dim3(256),

@@ -267,10 +267,53 @@ void transpose_mk_kernel(const bf16* __restrict__ in,  // [M][K]
   }
 }
 
+// Vectorized bwd-data weight rotation: wrot[c][R-1-r][S-1-s][k] =
+// w[k][r][s][c] — i.e. the 64x64 LDS transpose above with the OUTPUT row
+// index permuted ((r,s,c) -> (c, R-1-r, S-1-s)).  The scalar rot_weight
+// fallback below measured 9 GB/s (2-B scattered stores, 144 wavefronts);
+// this runs at transpose speed.  Host-gated on K % 64 == 0, R*S*C % 64 == 0.
+__global__ __launch_bounds__(256)
+void rot_weight64_kernel(const bf16* __restrict__ w,   // [K][R*S*C]
+                         bf16* __restrict__ wr,        // [R*S*C][K] permuted
+                         int K, int R, int S, int C) {
+  __shared__ char smem[64 * 128];
+  const int tid = threadIdx.x;
+  const int kt = blockIdx.x * 64;          // k tile (input rows)
+  const int jt = blockIdx.y * 64;          // rsc tile (input cols)
+  const int RSC = R * S * C;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {            // load: row = k, 8 rsc per lane
+    int d = i * 256 + tid;
+    int row = d >> 3, c = d & 7;
+    *(bf16x8*)(smem + tr_off(row, c * 8)) =
+        *(const bf16x8*)(w + (long long)(kt + row) * RSC + jt + c * 8);
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {            // store: row = rsc (permuted), 8 k
+    int d = i * 256 + tid;
+    int jr = d >> 3, c = d & 7;
+    bf16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      v[j] = *(const bf16*)(smem + tr_off(c * 8 + j, jr));
+    int rsc = jt + jr;
+    int cc = rsc % C;
+    int rs = rsc / C;
+    int s = rs % S;
+    int r = rs / S;
+    long long o = ((long long)cc * R + (R - 1 - r)) * S + (S - 1 - s);
+    *(bf16x8*)(wr + o * K + kt + c * 8) = v;
+  }
+}
+
 // dw GEMM on the transposed operands: D[kout][rsc] += A[kout][m] B[rsc][m]
 // with A = dyT (plain 2-D) and B = implicit transposed im2col of xpT.
-// Same 256-thread / 4-wave / 64x64x64-tile / 3-stage-glds structure as
-// conv_fwd_kernel; grid.z splits the m range into independent fp32 partials.
+// Same 256-thread / 4-wave / 3-stage-glds structure as conv_fwd_kernel;
+// BMK (kout tile) = 128 where K allows — the 64x64 tile has half the MFMA
+// work per LDS read and measured ~1.7x slower — else 64; grid.z splits the
+// m range into independent fp32 partials.
+template <int BMK>
 __global__ __launch_bounds__(256)
 void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
                     const bf16* __restrict__ xpT,  // [C][N][Hp][Wp]
@@ -279,34 +322,40 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
                     long long M, int RSC, int mtiles_per_split,
                     int qshift /* log2 Q */, int qmask,
                     int pshift /* log2 (P*Q) */, int pmask) {
-  constexpr int AB = 64 * 64 * 2;
-  __shared__ char smem[3 * 2 * AB];        // 3-stage A+B double tiles
+  constexpr int AB = BMK * BK * 2;         // A tile bytes
+  constexpr int BB = BN * BK * 2;          // B tile bytes
+  __shared__ char smem[3 * (AB + BB)];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
   const int rrow0 = blockIdx.x * BN;       // rsc tile base
-  const int krow0 = blockIdx.y * 64;       // kout tile base
+  const int krow0 = blockIdx.y * BMK;      // kout tile base
   const long long m0 =
       (long long)blockIdx.z * mtiles_per_split * BK;
   const int nkt = (int)std::min<long long>(mtiles_per_split,
                                            (M - m0 + BK - 1) / BK);
 
-  // fixed staging slots (2 A + 2 B per thread), fwd-kernel scheme: slot
-  // d = i*256 + tid -> LDS row d>>3, swizzled m-chunk (d&7)^(row&7)
-  int s_row[2], s_k8[2];
+  // fixed staging slots (A_SLOTS A + 2 B per thread), fwd-kernel scheme:
+  // slot d = i*256 + tid -> LDS row d>>3, swizzled m-chunk (d&7)^(row&7)
+  constexpr int A_SLOTS = (BMK * BK) / (256 * 8);  // 2 or 4
+  long long a_base[A_SLOTS];
+  int a_k8[A_SLOTS];
+#pragma unroll
+  for (int i = 0; i < A_SLOTS; ++i) {
+    int d = i * 256 + tid;
+    int row = d >> 3;
+    a_k8[i] = (d & 7) ^ (row & 7);
+    a_base[i] = (long long)(krow0 + row) * M;
+  }
+  long long b_base[2];                     // xpT (c, r-row, s-col) offset
+  int b_k8[2];
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
     int d = i * 256 + tid;
-    s_row[i] = d >> 3;
-    s_k8[i] = (d & 7) ^ (s_row[i] & 7);
-  }
-  long long a_base[2];                     // dyT row offset
-  long long b_base[2];                     // xpT (c, r-row, s-col) offset
-#pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    a_base[i] = (long long)(krow0 + s_row[i]) * M;
-    int rsc = rrow0 + s_row[i];
+    int row = d >> 3;
+    b_k8[i] = (d & 7) ^ (row & 7);
+    int rsc = rrow0 + row;
     int c = rsc % C;
     int rs = rsc / C;
     int s = rs % S;
@@ -314,8 +363,8 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
     b_base[i] = ((long long)c * N * Hp + r) * Wp + s;  // + n*Hp*Wp + p*Wp + q
   }
 
-  auto bufA = [&](int b) -> char* { return smem + b * 2 * AB; };
-  auto bufB = [&](int b) -> char* { return smem + b * 2 * AB + AB; };
+  auto bufA = [&](int b) -> char* { return smem + b * (AB + BB); };
+  auto bufB = [&](int b) -> char* { return smem + b * (AB + BB) + AB; };
 
   // A (dyT rows, always 16-B aligned) stages via glds; B (xpT rows shifted by
   // the filter column s, so only 2-B aligned) REGISTER-stages: an
@@ -326,13 +375,13 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
   auto stage = [&](int buf, int kt, bf16x8* br) {
     const long long mt = m0 + (long long)kt * BK;
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      const bf16* src = dyT + a_base[i] + mt + s_k8[i] * 8;
+    for (int i = 0; i < A_SLOTS; ++i) {
+      const bf16* src = dyT + a_base[i] + mt + a_k8[i] * 8;
       GLDS16(src, bufA(buf) + (i * 4 + wave) * 1024);
     }
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      long long mm = mt + s_k8[i] * 8;     // 8 consecutive m = 8 consecutive q
+      long long mm = mt + b_k8[i] * 8;     // 8 consecutive m = 8 consecutive q
       int q = (int)(mm & qmask);
       int p = (int)((mm >> qshift) & pmask);
       int n = (int)(mm >> pshift);
@@ -346,20 +395,21 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
       *(bf16x8*)(bufB(buf) + (i * 4 + wave) * 1024 + lane * 16) = br[i];
   };
 
+  constexpr int MR = BMK / 2 / 16;         // kout frags per wave (2 or 4)
   const int frag_row = lane & 15;
   const int frag_k = (lane >> 4) * 8;
-  f32x4 acc[2][2] = {};                    // wave tile 32(kout) x 32(rsc)
-  const int wk = (wave >> 1) * 32;
+  f32x4 acc[MR][2] = {};                   // wave tile (BMK/2) x 32(rsc)
+  const int wk = (wave >> 1) * (BMK / 2);
   const int wr = (wave & 1) * 32;
 
   if (nkt > 0) {
     stage(0, 0, breg[0]);
     if (nkt > 1) stage(1, 1, breg[1]);
     for (int kt = 0; kt < nkt; ++kt) {
-      // drain stage kt's 4 VMEM ops (2 glds + 2 B reg loads); stage kt+1's
-      // stay in flight
+      // drain stage kt's VMEM ops (A_SLOTS glds + 2 B reg loads); stage
+      // kt+1's stay in flight
       if (kt + 1 < nkt)
-        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(4) : "memory");
+        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(A_SLOTS + 2) : "memory");
       else
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       commitB(kt % 3, breg[kt % 2]);
@@ -370,9 +420,9 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
       const char* B = bufB(kt % 3);
 #pragma unroll
       for (int kk = 0; kk < BK; kk += 32) {
-        bf16x8 a[2], b[2];
+        bf16x8 a[MR], b[2];
 #pragma unroll
-        for (int f = 0; f < 2; ++f)
+        for (int f = 0; f < MR; ++f)
           a[f] = *(const bf16x8*)(A + lds_off(wk + f * 16 + frag_row,
                                               kk + frag_k));
 #pragma unroll
@@ -380,7 +430,7 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
           b[f] = *(const bf16x8*)(B + lds_off(wr + f * 16 + frag_row,
                                               kk + frag_k));
 #pragma unroll
-        for (int fa = 0; fa < 2; ++fa)
+        for (int fa = 0; fa < MR; ++fa)
 #pragma unroll
           for (int fb = 0; fb < 2; ++fb)
             acc[fa][fb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -393,7 +443,7 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
   float* out = part + (long long)blockIdx.z * K * RSC;
   const int col = rrow0 + wr + frag_row;
 #pragma unroll
-  for (int fa = 0; fa < 2; ++fa) {
+  for (int fa = 0; fa < MR; ++fa) {
 #pragma unroll
     for (int v = 0; v < 4; ++v) {
       int k = krow0 + wk + fa * 16 + (lane >> 4) * 4 + v;
@@ -414,6 +464,23 @@ __global__ void colsum_to_bf16_kernel(const float* __restrict__ part,
     float s = 0.f;
     for (int b = 0; b < splits; ++b) s += part[(long long)b * L + i];
     out[i] = __float2bfloat16(s);
+  }
+}
+
+// first-stage column reduction [SPLITS][L] -> [G][L]: with L as small as
+// 36 K (layer1 dw) a single-stage colsum has too few threads to cover the
+// read latency (measured 0.57 TB/s); grid.y = G groups raise parallelism
+// G-fold and the [G][L] slab is finished by colsum_to_bf16_kernel.
+__global__ void colsum_stage_kernel(const float* __restrict__ part,
+                                    int rows_per_group, long long L,
+                                    float* __restrict__ out) {
+  const float* src = part + (long long)blockIdx.y * rows_per_group * L;
+  float* dst = out + (long long)blockIdx.y * L;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < L; i += (long long)gridDim.x * blockDim.x) {
+    float s = 0.f;
+    for (int b = 0; b < rows_per_group; ++b) s += src[(long long)b * L + i];
+    dst[i] = s;
   }
 }
 
@@ -630,9 +697,15 @@ at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
   {
     auto stream = fedkit_stream();
     long long total = (long long)K * R * S * C;
-    hipLaunchKernelGGL((rot_weight_kernel<bf16>), dim3(grid_1d(total, 256)),
-                       dim3(256), 0, stream, (const bf16*)w.data_ptr(),
-                       (bf16*)wrot.data_ptr(), K, R, S, C);
+    int RSC = R * S * C;
+    if (K % 64 == 0 && RSC % 64 == 0)
+      hipLaunchKernelGGL(rot_weight64_kernel, dim3(K / 64, RSC / 64),
+                         dim3(256), 0, stream, (const bf16*)w.data_ptr(),
+                         (bf16*)wrot.data_ptr(), K, R, S, C);
+    else
+      hipLaunchKernelGGL((rot_weight_kernel<bf16>), dim3(grid_1d(total, 256)),
+                         dim3(256), 0, stream, (const bf16*)w.data_ptr(),
+                         (bf16*)wrot.data_ptr(), K, R, S, C);
   }
   // dilate+pad gy: pl = R-1-pad, pr = pl + a with a = (H + 2p - R) % stride
   int pl = R - 1 - (int)padding;
@@ -674,8 +747,9 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
                        (bf16*)xpT.data_ptr(), NHW, C);
     int qshift = __builtin_ctz((unsigned)Q);
     int pshift = qshift + __builtin_ctz((unsigned)P);
+    int BMK = K % 128 == 0 ? 128 : 64;     // kout tile (128 ~1.7x faster)
     long long mtiles = M / 64;
-    long long tiles_xy = (RSC / 64) * (K / 64);
+    long long tiles_xy = (RSC / 64) * (K / BMK);
     int splits = 1;
     while (splits < 64 && tiles_xy * splits < 512 &&
            (long long)splits * 2 <= mtiles)
@@ -685,16 +759,34 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
                           xp.options().dtype(at::kFloat));
     auto dw = at::empty({K, C, R, S},
                         xp.options().memory_format(at::MemoryFormat::ChannelsLast));
-    dim3 grid((unsigned)(RSC / 64), K / 64, splits);
-    hipLaunchKernelGGL(dw_gemm_kernel, grid, dim3(256), 0, stream,
-                       (const bf16*)dyT.data_ptr(),
-                       (const bf16*)xpT.data_ptr(), part.data_ptr<float>(),
-                       K, C, N, Hp, Wp, S, M, (int)RSC, mps,
-                       qshift, Q - 1, pshift, P - 1);
+    dim3 grid((unsigned)(RSC / 64), K / BMK, splits);
+    if (BMK == 128)
+      hipLaunchKernelGGL((dw_gemm_kernel<128>), grid, dim3(256), 0, stream,
+                         (const bf16*)dyT.data_ptr(),
+                         (const bf16*)xpT.data_ptr(), part.data_ptr<float>(),
+                         K, C, N, Hp, Wp, S, M, (int)RSC, mps,
+                         qshift, Q - 1, pshift, P - 1);
+    else
+      hipLaunchKernelGGL((dw_gemm_kernel<64>), grid, dim3(256), 0, stream,
+                         (const bf16*)dyT.data_ptr(),
+                         (const bf16*)xpT.data_ptr(), part.data_ptr<float>(),
+                         K, C, N, Hp, Wp, S, M, (int)RSC, mps,
+                         qshift, Q - 1, pshift, P - 1);
     long long L = (long long)K * RSC;
-    hipLaunchKernelGGL(colsum_to_bf16_kernel, dim3(grid_1d(L, 256)), dim3(256),
-                       0, stream, part.data_ptr<float>(), splits, L,
-                       (bf16*)dw.data_ptr());
+    if (splits > 8) {
+      auto part2 = at::empty({8, L}, xp.options().dtype(at::kFloat));
+      hipLaunchKernelGGL(colsum_stage_kernel,
+                         dim3(grid_1d(L, 256, 512), 8), dim3(256), 0, stream,
+                         part.data_ptr<float>(), splits / 8, L,
+                         part2.data_ptr<float>());
+      hipLaunchKernelGGL(colsum_to_bf16_kernel, dim3(grid_1d(L, 256)),
+                         dim3(256), 0, stream, part2.data_ptr<float>(), 8, L,
+                         (bf16*)dw.data_ptr());
+    } else {
+      hipLaunchKernelGGL(colsum_to_bf16_kernel, dim3(grid_1d(L, 256)),
+                         dim3(256), 0, stream, part.data_ptr<float>(), splits,
+                         L, (bf16*)dw.data_ptr());
+    }
     return dw;
   }
 
