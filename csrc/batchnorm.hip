@@ -1,12 +1,19 @@
 // NHWC BatchNorm2d fwd/bwd for MI355X (SURVEY.md §2a "BatchNorm2d").
 // channels_last layout puts C innermost, so per-channel reductions read
-// coalesced 16 B/lane vectors.  Structure:
-//   fwd (training): partial-sums kernel (fp32 atomics into a [2][C]
-//     workspace) -> finalize (mean/invstd + running-stat update) -> apply
-//     (normalize + scale/shift, vectorized, output in input dtype).
-//   fwd (eval): apply with running stats.
-//   bwd: partial sums of (dy, dy*xhat) -> finalize (gw, gb, coefficients)
-//     -> apply dx = gamma*invstd*(dy - mean(dy) - xhat*mean(dy*xhat)).
+// coalesced 16 B/lane vectors.  Launch-count is the binding constraint at
+// CIFAR sizes (2-17 MB activations: every extra kernel is ~6 us of latency
+// on ~2 us of work), so each direction is TWO kernels:
+//   fwd (training): partials kernel — block-level (sum, sumsq) rows, and the
+//     LAST block to finish (atomic counter + threadfence) column-sums the
+//     rows and finalizes mean/invstd + running stats in-kernel -> apply
+//     (normalize + scale/shift + optional fused residual-add / ELU).
+//   fwd (eval): finalize-from-running-stats -> apply.
+//   bwd: partials of (g, g*xhat) with the same last-block column sum into
+//     ws[2C] (g = dy, or dy*elu'(y) fused when the forward fused ELU)
+//     -> apply dx = gamma*invstd*(g - mean(g) - xhat*mean(g*xhat)), which
+//     also recomputes g and (optionally) writes it out as the residual
+//     gradient — the standalone elu_bwd launch and the g round-trip through
+//     HBM are gone.
 // All statistics fp32 regardless of activation dtype (bf16 data paths keep
 // fp32 BN stats — SURVEY.md §7 hard part 4).  C must divide 256 or be a
 // multiple of 256 (ResNet: 64/128/256/512) so each thread owns ONE channel
@@ -14,21 +21,49 @@
 
 #include "fedkit_common.h"
 
+#include <c10/hip/HIPFunctions.h>
+
 namespace {
 
 template <typename T, int VEC>
 struct alignas(sizeof(T) * VEC) VecT { T v[VEC]; };
 
-// Vectorized (16 B/lane) per-channel reduction, stage 1: each thread owns
-// VEC consecutive channels fixed across its grid-stride walk (C/VEC divides
-// the 256-thread block), accumulates in registers (4x unrolled so four 16-B
-// loads are in flight per lane), reduces across the block via LDS, and each
-// block writes its private partial row [2][C] — NO atomics; stage 2 (the
-// finalize / bwd-reduce kernels) sums the <=1024 partial rows.
+// elu'(y) from the SAVED OUTPUT y = elu(z): dy/dz = y > 0 ? 1 : y + 1
+__device__ __forceinline__ float elu_bwd_f(float y) {
+  return y > 0.f ? 1.f : y + 1.f;
+}
+
+// one zeroed int per device, allocated lazily outside the caching allocator;
+// the last block of every fused-reduction kernel resets it to 0 on its way
+// out, so consecutive launches on the stream can share it.
+int* bn_sync_counter() {
+  static int* counters[64] = {};
+  int dev = c10::hip::current_device();
+  if (!counters[dev]) {
+    void* p;
+    C10_HIP_CHECK(hipMalloc(&p, sizeof(int)));
+    C10_HIP_CHECK(hipMemset(p, 0, sizeof(int)));
+    counters[dev] = (int*)p;
+  }
+  return counters[dev];
+}
+
+// Vectorized (16 B/lane) per-channel reduction: each thread owns VEC
+// consecutive channels fixed across its grid-stride walk (C/VEC divides the
+// 256-thread block), accumulates in registers (4x unrolled so four 16-B
+// loads are in flight per lane), reduces across the block via LDS, writes
+// its private partial row [nb][2][C] (coalesced, NO atomics), and the last
+// block to finish reduces the rows (L2-hot) and finalizes the statistics.
 template <typename T, int VEC>
 __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
-                                   int Cv /* C/VEC */, int nb,
-                                   float* __restrict__ part /* [2][C][nb] */) {
+                                   int Cv /* C/VEC */,
+                                   float* __restrict__ part /* [nb][2][C] */,
+                                   int* __restrict__ counter, float eps,
+                                   float momentum, bool track,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var,
+                                   float* __restrict__ save_mean,
+                                   float* __restrict__ save_invstd) {
   using V = VecT<T, VEC>;
   const int C = Cv * VEC;
   const V* xv = reinterpret_cast<const V*>(x);
@@ -62,8 +97,7 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
   }
   const int members = 256 / Cv;
   const int g = tid % Cv;
-  // partials land TRANSPOSED ([2][C][nb], nb innermost) so the stage-2
-  // column sums read each channel's rows as one contiguous run
+  float* out = part + (long long)blockIdx.x * 2 * C;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     float* src = pass == 0 ? s : sq;
@@ -77,87 +111,36 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
 #pragma unroll
         for (int j = 0; j < VEC; ++j) acc[j] += red[(g + k * Cv) * VEC + j];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j)
-        part[(long long)(pass * C + c0 + j) * nb + blockIdx.x] = acc[j];
+      for (int j = 0; j < VEC; ++j) out[pass * C + c0 + j] = acc[j];
     }
   }
-}
 
-// stage 2: sum the per-block partial rows, then mean/invstd + running update
-__global__ void bn_finalize_kernel(const float* __restrict__ part, int nb,
-                                   int C, long long count, float eps,
-                                   float momentum, bool training, bool track,
-                                   float* __restrict__ running_mean,
-                                   float* __restrict__ running_var,
-                                   float* __restrict__ save_mean,
-                                   float* __restrict__ save_invstd) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float mean, var;
-  if (training) {
-    float s = 0.f, sq = 0.f;
-    for (int b = 0; b < nb; ++b) {
-      s += part[(long long)c * nb + b];
-      sq += part[(long long)(C + c) * nb + b];
+  // ---- last block: column-sum the partial rows + finalize statistics
+  __shared__ int is_last;
+  __threadfence();
+  if (tid == 0) is_last = (atomicAdd(counter, 1) == (int)gridDim.x - 1);
+  __syncthreads();
+  if (!is_last) return;
+  const int nb = gridDim.x;
+  const long long count = M;
+  for (int c = tid; c < C; c += blockDim.x) {
+    float cs = 0.f, csq = 0.f;
+    int b = 0;
+    for (; b + 3 < nb; b += 4) {
+      const float* r0 = part + (long long)b * 2 * C;
+      const float* r1 = r0 + 2 * C;
+      const float* r2 = r1 + 2 * C;
+      const float* r3 = r2 + 2 * C;
+      cs += (r0[c] + r1[c]) + (r2[c] + r3[c]);
+      csq += (r0[C + c] + r1[C + c]) + (r2[C + c] + r3[C + c]);
     }
-    mean = s / count;
-    var = fmaxf(sq / count - mean * mean, 0.f);  // biased
-    if (track) {
-      float unbiased = count > 1 ? var * count / (count - 1) : var;
-      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
-      running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    for (; b < nb; ++b) {
+      const float* r = part + (long long)b * 2 * C;
+      cs += r[c];
+      csq += r[C + c];
     }
-  } else {
-    mean = running_mean[c];
-    var = running_var[c];
-  }
-  save_mean[c] = mean;
-  save_invstd[c] = rsqrtf(var + eps);
-}
-
-// stage 2: column sums of the [nb][C2] partials matrix into ws[C2].
-// One WAVE per column: lane l sums rows l, l+64, ... (64 independent load
-// chains), then a wave shuffle tree.  The previous single-workgroup serial
-// loop over nb<=1024 rows was latency-bound at ~210 us and dominated the
-// whole training step (65% of GPU time in rocprof); this form is ~5 us.
-__global__ void bn_colsum_kernel(const float* __restrict__ part, int nb,
-                                 int C2 /* 2*C */, float* __restrict__ ws) {
-  int col = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
-  int lane = threadIdx.x & 63;
-  if (col >= C2) return;
-  float s = 0.f;
-  for (int b = lane; b < nb; b += 64) s += part[(long long)col * nb + b];
-#pragma unroll
-  for (int off = 32; off; off >>= 1) s += __shfl_down(s, off, 64);
-  if (lane == 0) ws[col] = s;
-}
-
-// fused stage 2 + statistics for the TRAINING forward: one wave per channel
-// sums both partial columns (sum, sumsq) and lane 0 derives mean/invstd and
-// updates the running stats — one launch instead of colsum + finalize.
-__global__ void bn_colsum_finalize_kernel(
-    const float* __restrict__ part, int nb, int C, long long count, float eps,
-    float momentum, bool track, float* __restrict__ running_mean,
-    float* __restrict__ running_var, float* __restrict__ save_mean,
-    float* __restrict__ save_invstd) {
-  int c = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
-  int lane = threadIdx.x & 63;
-  if (c >= C) return;
-  float s = 0.f, sq = 0.f;
-  const float* col_s = part + (long long)c * nb;
-  const float* col_sq = part + (long long)(C + c) * nb;
-  for (int b = lane; b < nb; b += 64) {
-    s += col_s[b];
-    sq += col_sq[b];
-  }
-#pragma unroll
-  for (int off = 32; off; off >>= 1) {
-    s += __shfl_down(s, off, 64);
-    sq += __shfl_down(sq, off, 64);
-  }
-  if (lane == 0) {
-    float mean = s / count;
-    float var = fmaxf(sq / count - mean * mean, 0.f);  // biased
+    float mean = cs / count;
+    float var = fmaxf(csq / count - mean * mean, 0.f);  // biased
     if (track) {
       float unbiased = count > 1 ? var * count / (count - 1) : var;
       running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
@@ -166,6 +149,19 @@ __global__ void bn_colsum_finalize_kernel(
     save_mean[c] = mean;
     save_invstd[c] = rsqrtf(var + eps);
   }
+  if (tid == 0) *counter = 0;
+}
+
+// eval-mode statistics: mean/invstd straight from the running buffers
+__global__ void bn_eval_stats_kernel(int C, float eps,
+                                     const float* __restrict__ running_mean,
+                                     const float* __restrict__ running_var,
+                                     float* __restrict__ save_mean,
+                                     float* __restrict__ save_invstd) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  save_mean[c] = running_mean[c];
+  save_invstd[c] = rsqrtf(running_var[c] + eps);
 }
 
 // normalize + scale/shift, with optional fused residual add (RES) and ELU
@@ -202,17 +198,25 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
   }
 }
 
-template <typename T, int VEC>
+// backward partial sums of (g, g*xhat) where g = dy (plain BN) or
+// dy * elu'(y) (forward fused ELU; y is the saved post-activation output).
+// Same block-row + last-block-column-sum structure as the forward; the last
+// block writes the finished sums straight into ws[2C] (gb, gw).
+template <typename T, int VEC, bool ELU>
 __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
-                                       const T* __restrict__ gy, long long M,
-                                       int Cv, int nb,
+                                       const T* __restrict__ gy,
+                                       const T* __restrict__ yout,
+                                       long long M, int Cv,
                                        const float* __restrict__ mean,
                                        const float* __restrict__ invstd,
-                                       float* __restrict__ part /* [2][C][nb] */) {
+                                       float* __restrict__ part /* [nb][2][C] */,
+                                       int* __restrict__ counter,
+                                       float* __restrict__ ws /* [2][C] */) {
   using V = VecT<T, VEC>;
   const int C = Cv * VEC;
   const V* xv = reinterpret_cast<const V*>(x);
   const V* gv = reinterpret_cast<const V*>(gy);
+  const V* yv = reinterpret_cast<const V*>(yout);
   __shared__ float red[256 * VEC];
   long long total = M * Cv;
   long long stride = (long long)gridDim.x * blockDim.x;
@@ -229,9 +233,18 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
   long long i = i0;
   for (; i + stride < total; i += 2 * stride) {
     V a0 = xv[i], g0 = gv[i], a1 = xv[i + stride], g1 = gv[i + stride];
+    V y0, y1;
+    if (ELU) {
+      y0 = yv[i];
+      y1 = yv[i + stride];
+    }
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       float gg0 = to_f32(g0.v[j]), gg1 = to_f32(g1.v[j]);
+      if (ELU) {
+        gg0 *= elu_bwd_f(to_f32(y0.v[j]));
+        gg1 *= elu_bwd_f(to_f32(y1.v[j]));
+      }
       sdy[j] += gg0 + gg1;
       sdyx[j] += gg0 * (to_f32(a0.v[j]) - m[j]) * is[j]
                + gg1 * (to_f32(a1.v[j]) - m[j]) * is[j];
@@ -239,15 +252,19 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
   }
   for (; i < total; i += stride) {
     V a = xv[i], g = gv[i];
+    V yy;
+    if (ELU) yy = yv[i];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       float gg = to_f32(g.v[j]);
+      if (ELU) gg *= elu_bwd_f(to_f32(yy.v[j]));
       sdy[j] += gg;
       sdyx[j] += gg * (to_f32(a.v[j]) - m[j]) * is[j];
     }
   }
   const int members = 256 / Cv;
   const int g8 = tid % Cv;
+  float* out = part + (long long)blockIdx.x * 2 * C;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     float* src = pass == 0 ? sdy : sdyx;
@@ -261,45 +278,71 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
 #pragma unroll
         for (int j = 0; j < VEC; ++j) acc[j] += red[(g8 + k * Cv) * VEC + j];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j)
-        part[(long long)(pass * C + c0 + j) * nb + blockIdx.x] = acc[j];
+      for (int j = 0; j < VEC; ++j) out[pass * C + c0 + j] = acc[j];
     }
   }
+
+  __shared__ int is_last;
+  __threadfence();
+  if (tid == 0) is_last = (atomicAdd(counter, 1) == (int)gridDim.x - 1);
+  __syncthreads();
+  if (!is_last) return;
+  const int nb = gridDim.x;
+  for (int c = tid; c < 2 * C; c += blockDim.x) {
+    float cs = 0.f;
+    int b = 0;
+    for (; b + 3 < nb; b += 4) {
+      const float* r0 = part + (long long)b * 2 * C + c;
+      cs += (r0[0] + r0[2 * C]) + (r0[4 * C] + r0[6 * C]);
+    }
+    for (; b < nb; ++b) cs += part[(long long)b * 2 * C + c];
+    ws[c] = cs;
+  }
+  if (tid == 0) *counter = 0;
 }
 
-template <typename T, int VEC>
+// dx = gamma*invstd*(g - mean(g) - xhat*mean(g*xhat)); recomputes
+// g = dy * elu'(y) when fused and optionally streams g out (the residual
+// branch gradient) — WANTG costs one extra write but no extra launch.
+template <typename T, int VEC, bool ELU, bool WANTG>
 __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
                                     const T* __restrict__ gy,
-                                    T* __restrict__ gx,
+                                    const T* __restrict__ yout,
+                                    T* __restrict__ gx, T* __restrict__ gout,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
                                     const float* __restrict__ gamma,
                                     const float* __restrict__ ws,
-                                    long long M, long long nvec, int Cv,
-                                    int VECC) {
+                                    long long M, long long nvec, int Cv) {
   using V = VecT<T, VEC>;
   const V* xv = reinterpret_cast<const V*>(x);
   const V* gv = reinterpret_cast<const V*>(gy);
+  const V* yv = reinterpret_cast<const V*>(yout);
   V* ov = reinterpret_cast<V*>(gx);
+  V* gov = reinterpret_cast<V*>(gout);
   int C = Cv * VEC;
   float inv_count = 1.f / (float)M;
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
        i < nvec; i += (long long)gridDim.x * blockDim.x) {
     int c0 = (int)(i % Cv) * VEC;
-    V a = xv[i], g = gv[i], r;
+    V a = xv[i], g = gv[i], r, go;
+    V yy;
+    if (ELU) yy = yv[i];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       int c = c0 + j;
       float is = invstd[c];
       float xhat = (to_f32(a.v[j]) - mean[c]) * is;
       float dy = to_f32(g.v[j]);
+      if (ELU) dy *= elu_bwd_f(to_f32(yy.v[j]));
+      if (WANTG) from_f32(dy, go.v[j]);
       float val = gamma[c] * is *
           (dy - ws[c] * inv_count - xhat * ws[C + c] * inv_count);
       from_f32(val, r.v[j]);
     }
     ov[i] = r;
+    if (WANTG) gov[i] = go;
   }
-  (void)VECC;
 }
 
 void check_nhwc(const at::Tensor& x) {
@@ -336,25 +379,22 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
       TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
                   "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                   ") | 256, got C=", C);
-      int nb = grid_1d(M * C / VEC, 256, 1024);
-      auto part = at::empty({2 * C, nb}, fopts);
+      int nb = grid_1d(M * C / VEC, 256, 640);
+      auto part = at::empty({nb, 2, C}, fopts);
+      bool track = running_mean.defined();
       hipLaunchKernelGGL((bn_partials_kernel<scalar_t, VEC>),
                          dim3(nb), dim3(256), 0,
                          stream, (const scalar_t*)x.data_ptr(), M, C / VEC,
-                         nb, part.data_ptr<float>());
-      hipLaunchKernelGGL(bn_colsum_finalize_kernel, dim3((C + 3) / 4),
-                         dim3(256), 0, stream, part.data_ptr<float>(), nb, C,
-                         M, (float)eps, (float)momentum,
-                         running_mean.defined(),
-                         running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
-                         running_var.defined() ? running_var.data_ptr<float>() : nullptr,
+                         part.data_ptr<float>(), bn_sync_counter(),
+                         (float)eps, (float)momentum, track,
+                         track ? running_mean.data_ptr<float>() : nullptr,
+                         track ? running_var.data_ptr<float>() : nullptr,
                          save_mean.data_ptr<float>(),
                          save_invstd.data_ptr<float>());
     });
   } else {
-    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
-                       0, stream, (const float*)nullptr, 0, C, M, (float)eps,
-                       (float)momentum, false, false,
+    hipLaunchKernelGGL(bn_eval_stats_kernel, dim3((C + 255) / 256), dim3(256),
+                       0, stream, C, (float)eps,
                        running_mean.data_ptr<float>(),
                        running_var.data_ptr<float>(),
                        save_mean.data_ptr<float>(),
@@ -393,10 +433,15 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
   return {y, save_mean, save_invstd};
 }
 
+// elu_y: the saved post-ELU output when the forward fused ELU (elu' is
+// recomputed from it inside both backward kernels); want_g additionally
+// returns g = dy*elu'(y) (the residual-branch gradient) as a 4th output.
 std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& gamma,
                                       const at::Tensor& save_mean,
-                                      const at::Tensor& save_invstd) {
+                                      const at::Tensor& save_invstd,
+                                      c10::optional<at::Tensor> elu_y,
+                                      bool want_g) {
   check_nhwc(x);
   check_nhwc(gy);
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
@@ -406,35 +451,53 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
   auto gx = at::empty_like(x);
   auto stream = fedkit_stream();
   auto gamma_f = gamma.contiguous();
+  const bool elu = elu_y.has_value();
+  const void* y_ptr = elu ? elu_y->data_ptr() : nullptr;
+  TORCH_CHECK(!want_g || elu, "want_g requires the fused-ELU backward");
+  at::Tensor gout;
+  if (want_g) gout = at::empty_like(gy);
   DISPATCH_F32_BF16(x, "bn_bwd_partials", {
     constexpr int VEC = 16 / sizeof(scalar_t);
     TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
                 "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                 ") | 256, got C=", C);
-    int nb = grid_1d(M * C / VEC, 256, 1024);
-    auto part = at::empty({2 * C, nb}, fopts);
-    hipLaunchKernelGGL((bn_bwd_partials_kernel<scalar_t, VEC>),
-                       dim3(nb), dim3(256), 0,
-                       stream, (const scalar_t*)x.data_ptr(),
-                       (const scalar_t*)gy.data_ptr(), M, C / VEC, nb,
-                       save_mean.data_ptr<float>(),
-                       save_invstd.data_ptr<float>(), part.data_ptr<float>());
-    hipLaunchKernelGGL(bn_colsum_kernel, dim3((2 * C + 3) / 4), dim3(256),
-                       0, stream, part.data_ptr<float>(), nb, 2 * C,
-                       ws.data_ptr<float>());
-  });
-  DISPATCH_F32_BF16(x, "bn_bwd_apply", {
-    constexpr int VEC = 16 / sizeof(scalar_t);
+    int nb = grid_1d(M * C / VEC, 256, 640);
+    auto part = at::empty({nb, 2, C}, fopts);
+    auto launch1 = [&](auto ekind) {
+      hipLaunchKernelGGL((bn_bwd_partials_kernel<scalar_t, VEC,
+                                                 decltype(ekind)::value>),
+                         dim3(nb), dim3(256), 0,
+                         stream, (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)gy.data_ptr(),
+                         (const scalar_t*)y_ptr, M, C / VEC,
+                         save_mean.data_ptr<float>(),
+                         save_invstd.data_ptr<float>(), part.data_ptr<float>(),
+                         bn_sync_counter(), ws.data_ptr<float>());
+    };
     long long nvec = M * C / VEC;
-    hipLaunchKernelGGL((bn_bwd_apply_kernel<scalar_t, VEC>),
-                       dim3(grid_1d(nvec, 256)), dim3(256), 0, stream,
-                       (const scalar_t*)x.data_ptr(),
-                       (const scalar_t*)gy.data_ptr(), (scalar_t*)gx.data_ptr(),
-                       save_mean.data_ptr<float>(),
-                       save_invstd.data_ptr<float>(),
-                       gamma_f.data_ptr<float>(), ws.data_ptr<float>(), M,
-                       nvec, C / VEC, VEC);
+    auto launch2 = [&](auto ekind, auto wkind) {
+      hipLaunchKernelGGL((bn_bwd_apply_kernel<scalar_t, VEC,
+                                              decltype(ekind)::value,
+                                              decltype(wkind)::value>),
+                         dim3(grid_1d(nvec, 256)), dim3(256), 0, stream,
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)gy.data_ptr(),
+                         (const scalar_t*)y_ptr, (scalar_t*)gx.data_ptr(),
+                         want_g ? (scalar_t*)gout.data_ptr() : nullptr,
+                         save_mean.data_ptr<float>(),
+                         save_invstd.data_ptr<float>(),
+                         gamma_f.data_ptr<float>(), ws.data_ptr<float>(), M,
+                         nvec, C / VEC);
+    };
+    using T0 = std::integral_constant<bool, false>;
+    using T1 = std::integral_constant<bool, true>;
+    if (elu) launch1(T1{}); else launch1(T0{});
+    if (elu && want_g)  launch2(T1{}, T1{});
+    else if (elu)       launch2(T1{}, T0{});
+    else                launch2(T0{}, T0{});
   });
-  // gw = sum(dy * xhat), gb = sum(dy): zero-copy views into ws
-  return {gx, ws.select(0, 1), ws.select(0, 0)};
+  // gw = sum(g * xhat), gb = sum(g): zero-copy views into ws
+  std::vector<at::Tensor> out = {gx, ws.select(0, 1), ws.select(0, 0)};
+  if (want_g) out.push_back(gout);
+  return out;
 }

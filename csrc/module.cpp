@@ -31,7 +31,9 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
 std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& gamma,
                                       const at::Tensor& save_mean,
-                                      const at::Tensor& save_invstd);
+                                      const at::Tensor& save_invstd,
+                                      c10::optional<at::Tensor> elu_y,
+                                      bool want_g);
 
 at::Tensor fedkit_conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                              long stride, long padding);
@@ -65,7 +67,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("momentum"), py::arg("eps"),
         py::arg("residual") = c10::nullopt, py::arg("elu") = false);
   m.def("bn_bwd", &fedkit_bn_bwd,
-        "NHWC BatchNorm bwd: returns (gx, gw, gb)");
+        "NHWC BatchNorm bwd (optionally fused with ELU backward from the "
+        "saved output): returns (gx, gw, gb[, g])",
+        py::arg("gy"), py::arg("x"), py::arg("gamma"), py::arg("save_mean"),
+        py::arg("save_invstd"), py::arg("elu_y") = c10::nullopt,
+        py::arg("want_g") = false);
   m.def("conv2d_fwd", &fedkit_conv2d_fwd,
         "NHWC implicit-GEMM conv fwd on MFMA (3x3/1x1, stride 1/2)");
   m.def("conv2d_pad_input", &fedkit_conv2d_pad_input, "zero-pad NHWC input");

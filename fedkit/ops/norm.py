@@ -72,10 +72,13 @@ class _BnActFn(torch.autograd.Function):
     def backward(ctx, gy):
         x, weight, save_mean, save_invstd, y = ctx.saved_tensors
         gy = gy.contiguous(memory_format=torch.channels_last)
-        g = _ext().elu_bwd(gy, y)
-        gx, gw, gb = _ext().bn_bwd(g, x, weight, save_mean, save_invstd)
-        gres = g if ctx.has_res else None
-        return gx, gw, gb, None, None, None, None, None, gres
+        # elu' is fused into both bn_bwd kernels (recomputed from the saved
+        # output y); the residual grad g is streamed out of the apply kernel
+        # only when the residual branch needs it
+        out = _ext().bn_bwd(gy, x, weight, save_mean, save_invstd,
+                            elu_y=y, want_g=ctx.has_res)
+        gres = out[3] if ctx.has_res else None
+        return out[0], out[1], out[2], None, None, None, None, None, gres
 
 
 class FedBatchNorm2d(nn.BatchNorm2d):
