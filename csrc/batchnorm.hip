@@ -33,19 +33,28 @@ __device__ __forceinline__ float elu_bwd_f(float y) {
   return y > 0.f ? 1.f : y + 1.f;
 }
 
-// one zeroed int per device, allocated lazily outside the caching allocator;
-// the last block of every fused-reduction kernel resets it to 0 on its way
-// out, so consecutive launches on the stream can share it.
-int* bn_sync_counter() {
-  static int* counters[64] = {};
+// Per-device persistent sync state, allocated lazily outside the caching
+// allocator and zero-initialized ONCE: a block-arrival counter followed by
+// a 2*Cmax fp32 accumulator.  Every fused-reduction kernel's blocks
+// unsafeAtomicAdd their channel sums into the accumulator; the LAST block
+// (counter) finalizes from it and re-zeros both on its way out, so
+// consecutive launches on a stream can share the buffer.  (A per-call
+// partials slab + last-block column sum was 10-20x slower: the rows land
+// in OTHER XCDs' L2s, so the reading block pays full HBM latency with only
+// C threads of parallelism.)  One stream per device at a time — which is
+// how fedkit runs BN — is a precondition.
+constexpr int kBnCMax = 4096;
+float* bn_sync_state() {
+  static float* state[64] = {};
   int dev = c10::hip::current_device();
-  if (!counters[dev]) {
+  if (!state[dev]) {
     void* p;
-    C10_HIP_CHECK(hipMalloc(&p, sizeof(int)));
-    C10_HIP_CHECK(hipMemset(p, 0, sizeof(int)));
-    counters[dev] = (int*)p;
+    size_t bytes = sizeof(float) * (2 * kBnCMax + 2);
+    C10_HIP_CHECK(hipMalloc(&p, bytes));
+    C10_HIP_CHECK(hipMemset(p, 0, bytes));
+    state[dev] = (float*)p;
   }
-  return counters[dev];
+  return state[dev];
 }
 
 // Vectorized (16 B/lane) per-channel reduction: each thread owns VEC
@@ -57,8 +66,8 @@ int* bn_sync_counter() {
 template <typename T, int VEC>
 __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
                                    int Cv /* C/VEC */,
-                                   float* __restrict__ part /* [nb][2][C] */,
-                                   int* __restrict__ counter, float eps,
+                                   float* __restrict__ state /* counter+acc */,
+                                   float eps,
                                    float momentum, bool track,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var,
@@ -97,7 +106,8 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
   }
   const int members = 256 / Cv;
   const int g = tid % Cv;
-  float* out = part + (long long)blockIdx.x * 2 * C;
+  int* counter = (int*)state;
+  float* acc_buf = state + 2;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     float* src = pass == 0 ? s : sq;
@@ -111,34 +121,25 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
 #pragma unroll
         for (int j = 0; j < VEC; ++j) acc[j] += red[(g + k * Cv) * VEC + j];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j) out[pass * C + c0 + j] = acc[j];
+      for (int j = 0; j < VEC; ++j)
+        unsafeAtomicAdd(&acc_buf[pass * C + c0 + j], acc[j]);
     }
   }
 
-  // ---- last block: column-sum the partial rows + finalize statistics
+  // ---- last block: finalize statistics from the 2C accumulated sums
+  // (the barrier makes every thread's atomics precede tid 0's counter
+  // bump: __syncthreads waits each thread's own vmcnt before the barrier)
   __shared__ int is_last;
+  __syncthreads();
   __threadfence();
   if (tid == 0) is_last = (atomicAdd(counter, 1) == (int)gridDim.x - 1);
   __syncthreads();
   if (!is_last) return;
-  const int nb = gridDim.x;
   const long long count = M;
   for (int c = tid; c < C; c += blockDim.x) {
-    float cs = 0.f, csq = 0.f;
-    int b = 0;
-    for (; b + 3 < nb; b += 4) {
-      const float* r0 = part + (long long)b * 2 * C;
-      const float* r1 = r0 + 2 * C;
-      const float* r2 = r1 + 2 * C;
-      const float* r3 = r2 + 2 * C;
-      cs += (r0[c] + r1[c]) + (r2[c] + r3[c]);
-      csq += (r0[C + c] + r1[C + c]) + (r2[C + c] + r3[C + c]);
-    }
-    for (; b < nb; ++b) {
-      const float* r = part + (long long)b * 2 * C;
-      cs += r[c];
-      csq += r[C + c];
-    }
+    float cs = acc_buf[c], csq = acc_buf[C + c];
+    acc_buf[c] = 0.f;                      // re-zero for the next launch
+    acc_buf[C + c] = 0.f;
     float mean = cs / count;
     float var = fmaxf(csq / count - mean * mean, 0.f);  // biased
     if (track) {
@@ -209,8 +210,7 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
                                        long long M, int Cv,
                                        const float* __restrict__ mean,
                                        const float* __restrict__ invstd,
-                                       float* __restrict__ part /* [nb][2][C] */,
-                                       int* __restrict__ counter,
+                                       float* __restrict__ state,
                                        float* __restrict__ ws /* [2][C] */) {
   using V = VecT<T, VEC>;
   const int C = Cv * VEC;
@@ -264,7 +264,8 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
   }
   const int members = 256 / Cv;
   const int g8 = tid % Cv;
-  float* out = part + (long long)blockIdx.x * 2 * C;
+  int* counter = (int*)state;
+  float* acc_buf = state + 2;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     float* src = pass == 0 ? sdy : sdyx;
@@ -278,25 +279,20 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
 #pragma unroll
         for (int j = 0; j < VEC; ++j) acc[j] += red[(g8 + k * Cv) * VEC + j];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j) out[pass * C + c0 + j] = acc[j];
+      for (int j = 0; j < VEC; ++j)
+        unsafeAtomicAdd(&acc_buf[pass * C + c0 + j], acc[j]);
     }
   }
 
   __shared__ int is_last;
+  __syncthreads();
   __threadfence();
   if (tid == 0) is_last = (atomicAdd(counter, 1) == (int)gridDim.x - 1);
   __syncthreads();
   if (!is_last) return;
-  const int nb = gridDim.x;
   for (int c = tid; c < 2 * C; c += blockDim.x) {
-    float cs = 0.f;
-    int b = 0;
-    for (; b + 3 < nb; b += 4) {
-      const float* r0 = part + (long long)b * 2 * C + c;
-      cs += (r0[0] + r0[2 * C]) + (r0[4 * C] + r0[6 * C]);
-    }
-    for (; b < nb; ++b) cs += part[(long long)b * 2 * C + c];
-    ws[c] = cs;
+    ws[c] = acc_buf[c];
+    acc_buf[c] = 0.f;                      // re-zero for the next launch
   }
   if (tid == 0) *counter = 0;
 }
@@ -379,13 +375,13 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
       TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
                   "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                   ") | 256, got C=", C);
+      TORCH_CHECK(C <= kBnCMax, "bn supports C <= ", kBnCMax);
       int nb = grid_1d(M * C / VEC, 256, 640);
-      auto part = at::empty({nb, 2, C}, fopts);
       bool track = running_mean.defined();
       hipLaunchKernelGGL((bn_partials_kernel<scalar_t, VEC>),
                          dim3(nb), dim3(256), 0,
                          stream, (const scalar_t*)x.data_ptr(), M, C / VEC,
-                         part.data_ptr<float>(), bn_sync_counter(),
+                         bn_sync_state(),
                          (float)eps, (float)momentum, track,
                          track ? running_mean.data_ptr<float>() : nullptr,
                          track ? running_var.data_ptr<float>() : nullptr,
@@ -461,8 +457,8 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
     TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
                 "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                 ") | 256, got C=", C);
+    TORCH_CHECK(C <= kBnCMax, "bn supports C <= ", kBnCMax);
     int nb = grid_1d(M * C / VEC, 256, 640);
-    auto part = at::empty({nb, 2, C}, fopts);
     auto launch1 = [&](auto ekind) {
       hipLaunchKernelGGL((bn_bwd_partials_kernel<scalar_t, VEC,
                                                  decltype(ekind)::value>),
@@ -471,8 +467,8 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                          (const scalar_t*)gy.data_ptr(),
                          (const scalar_t*)y_ptr, M, C / VEC,
                          save_mean.data_ptr<float>(),
-                         save_invstd.data_ptr<float>(), part.data_ptr<float>(),
-                         bn_sync_counter(), ws.data_ptr<float>());
+                         save_invstd.data_ptr<float>(),
+                         bn_sync_state(), ws.data_ptr<float>());
     };
     long long nvec = M * C / VEC;
     auto launch2 = [&](auto ekind, auto wkind) {
