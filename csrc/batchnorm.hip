@@ -1,19 +1,21 @@
 // NHWC BatchNorm2d fwd/bwd for MI355X (SURVEY.md §2a "BatchNorm2d").
 // channels_last layout puts C innermost, so per-channel reductions read
-// coalesced 16 B/lane vectors.  Launch-count is the binding constraint at
-// CIFAR sizes (2-17 MB activations: every extra kernel is ~6 us of latency
-// on ~2 us of work), so each direction is TWO kernels:
-//   fwd (training): partials kernel — block-level (sum, sumsq) rows, and the
-//     LAST block to finish (atomic counter + threadfence) column-sums the
-//     rows and finalizes mean/invstd + running stats in-kernel -> apply
+// coalesced 16 B/lane vectors.  Structure:
+//   fwd (training): partial-sums kernel (per-block rows, no atomics)
+//     -> fused colsum+finalize (mean/invstd + running-stat update) -> apply
 //     (normalize + scale/shift + optional fused residual-add / ELU).
-//   fwd (eval): finalize-from-running-stats -> apply.
-//   bwd: partials of (g, g*xhat) with the same last-block column sum into
-//     ws[2C] (g = dy, or dy*elu'(y) fused when the forward fused ELU)
-//     -> apply dx = gamma*invstd*(g - mean(g) - xhat*mean(g*xhat)), which
-//     also recomputes g and (optionally) writes it out as the residual
-//     gradient — the standalone elu_bwd launch and the g round-trip through
-//     HBM are gone.
+//   fwd (eval): stats-from-running -> apply.
+//   bwd: partial sums of (g, g*xhat) -> colsum -> apply
+//     dx = gamma*invstd*(g - mean(g) - xhat*mean(g*xhat)), where
+//     g = dy, or dy*elu'(y) fused in-kernel when the forward fused ELU
+//     (kills the standalone elu_bwd launch and the g HBM round-trip; the
+//     residual-branch gradient streams out of apply only when needed).
+// Measured dead ends, kept out on purpose: a last-block column sum of the
+// partials slab (rows land in other XCDs' L2s -> HBM latency on C lanes,
+// 10-20x slower) and cross-block fp32 atomics into a shared 2C accumulator
+// (same-address atomics from 8 XCDs serialize at coherence-point latency,
+// ~190 us flat).  The separate wave-per-column colsum launch is ~6 us and
+// is the latency floor here.
 // All statistics fp32 regardless of activation dtype (bf16 data paths keep
 // fp32 BN stats — SURVEY.md §7 hard part 4).  C must divide 256 or be a
 // multiple of 256 (ResNet: 64/128/256/512) so each thread owns ONE channel
@@ -21,58 +23,21 @@
 
 #include "fedkit_common.h"
 
-#include <c10/hip/HIPFunctions.h>
-
 namespace {
 
 template <typename T, int VEC>
 struct alignas(sizeof(T) * VEC) VecT { T v[VEC]; };
 
-// elu'(y) from the SAVED OUTPUT y = elu(z): dy/dz = y > 0 ? 1 : y + 1
-__device__ __forceinline__ float elu_bwd_f(float y) {
-  return y > 0.f ? 1.f : y + 1.f;
-}
-
-// Per-device persistent sync state, allocated lazily outside the caching
-// allocator and zero-initialized ONCE: a block-arrival counter followed by
-// a 2*Cmax fp32 accumulator.  Every fused-reduction kernel's blocks
-// unsafeAtomicAdd their channel sums into the accumulator; the LAST block
-// (counter) finalizes from it and re-zeros both on its way out, so
-// consecutive launches on a stream can share the buffer.  (A per-call
-// partials slab + last-block column sum was 10-20x slower: the rows land
-// in OTHER XCDs' L2s, so the reading block pays full HBM latency with only
-// C threads of parallelism.)  One stream per device at a time — which is
-// how fedkit runs BN — is a precondition.
-constexpr int kBnCMax = 4096;
-float* bn_sync_state() {
-  static float* state[64] = {};
-  int dev = c10::hip::current_device();
-  if (!state[dev]) {
-    void* p;
-    size_t bytes = sizeof(float) * (2 * kBnCMax + 2);
-    C10_HIP_CHECK(hipMalloc(&p, bytes));
-    C10_HIP_CHECK(hipMemset(p, 0, bytes));
-    state[dev] = (float*)p;
-  }
-  return state[dev];
-}
-
-// Vectorized (16 B/lane) per-channel reduction: each thread owns VEC
-// consecutive channels fixed across its grid-stride walk (C/VEC divides the
-// 256-thread block), accumulates in registers (4x unrolled so four 16-B
-// loads are in flight per lane), reduces across the block via LDS, writes
-// its private partial row [nb][2][C] (coalesced, NO atomics), and the last
-// block to finish reduces the rows (L2-hot) and finalizes the statistics.
+// Vectorized (16 B/lane) per-channel reduction, stage 1: each thread owns
+// VEC consecutive channels fixed across its grid-stride walk (C/VEC divides
+// the 256-thread block), accumulates in registers (4x unrolled so four 16-B
+// loads are in flight per lane), reduces across the block via LDS, and each
+// block writes its private partial row [2][C] — NO atomics; stage 2 (the
+// finalize / bwd-reduce kernels) sums the <=1024 partial rows.
 template <typename T, int VEC>
 __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
                                    int Cv /* C/VEC */,
-                                   float* __restrict__ state /* counter+acc */,
-                                   float eps,
-                                   float momentum, bool track,
-                                   float* __restrict__ running_mean,
-                                   float* __restrict__ running_var,
-                                   float* __restrict__ save_mean,
-                                   float* __restrict__ save_invstd) {
+                                   float* __restrict__ part /* [nb][2][C] */) {
   using V = VecT<T, VEC>;
   const int C = Cv * VEC;
   const V* xv = reinterpret_cast<const V*>(x);
@@ -106,8 +71,7 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
   }
   const int members = 256 / Cv;
   const int g = tid % Cv;
-  int* counter = (int*)state;
-  float* acc_buf = state + 2;
+  float* out = part + (long long)blockIdx.x * 2 * C;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     float* src = pass == 0 ? s : sq;
@@ -121,27 +85,85 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
 #pragma unroll
         for (int j = 0; j < VEC; ++j) acc[j] += red[(g + k * Cv) * VEC + j];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j)
-        unsafeAtomicAdd(&acc_buf[pass * C + c0 + j], acc[j]);
+      for (int j = 0; j < VEC; ++j) out[pass * C + c0 + j] = acc[j];
     }
   }
+}
 
-  // ---- last block: finalize statistics from the 2C accumulated sums
-  // (the barrier makes every thread's atomics precede tid 0's counter
-  // bump: __syncthreads waits each thread's own vmcnt before the barrier)
-  __shared__ int is_last;
-  __syncthreads();
-  __threadfence();
-  if (tid == 0) is_last = (atomicAdd(counter, 1) == (int)gridDim.x - 1);
-  __syncthreads();
-  if (!is_last) return;
-  const long long count = M;
-  for (int c = tid; c < C; c += blockDim.x) {
-    float cs = acc_buf[c], csq = acc_buf[C + c];
-    acc_buf[c] = 0.f;                      // re-zero for the next launch
-    acc_buf[C + c] = 0.f;
-    float mean = cs / count;
-    float var = fmaxf(csq / count - mean * mean, 0.f);  // biased
+// stage 2: sum the per-block partial rows, then mean/invstd + running update
+__global__ void bn_finalize_kernel(const float* __restrict__ part, int nb,
+                                   int C, long long count, float eps,
+                                   float momentum, bool training, bool track,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var,
+                                   float* __restrict__ save_mean,
+                                   float* __restrict__ save_invstd) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float mean, var;
+  if (training) {
+    float s = 0.f, sq = 0.f;
+    for (int b = 0; b < nb; ++b) {
+      s += part[(long long)b * 2 * C + c];
+      sq += part[(long long)b * 2 * C + C + c];
+    }
+    mean = s / count;
+    var = fmaxf(sq / count - mean * mean, 0.f);  // biased
+    if (track) {
+      float unbiased = count > 1 ? var * count / (count - 1) : var;
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    }
+  } else {
+    mean = running_mean[c];
+    var = running_var[c];
+  }
+  save_mean[c] = mean;
+  save_invstd[c] = rsqrtf(var + eps);
+}
+
+// stage 2: column sums of the [nb][C2] partials matrix into ws[C2].
+// One WAVE per column: lane l sums rows l, l+64, ... (64 independent load
+// chains), then a wave shuffle tree.  The previous single-workgroup serial
+// loop over nb<=1024 rows was latency-bound at ~210 us and dominated the
+// whole training step (65% of GPU time in rocprof); this form is ~5 us.
+__global__ void bn_colsum_kernel(const float* __restrict__ part, int nb,
+                                 int C2 /* 2*C */, float* __restrict__ ws) {
+  int col = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  int lane = threadIdx.x & 63;
+  if (col >= C2) return;
+  float s = 0.f;
+  for (int b = lane; b < nb; b += 64) s += part[(long long)b * C2 + col];
+#pragma unroll
+  for (int off = 32; off; off >>= 1) s += __shfl_down(s, off, 64);
+  if (lane == 0) ws[col] = s;
+}
+
+// fused stage 2 + statistics for the TRAINING forward: one wave per channel
+// sums both partial columns (sum, sumsq) and lane 0 derives mean/invstd and
+// updates the running stats — one launch instead of colsum + finalize.
+__global__ void bn_colsum_finalize_kernel(
+    const float* __restrict__ part, int nb, int C, long long count, float eps,
+    float momentum, bool track, float* __restrict__ running_mean,
+    float* __restrict__ running_var, float* __restrict__ save_mean,
+    float* __restrict__ save_invstd) {
+  int c = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  int lane = threadIdx.x & 63;
+  if (c >= C) return;
+  float s = 0.f, sq = 0.f;
+  for (int b = lane; b < nb; b += 64) {
+    const float* row = part + (long long)b * 2 * C;
+    s += row[c];
+    sq += row[C + c];
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) {
+    s += __shfl_down(s, off, 64);
+    sq += __shfl_down(sq, off, 64);
+  }
+  if (lane == 0) {
+    float mean = s / count;
+    float var = fmaxf(sq / count - mean * mean, 0.f);  // biased
     if (track) {
       float unbiased = count > 1 ? var * count / (count - 1) : var;
       running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
@@ -150,19 +172,6 @@ __global__ void bn_partials_kernel(const T* __restrict__ x, long long M,
     save_mean[c] = mean;
     save_invstd[c] = rsqrtf(var + eps);
   }
-  if (tid == 0) *counter = 0;
-}
-
-// eval-mode statistics: mean/invstd straight from the running buffers
-__global__ void bn_eval_stats_kernel(int C, float eps,
-                                     const float* __restrict__ running_mean,
-                                     const float* __restrict__ running_var,
-                                     float* __restrict__ save_mean,
-                                     float* __restrict__ save_invstd) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  save_mean[c] = running_mean[c];
-  save_invstd[c] = rsqrtf(running_var[c] + eps);
 }
 
 // normalize + scale/shift, with optional fused residual add (RES) and ELU
@@ -199,19 +208,18 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
   }
 }
 
-// backward partial sums of (g, g*xhat) where g = dy (plain BN) or
-// dy * elu'(y) (forward fused ELU; y is the saved post-activation output).
-// Same block-row + last-block-column-sum structure as the forward; the last
-// block writes the finished sums straight into ws[2C] (gb, gw).
+// elu'(y) from the SAVED OUTPUT y = elu(z): dy/dz = y > 0 ? 1 : y + 1
+__device__ __forceinline__ float elu_bwd_f(float y) {
+  return y > 0.f ? 1.f : y + 1.f;
+}
+
 template <typename T, int VEC, bool ELU>
 __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
                                        const T* __restrict__ gy,
-                                       const T* __restrict__ yout,
-                                       long long M, int Cv,
-                                       const float* __restrict__ mean,
+                                       const T* __restrict__ yout, long long M,
+                                       int Cv, const float* __restrict__ mean,
                                        const float* __restrict__ invstd,
-                                       float* __restrict__ state,
-                                       float* __restrict__ ws /* [2][C] */) {
+                                       float* __restrict__ part /* [nb][2][C] */) {
   using V = VecT<T, VEC>;
   const int C = Cv * VEC;
   const V* xv = reinterpret_cast<const V*>(x);
@@ -264,8 +272,7 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
   }
   const int members = 256 / Cv;
   const int g8 = tid % Cv;
-  int* counter = (int*)state;
-  float* acc_buf = state + 2;
+  float* out = part + (long long)blockIdx.x * 2 * C;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     float* src = pass == 0 ? sdy : sdyx;
@@ -279,22 +286,9 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
 #pragma unroll
         for (int j = 0; j < VEC; ++j) acc[j] += red[(g8 + k * Cv) * VEC + j];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j)
-        unsafeAtomicAdd(&acc_buf[pass * C + c0 + j], acc[j]);
+      for (int j = 0; j < VEC; ++j) out[pass * C + c0 + j] = acc[j];
     }
   }
-
-  __shared__ int is_last;
-  __syncthreads();
-  __threadfence();
-  if (tid == 0) is_last = (atomicAdd(counter, 1) == (int)gridDim.x - 1);
-  __syncthreads();
-  if (!is_last) return;
-  for (int c = tid; c < 2 * C; c += blockDim.x) {
-    ws[c] = acc_buf[c];
-    acc_buf[c] = 0.f;                      // re-zero for the next launch
-  }
-  if (tid == 0) *counter = 0;
 }
 
 // dx = gamma*invstd*(g - mean(g) - xhat*mean(g*xhat)); recomputes
@@ -375,22 +369,25 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
       TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
                   "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                   ") | 256, got C=", C);
-      TORCH_CHECK(C <= kBnCMax, "bn supports C <= ", kBnCMax);
-      int nb = grid_1d(M * C / VEC, 256, 640);
-      bool track = running_mean.defined();
+      int nb = grid_1d(M * C / VEC, 256, 1024);
+      auto part = at::empty({nb, 2, C}, fopts);
       hipLaunchKernelGGL((bn_partials_kernel<scalar_t, VEC>),
                          dim3(nb), dim3(256), 0,
                          stream, (const scalar_t*)x.data_ptr(), M, C / VEC,
-                         bn_sync_state(),
-                         (float)eps, (float)momentum, track,
-                         track ? running_mean.data_ptr<float>() : nullptr,
-                         track ? running_var.data_ptr<float>() : nullptr,
+                         part.data_ptr<float>());
+      hipLaunchKernelGGL(bn_colsum_finalize_kernel, dim3((C + 3) / 4),
+                         dim3(256), 0, stream, part.data_ptr<float>(), nb, C,
+                         M, (float)eps, (float)momentum,
+                         running_mean.defined(),
+                         running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
+                         running_var.defined() ? running_var.data_ptr<float>() : nullptr,
                          save_mean.data_ptr<float>(),
                          save_invstd.data_ptr<float>());
     });
   } else {
-    hipLaunchKernelGGL(bn_eval_stats_kernel, dim3((C + 255) / 256), dim3(256),
-                       0, stream, C, (float)eps,
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                       0, stream, (const float*)nullptr, 0, C, M, (float)eps,
+                       (float)momentum, false, false,
                        running_mean.data_ptr<float>(),
                        running_var.data_ptr<float>(),
                        save_mean.data_ptr<float>(),
@@ -452,13 +449,15 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
   TORCH_CHECK(!want_g || elu, "want_g requires the fused-ELU backward");
   at::Tensor gout;
   if (want_g) gout = at::empty_like(gy);
+  using T0 = std::integral_constant<bool, false>;
+  using T1 = std::integral_constant<bool, true>;
   DISPATCH_F32_BF16(x, "bn_bwd_partials", {
     constexpr int VEC = 16 / sizeof(scalar_t);
     TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
                 "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                 ") | 256, got C=", C);
-    TORCH_CHECK(C <= kBnCMax, "bn supports C <= ", kBnCMax);
-    int nb = grid_1d(M * C / VEC, 256, 640);
+    int nb = grid_1d(M * C / VEC, 256, 1024);
+    auto part = at::empty({nb, 2, C}, fopts);
     auto launch1 = [&](auto ekind) {
       hipLaunchKernelGGL((bn_bwd_partials_kernel<scalar_t, VEC,
                                                  decltype(ekind)::value>),
@@ -467,9 +466,15 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                          (const scalar_t*)gy.data_ptr(),
                          (const scalar_t*)y_ptr, M, C / VEC,
                          save_mean.data_ptr<float>(),
-                         save_invstd.data_ptr<float>(),
-                         bn_sync_state(), ws.data_ptr<float>());
+                         save_invstd.data_ptr<float>(), part.data_ptr<float>());
     };
+    if (elu) launch1(T1{}); else launch1(T0{});
+    hipLaunchKernelGGL(bn_colsum_kernel, dim3((2 * C + 3) / 4), dim3(256),
+                       0, stream, part.data_ptr<float>(), nb, 2 * C,
+                       ws.data_ptr<float>());
+  });
+  DISPATCH_F32_BF16(x, "bn_bwd_apply", {
+    constexpr int VEC = 16 / sizeof(scalar_t);
     long long nvec = M * C / VEC;
     auto launch2 = [&](auto ekind, auto wkind) {
       hipLaunchKernelGGL((bn_bwd_apply_kernel<scalar_t, VEC,
@@ -485,9 +490,6 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                          gamma_f.data_ptr<float>(), ws.data_ptr<float>(), M,
                          nvec, C / VEC);
     };
-    using T0 = std::integral_constant<bool, false>;
-    using T1 = std::integral_constant<bool, true>;
-    if (elu) launch1(T1{}); else launch1(T0{});
     if (elu && want_g)  launch2(T1{}, T1{});
     else if (elu)       launch2(T1{}, T0{});
     else                launch2(T0{}, T0{});
