@@ -49,7 +49,13 @@ __device__ __forceinline__ int lds_off(int row, int k) {
   return row * 128 + chunk * 16 + (k & 7) * 2;
 }
 
-template <int BM, int STRIDE>
+// REGSTAGE: 0 = LDS-DMA (glds) staging, 1 = register staging (global
+// vector loads into VGPRs + ds_write_b128 committed right before the
+// barrier).  A glds instruction's ISSUE cost among MFMAs is ~60-185 cycles
+// (guide §LDS-DMA); at this kernel's small tiles (6 pieces per 16 MFMAs)
+// that is comparable to the MFMA work itself, so the cheap-issue register
+// path is worth an A/B (FEDKIT_CONV_REGSTAGE=0/1, default measured best).
+template <int BM, int STRIDE, int REGSTAGE>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      const bf16* __restrict__ w,   // [Kout][R*S*C]
@@ -115,7 +121,8 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   auto bufA = [&](int b) -> char* { return smem + b * (AB + BB); };
   auto bufB = [&](int b) -> char* { return smem + b * (AB + BB) + AB; };
 
-  auto stage = [&](int buf, int kt) {
+  bf16x8 areg[2][A_SLOTS], brg[2][B_SLOTS];
+  auto stage = [&](int buf, int kt, int rset) {
     // A tile: per-slot source (r,s,c) from k_global; 16-B LDS-DMA.
     // LDS dest for a glds is wave-uniform base + lane*16: slot d = pass*256
     // + wave*64 + lane matches d = pass*256 + tid exactly.
@@ -127,15 +134,29 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
       int s = rs % S;
       int r = rs / S;
       const bf16* src = xp + a_rowbase[i] + ((long long)r * Wp + s) * C + c;
-      char* dst = bufA(buf) + (i * 4 + wave) * 1024;
-      GLDS16(src, dst);
+      if (REGSTAGE)
+        __builtin_memcpy(&areg[rset][i], src, sizeof(bf16x8));
+      else
+        GLDS16(src, bufA(buf) + (i * 4 + wave) * 1024);
     }
 #pragma unroll
     for (int i = 0; i < B_SLOTS; ++i) {
       const bf16* src = w + b_rowbase[i] + kt * BK + b_k8[i] * 8;
-      char* dst = bufB(buf) + (i * 4 + wave) * 1024;
-      GLDS16(src, dst);
+      if (REGSTAGE)
+        __builtin_memcpy(&brg[rset][i], src, sizeof(bf16x8));
+      else
+        GLDS16(src, bufB(buf) + (i * 4 + wave) * 1024);
     }
+  };
+  auto commit = [&](int buf, int rset) {
+#pragma unroll
+    for (int i = 0; i < A_SLOTS; ++i)
+      *(bf16x8*)(bufA(buf) + (i * 4 + wave) * 1024 + lane * 16) =
+          areg[rset][i];
+#pragma unroll
+    for (int i = 0; i < B_SLOTS; ++i)
+      *(bf16x8*)(bufB(buf) + (i * 4 + wave) * 1024 + lane * 16) =
+          brg[rset][i];
   };
 
   // ---- wave -> output sub-tile: 2x2 waves, wave tile (BM/2) x 32
@@ -158,16 +179,20 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   // single raw s_barrier does double duty: every wave's tile-kt loads have
   // landed, and every wave is done reading buffer (kt+2)%3 (used by kt-1).
   constexpr int LPS = A_SLOTS + B_SLOTS;
-  stage(0, 0);
-  if (nkt > 1) stage(1, 1);
+  stage(0, 0, 0);
+  if (nkt > 1) stage(1, 1, 1);
 
   for (int kt = 0; kt < nkt; ++kt) {
     if (kt + 1 < nkt)
       asm volatile("s_waitcnt vmcnt(%0)" ::"n"(LPS) : "memory");
     else
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (REGSTAGE) {
+      commit(kt % 3, kt % 2);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    }
     __builtin_amdgcn_s_barrier();
-    if (kt + 2 < nkt) stage((kt + 2) % 3, kt + 2);
+    if (kt + 2 < nkt) stage((kt + 2) % 3, kt + 2, kt % 2);
     const char* A = bufA(kt % 3);
     const char* B = bufB(kt % 3);
 #pragma unroll
@@ -625,25 +650,27 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
   bool bm64 = ((M + 127) / 128) * (Kout / BN) < 256;
   int BM = bm64 ? 64 : 128;
   dim3 grid((unsigned)((M + BM - 1) / BM), Kout / BN);
-  if (stride == 1) {
-    if (bm64)
-      hipLaunchKernelGGL((conv_fwd_kernel<64, 1>), grid, dim3(256), 0, stream,
-                         (const bf16*)xp.data_ptr(), (const bf16*)w_krs_c.data_ptr(),
-                         (bf16*)y.data_ptr(), N, Hp, Wp, C, Kout, R, S, P, Q, Kg);
+  static const int regstage = []() {
+    const char* e = getenv("FEDKIT_CONV_REGSTAGE");
+    return e ? atoi(e) : 0;
+  }();
+  auto L = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
+                       (const bf16*)xp.data_ptr(),
+                       (const bf16*)w_krs_c.data_ptr(), (bf16*)y.data_ptr(),
+                       N, Hp, Wp, C, Kout, R, S, P, Q, Kg);
+  };
+  TORCH_CHECK(stride == 1 || stride == 2, "conv kernel supports stride 1/2");
+  if (regstage) {
+    if (stride == 1)
+      bm64 ? L(conv_fwd_kernel<64, 1, 1>) : L(conv_fwd_kernel<128, 1, 1>);
     else
-      hipLaunchKernelGGL((conv_fwd_kernel<128, 1>), grid, dim3(256), 0, stream,
-                         (const bf16*)xp.data_ptr(), (const bf16*)w_krs_c.data_ptr(),
-                         (bf16*)y.data_ptr(), N, Hp, Wp, C, Kout, R, S, P, Q, Kg);
+      bm64 ? L(conv_fwd_kernel<64, 2, 1>) : L(conv_fwd_kernel<128, 2, 1>);
   } else {
-    TORCH_CHECK(stride == 2, "conv kernel supports stride 1/2");
-    if (bm64)
-      hipLaunchKernelGGL((conv_fwd_kernel<64, 2>), grid, dim3(256), 0, stream,
-                         (const bf16*)xp.data_ptr(), (const bf16*)w_krs_c.data_ptr(),
-                         (bf16*)y.data_ptr(), N, Hp, Wp, C, Kout, R, S, P, Q, Kg);
+    if (stride == 1)
+      bm64 ? L(conv_fwd_kernel<64, 1, 0>) : L(conv_fwd_kernel<128, 1, 0>);
     else
-      hipLaunchKernelGGL((conv_fwd_kernel<128, 2>), grid, dim3(256), 0, stream,
-                         (const bf16*)xp.data_ptr(), (const bf16*)w_krs_c.data_ptr(),
-                         (bf16*)y.data_ptr(), N, Hp, Wp, C, Kout, R, S, P, Q, Kg);
+      bm64 ? L(conv_fwd_kernel<64, 2, 0>) : L(conv_fwd_kernel<128, 2, 0>);
   }
   return y;
 }
