@@ -124,8 +124,12 @@ def bench_conv_bwd(iters):
         t_dx = timeit(fed_dx, iters)
         t_dw = timeit(fed_dw, iters)
         t_e = timeit(eager_both, iters)
+        _, gw_ref = eager_both()
+        dwerr = ((fed_dw().float() - gw_ref).abs().max()
+                 / gw_ref.abs().max().clamp_min(1e-6)).item()
         name = f"{ks}x{ks} C{Cin} H{H} K{Kout} s{stride} x{count}"
-        print(f"{name:38s} {t_dx:8.1f} {t_dw:8.1f} {t_e:8.1f} (dx+dw fp32)")
+        print(f"{name:38s} {t_dx:8.1f} {t_dw:8.1f} {t_e:8.1f} (dx+dw fp32)"
+              + (f"  DWERR {dwerr:.3f}" if dwerr > 0.02 else ""))
 
 
 def bench_bn(iters):
