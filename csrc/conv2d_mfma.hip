@@ -59,9 +59,12 @@ template <int BM, int STRIDE, int REGSTAGE>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      const bf16* __restrict__ w,   // [Kout][R*S*C]
-                     bf16* __restrict__ y,         // [M][Kout_total]
+                     bf16* __restrict__ y,         // [M][Ktrue]
                      int N, int Hp, int Wp, int C, int Kout,
-                     int R, int S, int P, int Q, int Kg) {
+                     int R, int S, int P, int Q, int Kg, int dil,
+                     int Ktrue /* <= Kout: K rows beyond are zero-padding
+                                  (VAE/CPC channel counts), skipped on
+                                  store so y needs no unpad pass */) {
   constexpr int AB = BM * BK * 2;          // A tile bytes
   constexpr int BB = BN * BK * 2;          // B tile bytes
   __shared__ char smem[3 * (AB + BB)];     // 3-stage pipeline
@@ -133,7 +136,8 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
       int rs = kg / C;
       int s = rs % S;
       int r = rs / S;
-      const bf16* src = xp + a_rowbase[i] + ((long long)r * Wp + s) * C + c;
+      const bf16* src =
+          xp + a_rowbase[i] + ((long long)r * dil * Wp + s * dil) * C + c;
       if (REGSTAGE)
         __builtin_memcpy(&areg[rset][i], src, sizeof(bf16x8));
       else
@@ -215,7 +219,8 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
     }
   }
 
-  // ---- epilogue: D fragment lane l holds col = l&15, rows (l>>4)*4 + v
+  // ---- epilogue: D fragment lane l holds col = l&15, rows (l>>4)*4 + v;
+  // stores go at Ktrue stride and skip the zero-padded out channels
   const int col = bn * BN + wn + frag_row;
 #pragma unroll
   for (int mfrag = 0; mfrag < MR; ++mfrag) {
@@ -225,8 +230,9 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
       if (m < M) {
 #pragma unroll
         for (int nfrag = 0; nfrag < NR; ++nfrag)
-          y[m * Kout + col + nfrag * 16] =
-              __float2bfloat16(acc[mfrag][nfrag][v]);
+          if (col + nfrag * 16 < Ktrue)
+            y[m * Ktrue + col + nfrag * 16] =
+                __float2bfloat16(acc[mfrag][nfrag][v]);
       }
     }
   }
@@ -579,7 +585,8 @@ __global__ void rot_weight_kernel(const T* __restrict__ w, T* __restrict__ wr,
 template <int VEC>
 __global__ void im2col_kernel(const bf16* __restrict__ xp,
                               bf16* __restrict__ col, int N, int Hp, int Wp,
-                              int C, int R, int S, int P, int Q, int stride) {
+                              int C, int R, int S, int P, int Q, int stride,
+                              int dil) {
   using V = __attribute__((ext_vector_type(VEC))) __bf16;
   const V* xv = reinterpret_cast<const V*>(xp);
   V* cv = reinterpret_cast<V*>(col);
@@ -597,8 +604,8 @@ __global__ void im2col_kernel(const bf16* __restrict__ xp,
     int q = (int)(m % Q);
     int p = (int)((m / Q) % P);
     int n = (int)(m / ((long long)P * Q));
-    cv[i] = xv[((((long long)n * Hp + p * stride + r) * Wp +
-                 (q * stride + s)) * C) / VEC + c8];
+    cv[i] = xv[((((long long)n * Hp + p * stride + r * dil) * Wp +
+                 (q * stride + s * dil)) * C) / VEC + c8];
   }
 }
 
@@ -643,17 +650,20 @@ void check_conv_inputs(const at::Tensor& x, const at::Tensor& w) {
               "conv weight must be channels_last");
 }
 
-// launch on the PRE-PADDED input; pad already folded into Hp/Wp
+// launch on the PRE-PADDED input; pad already folded into Hp/Wp.
+// Ktrue < Kout means the trailing weight rows are channel zero-padding
+// (VAE/CPC shapes): y is allocated and stored dense at Ktrue channels.
 at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
-                     int stride, int P, int Q) {
+                     int stride, int P, int Q, int dil = 1, int Ktrue = -1) {
   int N = xp.size(0), C = xp.size(1), Hp = xp.size(2), Wp = xp.size(3);
   int Kout = w_krs_c.size(0), R = w_krs_c.size(2), S = w_krs_c.size(3);
   int Kg = R * S * C;
+  if (Ktrue < 0) Ktrue = Kout;
   TORCH_CHECK(C % 8 == 0, "conv kernel needs C % 8 == 0, got ", C);
   TORCH_CHECK(Kg % BK == 0, "conv kernel needs R*S*C % 64 == 0, got ", Kg);
   TORCH_CHECK(Kout % BN == 0, "conv kernel needs Kout % 64 == 0, got ", Kout);
   long long M = (long long)N * P * Q;
-  auto y = at::empty({N, Kout, P, Q},
+  auto y = at::empty({N, Ktrue, P, Q},
                      xp.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto stream = fedkit_stream();
   // pick BM so the grid fills the 256 CUs when possible
@@ -668,7 +678,7 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
                        (const bf16*)xp.data_ptr(),
                        (const bf16*)w_krs_c.data_ptr(), (bf16*)y.data_ptr(),
-                       N, Hp, Wp, C, Kout, R, S, P, Q, Kg);
+                       N, Hp, Wp, C, Kout, R, S, P, Q, Kg, dil, Ktrue);
   };
   TORCH_CHECK(stride == 1 || stride == 2, "conv kernel supports stride 1/2");
   if (regstage) {
@@ -690,6 +700,15 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
 at::Tensor fedkit_conv_small_fwd(const at::Tensor& x, const at::Tensor& w,
                                  long stride, long padding);  // conv_small.hip
 
+// public dilate+pad (transposed-conv input dilation):
+// out[n][h*str+pt][w*str+pl][c] = x[n][h][w][c], everything else zero
+at::Tensor fedkit_dilate_pad(const at::Tensor& x, long pt, long pb, long pl,
+                             long pr, long str) {
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "dilate_pad expects channels_last");
+  return dilate_pad_nhwc(x, (int)pt, (int)pb, (int)pl, (int)pr, (int)str);
+}
+
 at::Tensor fedkit_conv2d_pad_input(const at::Tensor& x, long padding) {
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
               "pad_input expects channels_last");
@@ -698,39 +717,59 @@ at::Tensor fedkit_conv2d_pad_input(const at::Tensor& x, long padding) {
 }
 
 // forward on a PRE-PADDED input (the autograd wrapper saves xp so backward
-// never re-pads): P = (Hp - R)/stride + 1
+// never re-pads): P = (Hp - Reff)/stride + 1 with Reff = (R-1)*dil + 1.
+// ktrue < Kout marks trailing zero-padded out channels (generic VAE/CPC
+// path); y comes back dense at ktrue channels.
 at::Tensor fedkit_conv2d_fwd_prepadded(const at::Tensor& xp,
-                                       const at::Tensor& w, long stride) {
+                                       const at::Tensor& w, long stride,
+                                       long dil, long ktrue) {
   check_conv_inputs(xp, w);
   int Hp = xp.size(2), Wp = xp.size(3);
   int R = w.size(2), S = w.size(3);
-  int P = (Hp - R) / (int)stride + 1;
-  int Q = (Wp - S) / (int)stride + 1;
-  return conv_core(xp, w, (int)stride, P, Q);
+  int Reff = (R - 1) * (int)dil + 1, Seff = (S - 1) * (int)dil + 1;
+  int P = (Hp - Reff) / (int)stride + 1;
+  int Q = (Wp - Seff) / (int)stride + 1;
+  return conv_core(xp, w, (int)stride, P, Q, (int)dil, (int)ktrue);
 }
 
 at::Tensor fedkit_conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
-                             long stride, long padding) {
+                             long stride, long padding, long dil,
+                             long ktrue) {
   check_conv_inputs(x, w);
   int C = x.size(1), H = x.size(2), W = x.size(3);
   int R = w.size(2);
-  if (C % 8 != 0)   // e.g. ResNet conv1 (C=3): direct small-C kernel
+  if (C % 8 != 0) {  // e.g. ResNet conv1 (C=3): direct small-C kernel
+    TORCH_CHECK(dil == 1 && ktrue < 0,
+                "small-C conv path has no dilation/ktrue support");
     return fedkit_conv_small_fwd(x, w, stride, padding);
-  int P = (H + 2 * (int)padding - R) / (int)stride + 1;
-  int Q = (W + 2 * (int)padding - R) / (int)stride + 1;
+  }
+  int Reff = (R - 1) * (int)dil + 1;
+  int P = (H + 2 * (int)padding - Reff) / (int)stride + 1;
+  int Q = (W + 2 * (int)padding - Reff) / (int)stride + 1;
   at::Tensor xp = padding > 0 ? pad_nhwc(x, padding, padding, padding, padding)
                               : x;
-  return conv_core(xp, w, (int)stride, P, Q);
+  return conv_core(xp, w, (int)stride, P, Q, (int)dil, (int)ktrue);
 }
 
+// dx = conv_dil(dilate_str(gy), rot180(w)); gy must carry the same
+// (possibly zero-padded) K channels as w; ctrue < C slices the channel
+// padding off dx at the store.
 at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
-                                  long stride, long padding, long H, long W) {
+                                  long stride, long padding, long H, long W,
+                                  long dil, long ctrue) {
   check_conv_inputs(gy, w);
   int K = w.size(0), C = w.size(1), R = w.size(2), S = w.size(3);
   TORCH_CHECK(K % 8 == 0, "bwd-data needs Kout % 8 == 0");
-  // rotate weights: wrot[c][r][s][k]
-  auto wrot = at::empty({C, K, R, S},
-                        w.options().memory_format(at::MemoryFormat::ChannelsLast));
+  // rotate weights: wrot[c][r][s][k].  The dx GEMM's out-channel count is
+  // C, which the generic (VAE/CPC) path only pads to a multiple of 8 —
+  // round the wrot rows up to the 64-row tile with zeros and store dx
+  // dense at ctrue via the kernel's Ktrue guard.
+  int Cpad = (C + 63) / 64 * 64;
+  auto wrot = Cpad == C
+      ? at::empty({C, K, R, S},
+                  w.options().memory_format(at::MemoryFormat::ChannelsLast))
+      : at::zeros({Cpad, K, R, S},
+                  w.options().memory_format(at::MemoryFormat::ChannelsLast));
   {
     auto stream = fedkit_stream();
     long long total = (long long)K * R * S * C;
@@ -744,18 +783,21 @@ at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
                          dim3(256), 0, stream, (const bf16*)w.data_ptr(),
                          (bf16*)wrot.data_ptr(), K, R, S, C);
   }
-  // dilate+pad gy: pl = R-1-pad, pr = pl + a with a = (H + 2p - R) % stride
-  int pl = R - 1 - (int)padding;
-  int a = (int)((H + 2 * padding - R) % stride);
+  if (ctrue < 0) ctrue = C;
+  // dilate+pad gy: pl = (R-1)*dil - pad, pr = pl + a with
+  // a = (H + 2p - Reff) % stride, Reff = (R-1)*dil + 1
+  int Reff = (R - 1) * (int)dil + 1;
+  int pl = (Reff - 1) - (int)padding;
+  int a = (int)((H + 2 * padding - Reff) % stride);
   at::Tensor gyp = dilate_pad_nhwc(gy, pl, pl + a, pl, pl + a, (int)stride);
-  return conv_core(gyp, wrot, 1, (int)H, (int)W);
+  return conv_core(gyp, wrot, 1, (int)H, (int)W, (int)dil, (int)ctrue);
 }
 
 // bwd-weight from the PRE-PADDED input saved by the forward
 at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
                                               const at::Tensor& xp,
                                               long stride, long R_in,
-                                              long S_in) {
+                                              long S_in, long dil) {
   check_conv_inputs(gy, xp);
   int N = xp.size(0), C = xp.size(1);
   int K = gy.size(1), P = gy.size(2), Q = gy.size(3);
@@ -770,8 +812,8 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
   int Hp = xp.size(2), Wp = xp.size(3);
   long long NHW = (long long)N * Hp * Wp;
   bool pow2 = P > 0 && Q > 0 && (P & (P - 1)) == 0 && (Q & (Q - 1)) == 0;
-  if (stride == 1 && pow2 && Q % 8 == 0 && K % 64 == 0 && RSC % 64 == 0 &&
-      C % 64 == 0 && M % 64 == 0 && NHW % 64 == 0) {
+  if (stride == 1 && dil == 1 && pow2 && Q % 8 == 0 && K % 64 == 0 &&
+      RSC % 64 == 0 && C % 64 == 0 && M % 64 == 0 && NHW % 64 == 0) {
     auto stream = fedkit_stream();
     auto dyT = at::empty({(long long)K, M}, gy.options());
     auto xpT = at::empty({(long long)C, NHW}, xp.options());
@@ -842,13 +884,13 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
       hipLaunchKernelGGL(im2col_kernel<8>, dim3(grid_1d(total, 512)),
                          dim3(256), 0, stream, (const bf16*)xp.data_ptr(),
                          (bf16*)col.data_ptr(), N, xp.size(2), xp.size(3), C,
-                         R, S, P, Q, (int)stride);
+                         R, S, P, Q, (int)stride, (int)dil);
     } else {
       long long total = M * R * S * C;
       hipLaunchKernelGGL(im2col_kernel<1>, dim3(grid_1d(total, 512)),
                          dim3(256), 0, stream, (const bf16*)xp.data_ptr(),
                          (bf16*)col.data_ptr(), N, xp.size(2), xp.size(3), C,
-                         R, S, P, Q, (int)stride);
+                         R, S, P, Q, (int)stride, (int)dil);
     }
   }
   // dw = dy^T @ col is a tall-skinny reduction GEMM ([K<=512, RSC<=4608]
@@ -875,8 +917,8 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
 
 at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
                                     long stride, long padding, long R_in,
-                                    long S_in) {
+                                    long S_in, long dil) {
   at::Tensor xp = padding > 0 ? pad_nhwc(x, padding, padding, padding, padding)
                               : x;
-  return fedkit_conv2d_bwd_weight_prepadded(gy, xp, stride, R_in, S_in);
+  return fedkit_conv2d_bwd_weight_prepadded(gy, xp, stride, R_in, S_in, dil);
 }

@@ -36,17 +36,23 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       bool want_g);
 
 at::Tensor fedkit_conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
-                             long stride, long padding);
+                             long stride, long padding, long dil, long ktrue);
 at::Tensor fedkit_conv2d_pad_input(const at::Tensor& x, long padding);
+at::Tensor fedkit_dilate_pad(const at::Tensor& x, long pt, long pb, long pl,
+                             long pr, long str);
 at::Tensor fedkit_conv2d_fwd_prepadded(const at::Tensor& xp,
-                                       const at::Tensor& w, long stride);
+                                       const at::Tensor& w, long stride,
+                                       long dil, long ktrue);
 at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
                                               const at::Tensor& xp,
-                                              long stride, long R, long S);
+                                              long stride, long R, long S,
+                                              long dil);
 at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
-                                  long stride, long padding, long H, long W);
+                                  long stride, long padding, long H, long W,
+                                  long dil, long ctrue);
 at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
-                                    long stride, long padding, long R, long S);
+                                    long stride, long padding, long R, long S,
+                                    long dil);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "fedkit hand-written CDNA4 (gfx950 / MI355X) kernels";
@@ -73,12 +79,26 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("save_invstd"), py::arg("elu_y") = c10::nullopt,
         py::arg("want_g") = false);
   m.def("conv2d_fwd", &fedkit_conv2d_fwd,
-        "NHWC implicit-GEMM conv fwd on MFMA (3x3/1x1, stride 1/2)");
+        "NHWC implicit-GEMM conv fwd on MFMA (square filter, stride 1/2, "
+        "optional dilation; ktrue marks channel-padded Kout)",
+        py::arg("x"), py::arg("w"), py::arg("stride"), py::arg("padding"),
+        py::arg("dil") = 1, py::arg("ktrue") = -1);
   m.def("conv2d_pad_input", &fedkit_conv2d_pad_input, "zero-pad NHWC input");
+  m.def("dilate_pad", &fedkit_dilate_pad,
+        "NHWC zero-insert dilation + border pad (transposed-conv input)");
   m.def("conv2d_fwd_prepadded", &fedkit_conv2d_fwd_prepadded,
-        "conv fwd on a pre-padded input");
+        "conv fwd on a pre-padded input",
+        py::arg("xp"), py::arg("w"), py::arg("stride"),
+        py::arg("dil") = 1, py::arg("ktrue") = -1);
   m.def("conv2d_bwd_weight_prepadded", &fedkit_conv2d_bwd_weight_prepadded,
-        "conv bwd-weight from the saved padded input");
-  m.def("conv2d_bwd_data", &fedkit_conv2d_bwd_data, "conv bwd-data");
-  m.def("conv2d_bwd_weight", &fedkit_conv2d_bwd_weight, "conv bwd-weight");
+        "conv bwd-weight from the saved padded input",
+        py::arg("gy"), py::arg("xp"), py::arg("stride"), py::arg("R"),
+        py::arg("S"), py::arg("dil") = 1);
+  m.def("conv2d_bwd_data", &fedkit_conv2d_bwd_data, "conv bwd-data",
+        py::arg("gy"), py::arg("w"), py::arg("stride"), py::arg("padding"),
+        py::arg("H"), py::arg("W"), py::arg("dil") = 1,
+        py::arg("ctrue") = -1);
+  m.def("conv2d_bwd_weight", &fedkit_conv2d_bwd_weight, "conv bwd-weight",
+        py::arg("gy"), py::arg("x"), py::arg("stride"), py::arg("padding"),
+        py::arg("R"), py::arg("S"), py::arg("dil") = 1);
 }

@@ -10,6 +10,8 @@ Architecture parity with reference src/simple_models.py:436-514:
 
 import torch
 import torch.nn as nn
+
+from ..ops.conv import FedConvGeneric
 import torch.nn.functional as F
 
 from ..ops.elu import elu
@@ -20,14 +22,14 @@ class EncoderCNN(nn.Module):
         super().__init__()
         self.latent_dim = latent_dim
         # dilated 4x4 stride-2 bank, all 8->8 channels, 32x32 -> 16x16
-        self.conv1_1 = nn.Conv2d(8, 8, 4, stride=2, dilation=1, padding=1)
-        self.conv1_2 = nn.Conv2d(8, 8, 4, stride=2, dilation=2, padding=3)
-        self.conv1_4 = nn.Conv2d(8, 8, 4, stride=2, dilation=4, padding=6)
-        self.conv1_8 = nn.Conv2d(8, 8, 4, stride=2, dilation=8, padding=12)
-        self.conv1_16 = nn.Conv2d(8, 8, 4, stride=2, dilation=16, padding=24)
-        self.conv2 = nn.Conv2d(8 * 5, latent_dim // 4, 4, stride=2, padding=1)   # 16 -> 8
-        self.conv3 = nn.Conv2d(latent_dim // 4, latent_dim // 2, 4, stride=2, padding=1)  # 8 -> 4
-        self.conv4 = nn.Conv2d(latent_dim // 2, latent_dim, 4, stride=2, padding=1)       # 4 -> 2
+        self.conv1_1 = FedConvGeneric(8, 8, 4, stride=2, dilation=1, padding=1)
+        self.conv1_2 = FedConvGeneric(8, 8, 4, stride=2, dilation=2, padding=3)
+        self.conv1_4 = FedConvGeneric(8, 8, 4, stride=2, dilation=4, padding=6)
+        self.conv1_8 = FedConvGeneric(8, 8, 4, stride=2, dilation=8, padding=12)
+        self.conv1_16 = FedConvGeneric(8, 8, 4, stride=2, dilation=16, padding=24)
+        self.conv2 = FedConvGeneric(8 * 5, latent_dim // 4, 4, stride=2, padding=1)   # 16 -> 8
+        self.conv3 = FedConvGeneric(latent_dim // 4, latent_dim // 2, 4, stride=2, padding=1)  # 8 -> 4
+        self.conv4 = FedConvGeneric(latent_dim // 2, latent_dim, 4, stride=2, padding=1)       # 4 -> 2
 
     def forward(self, x):
         bank = torch.cat(
@@ -49,10 +51,10 @@ class ContextgenCNN(nn.Module):
     def __init__(self, latent_dim=1024):
         super().__init__()
         self.latent_dim = latent_dim
-        self.conv1 = nn.Conv2d(latent_dim, latent_dim // 4, 1, stride=1, padding=0, bias=False)
-        self.conv2 = nn.Conv2d(latent_dim // 4, latent_dim // 4, 2, stride=1, padding=1, bias=False)
-        self.conv3 = nn.Conv2d(latent_dim // 4, latent_dim // 2, 2, stride=1, padding=0, bias=False)
-        self.conv4 = nn.Conv2d(latent_dim // 2, latent_dim, 1, stride=1, padding=0, bias=False)
+        self.conv1 = FedConvGeneric(latent_dim, latent_dim // 4, 1, stride=1, padding=0, bias=False)
+        self.conv2 = FedConvGeneric(latent_dim // 4, latent_dim // 4, 2, stride=1, padding=1, bias=False)
+        self.conv3 = FedConvGeneric(latent_dim // 4, latent_dim // 2, 2, stride=1, padding=0, bias=False)
+        self.conv4 = FedConvGeneric(latent_dim // 2, latent_dim, 1, stride=1, padding=0, bias=False)
 
     def forward(self, x):
         x = elu(self.conv1(x))
@@ -72,8 +74,8 @@ class PredictorCNN(nn.Module):
         super().__init__()
         self.latent_dim = latent_dim
         self.reduced_dim = reduced_dim
-        self.conv1 = nn.Conv2d(latent_dim, reduced_dim, 1, bias=False)
-        self.conv2 = nn.Conv2d(latent_dim, reduced_dim, 1, bias=False)
+        self.conv1 = FedConvGeneric(latent_dim, reduced_dim, 1, bias=False)
+        self.conv2 = FedConvGeneric(latent_dim, reduced_dim, 1, bias=False)
 
     def forward(self, latents, context):
         return self.conv1(latents), self.conv2(context)

@@ -20,6 +20,8 @@ reparametrization is never actually disabled (simple_models.py:344-345).
 
 import torch
 import torch.nn as nn
+
+from ..ops.conv import FedConvGeneric, FedConvTranspose2d
 import torch.nn.functional as F
 
 from ..ops.elu import elu
@@ -31,18 +33,18 @@ class AutoEncoderCNN(nn.Module):
     def __init__(self):
         super().__init__()
         self.latent_dim = 10
-        self.conv1 = nn.Conv2d(3, 12, 4, stride=2, padding=1)    # 32 -> 16
-        self.conv2 = nn.Conv2d(12, 24, 4, stride=2, padding=1)   # 16 -> 8
-        self.conv3 = nn.Conv2d(24, 48, 4, stride=2, padding=1)   # 8 -> 4
-        self.conv4 = nn.Conv2d(48, 96, 4, stride=2, padding=1)   # 4 -> 2
+        self.conv1 = FedConvGeneric(3, 12, 4, stride=2, padding=1)    # 32 -> 16
+        self.conv2 = FedConvGeneric(12, 24, 4, stride=2, padding=1)   # 16 -> 8
+        self.conv3 = FedConvGeneric(24, 48, 4, stride=2, padding=1)   # 8 -> 4
+        self.conv4 = FedConvGeneric(48, 96, 4, stride=2, padding=1)   # 4 -> 2
         self.fc1 = nn.Linear(384, 16)
         self.fc21 = nn.Linear(16, self.latent_dim)
         self.fc22 = nn.Linear(16, self.latent_dim)
         self.fc3 = nn.Linear(self.latent_dim, 384)
-        self.tconv1 = nn.ConvTranspose2d(96, 48, 4, stride=2, padding=1)
-        self.tconv2 = nn.ConvTranspose2d(48, 24, 4, stride=2, padding=1)
-        self.tconv3 = nn.ConvTranspose2d(24, 12, 4, stride=2, padding=1)
-        self.tconv4 = nn.ConvTranspose2d(12, 3, 4, stride=2, padding=1)
+        self.tconv1 = FedConvTranspose2d(96, 48, 4, stride=2, padding=1)
+        self.tconv2 = FedConvTranspose2d(48, 24, 4, stride=2, padding=1)
+        self.tconv3 = FedConvTranspose2d(24, 12, 4, stride=2, padding=1)
+        self.tconv4 = FedConvTranspose2d(12, 3, 4, stride=2, padding=1)
 
     def encode(self, x):
         x = elu(self.conv1(x))
@@ -89,10 +91,10 @@ class AutoEncoderCNNCL(nn.Module):
         self.K = K   # clusters
         self.L = L   # latent dimension
         self.repr_flag = True
-        self.conv1 = nn.Conv2d(3, 12, 4, stride=2, padding=1)
-        self.conv2 = nn.Conv2d(12, 24, 4, stride=2, padding=1)
-        self.conv3 = nn.Conv2d(24, 48, 4, stride=2, padding=1)
-        self.conv4 = nn.Conv2d(48, 96, 4, stride=2, padding=1)
+        self.conv1 = FedConvGeneric(3, 12, 4, stride=2, padding=1)
+        self.conv2 = FedConvGeneric(12, 24, 4, stride=2, padding=1)
+        self.conv3 = FedConvGeneric(24, 48, 4, stride=2, padding=1)
+        self.conv4 = FedConvGeneric(48, 96, 4, stride=2, padding=1)
         # cluster head q(k|x)
         self.fc11 = nn.Linear(384, 128)
         self.fc12 = nn.Linear(128, 64)
@@ -109,11 +111,11 @@ class AutoEncoderCNNCL(nn.Module):
         self.fc17 = nn.Linear(64, self.L)
         # decoder p(x|z)
         self.fc25 = nn.Linear(self.L, 384)
-        self.tconv1 = nn.ConvTranspose2d(96, 48, 4, stride=2, padding=1)
-        self.tconv2 = nn.ConvTranspose2d(48, 24, 4, stride=2, padding=1)
-        self.tconv3 = nn.ConvTranspose2d(24, 12, 4, stride=2, padding=1)
-        self.tconv4 = nn.ConvTranspose2d(12, 3, 4, stride=2, padding=1)
-        self.tconv5 = nn.ConvTranspose2d(12, 3, 4, stride=2, padding=1)
+        self.tconv1 = FedConvTranspose2d(96, 48, 4, stride=2, padding=1)
+        self.tconv2 = FedConvTranspose2d(48, 24, 4, stride=2, padding=1)
+        self.tconv3 = FedConvTranspose2d(24, 12, 4, stride=2, padding=1)
+        self.tconv4 = FedConvTranspose2d(12, 3, 4, stride=2, padding=1)
+        self.tconv5 = FedConvTranspose2d(12, 3, 4, stride=2, padding=1)
 
     def enable_repr(self):
         self.repr_flag = True

@@ -73,10 +73,12 @@ def native_enabled(t: torch.Tensor) -> bool:
 
 
 from . import elu, flat, losses          # noqa: E402,F401
-from .conv import FedConv2d              # noqa: E402
+from .conv import (FedConv2d, FedConvGeneric,  # noqa: E402
+                   FedConvTranspose2d)
 from .norm import FedBatchNorm2d         # noqa: E402
 
 __all__ = [
     "ext", "has_ext", "require_ext", "native_enabled",
-    "elu", "flat", "losses", "FedConv2d", "FedBatchNorm2d",
+    "elu", "flat", "losses", "FedConv2d", "FedConvGeneric",
+    "FedConvTranspose2d", "FedBatchNorm2d",
 ]

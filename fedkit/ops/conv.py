@@ -68,6 +68,154 @@ class _ConvFn(torch.autograd.Function):
         return gx, gw, None, None
 
 
+def _pad_mult(n, m):
+    return (n + m - 1) // m * m
+
+
+class _GenConvFn(torch.autograd.Function):
+    """Generic conv on the MFMA kernels: any in/out channel count via zero
+    channel padding (in -> mult of 8, out -> mult of 64, with the kernel
+    storing dense at the true K), square filter, stride 1/2, dilation.
+
+    Covers the VAE encoder/decoder 4x4-s2 convs, the CPC dilated conv bank
+    (d in 1..16) and the CPC 2x2/1x1 convs (SURVEY.md §2a rows 3, 6;
+    BASELINE north star: "the VAE/CPC encoder convs — hand-written CDNA4").
+    Inputs arrive already channel-padded (xpch [N][C8][H][W] bf16 NHWC,
+    wp [K64][C8][R][S]); the callers pad via F.pad so autograd slices the
+    gradients back down.
+    """
+
+    @staticmethod
+    def forward(ctx, xpch, wp, stride, padding, dil, ktrue):
+        ext = _ext()
+        xp = ext.conv2d_pad_input(xpch, padding)
+        y = ext.conv2d_fwd_prepadded(xp, wp, stride, dil=dil, ktrue=ktrue)
+        ctx.save_for_backward(xp, wp)
+        ctx.meta = (stride, padding, dil, ktrue,
+                    xpch.shape[2], xpch.shape[3])
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        xp, wp = ctx.saved_tensors
+        stride, padding, dil, ktrue, H, W = ctx.meta
+        K64 = wp.shape[0]
+        gy = gy.contiguous(memory_format=torch.channels_last)
+        if K64 != ktrue:  # re-pad the out channels the forward stored dense
+            gy = F.pad(gy, (0, 0, 0, 0, 0, K64 - ktrue)).contiguous(
+                memory_format=torch.channels_last)
+        gx = gw = None
+        if ctx.needs_input_grad[0]:
+            gx = _ext().conv2d_bwd_data(gy, wp, stride, padding, H, W,
+                                        dil=dil, ctrue=-1)
+        if ctx.needs_input_grad[1]:
+            gw = _ext().conv2d_bwd_weight_prepadded(
+                gy, xp, stride, wp.shape[2], wp.shape[3], dil=dil)
+        return gx, gw, None, None, None, None
+
+
+def _chanpad_nhwc(x, c_to):
+    if x.shape[1] != c_to:
+        x = F.pad(x, (0, 0, 0, 0, 0, c_to - x.shape[1]))
+    return x.contiguous(memory_format=torch.channels_last)
+
+
+def _gen_conv(x, weight, bias, stride, padding, dil):
+    """bf16 NHWC generic conv via _GenConvFn, channel-padding both sides."""
+    K, C, R, S = weight.shape
+    C8 = _pad_mult(C, 8)
+    K64 = _pad_mult(K, 64)
+    xb = _chanpad_nhwc(x, C8)
+    wb = weight.to(torch.bfloat16)
+    if (K64, C8) != (K, C):
+        wb = F.pad(wb, (0, 0, 0, 0, 0, C8 - C, 0, K64 - K))
+    wb = wb.contiguous(memory_format=torch.channels_last)
+    y = _GenConvFn.apply(xb, wb, stride, padding, dil, K)
+    if bias is not None:
+        y = y + bias.to(y.dtype).view(1, -1, 1, 1)
+    return y
+
+
+def _gen_conv_supported(mod, x):
+    if not _native(x):
+        return False
+    k = mod.kernel_size
+    d = mod.dilation
+    s = mod.stride
+    if k[0] != k[1] or d[0] != d[1] or s[0] != s[1] or s[0] not in (1, 2) \
+            or mod.groups != 1:
+        return False
+    C8 = _pad_mult(mod.in_channels, 8)
+    return (k[0] * k[1] * C8) % 64 == 0
+
+
+class FedConvGeneric(nn.Conv2d):
+    """Conv2d routed through the generic MFMA path when the shape allows
+    (R*S*pad8(C) % 64 == 0 — true for all VAE/CPC convs); stock F.conv2d
+    otherwise and on CPU."""
+
+    def forward(self, x):
+        if _gen_conv_supported(self, x) and self.padding[0] == self.padding[1]:
+            if x.dtype != torch.bfloat16 and torch.is_autocast_enabled():
+                x = x.to(torch.bfloat16)
+            if x.dtype == torch.bfloat16:
+                return _gen_conv(x, self.weight, self.bias, self.stride[0],
+                                 self.padding[0], self.dilation[0])
+        return F.conv2d(x, self.weight.to(x.dtype),
+                        self.bias.to(x.dtype) if self.bias is not None else None,
+                        self.stride, self.padding, self.dilation, self.groups)
+
+
+class _DilatePadFn(torch.autograd.Function):
+    """y[n][c][h*str+pt][w*str+pl] = x: the input-dilation step of a
+    transposed conv, on the vectorized NHWC pad/dilate kernel; backward is
+    a strided slice."""
+
+    @staticmethod
+    def forward(ctx, x, str_, pt, pb, pl, pr):
+        ctx.meta = (str_, pt, pl, x.shape[2], x.shape[3])
+        return _ext().dilate_pad(x, pt, pb, pl, pr, str_)
+
+    @staticmethod
+    def backward(ctx, gy):
+        str_, pt, pl, H, W = ctx.meta
+        gx = gy[:, :, pt:pt + (H - 1) * str_ + 1:str_,
+                pl:pl + (W - 1) * str_ + 1:str_]
+        return gx.contiguous(memory_format=torch.channels_last), \
+            None, None, None, None, None
+
+
+class FedConvTranspose2d(nn.ConvTranspose2d):
+    """ConvTranspose2d as dilate-input + stride-1 MFMA conv with the
+    spatially-flipped, in/out-swapped weight (the exact adjoint of the
+    forward conv; same identity the ResNet bwd-data path uses).  Covers the
+    VAE decoder 4x4-stride-2 deconvs (simple_models.py:262-265, 336-340)."""
+
+    def forward(self, x, output_size=None):
+        k = self.kernel_size
+        s = self.stride
+        p = self.padding
+        op = self.output_padding
+        C8 = _pad_mult(self.in_channels, 8)
+        ok = (output_size is None and k[0] == k[1] and s[0] == s[1]
+              and p[0] == p[1] and self.dilation == (1, 1)
+              and self.groups == 1 and k[0] - 1 - p[0] >= 0
+              and (k[0] * k[1] * C8) % 64 == 0 and _native(x))
+        if ok:
+            if x.dtype != torch.bfloat16 and torch.is_autocast_enabled():
+                x = x.to(torch.bfloat16)
+            ok = x.dtype == torch.bfloat16
+        if not ok:
+            return super().forward(x, output_size)
+        # weight [Cin][Cout][R][S] -> conv weight [Cout][Cin][R][S], rot180
+        wr = self.weight.flip(2, 3).transpose(0, 1)
+        pt = k[0] - 1 - p[0]
+        xb = _chanpad_nhwc(x, C8)
+        xd = _DilatePadFn.apply(xb, s[0], pt, pt + op[0], pt, pt + op[1])
+        yb = _gen_conv(xd, wr, self.bias, 1, 0, 1)
+        return yb
+
+
 class FedConv2d(nn.Conv2d):
     def forward(self, x):
         if _native(x) and self.bias is None and self.groups == 1 \

@@ -345,3 +345,80 @@ def test_resnet18_training_step_bf16():
     assert torch.isfinite(loss)
     for p in net.parameters():
         assert torch.isfinite(p).all()
+
+
+# ------------------------------------------------------- generic conv (VAE/CPC)
+
+VAE_CPC_CONVS = [
+    # (Cin, Kout, H, ksize, stride, pad, dil) — VAE encoder 4x4-s2 chain
+    # (simple_models.py:249-255) and the CPC dilated bank / 2x2 / 1x1 convs
+    # (simple_models.py:441-460, 478-481, 503-504)
+    (3, 12, 32, 4, 2, 1, 1),
+    (12, 24, 16, 4, 2, 1, 1),
+    (24, 48, 8, 4, 2, 1, 1),
+    (48, 96, 4, 4, 2, 1, 1),
+    (8, 8, 32, 4, 2, 3, 2),      # CPC dilated bank d=2
+    (8, 8, 32, 4, 2, 6, 4),      # d=4
+    (8, 8, 32, 4, 2, 24, 16),    # d=16
+    (40, 256, 16, 4, 2, 1, 1),   # CPC conv2 (8*5 -> latent/4)
+    (256, 256, 6, 2, 1, 1, 1),   # CPC contextgen 2x2 pad 1
+    (1024, 256, 4, 1, 1, 0, 1),  # CPC predictor 1x1
+]
+
+
+@pytest.mark.parametrize("Cin,Kout,H,ks,stride,pad,dil", VAE_CPC_CONVS)
+def test_gen_conv_fwd_bwd_vs_torch(Cin, Kout, H, ks, stride, pad, dil):
+    from fedkit.ops.conv import _gen_conv
+    torch.manual_seed(11)
+    N = 16
+    x = (torch.randn(N, Cin, H, H, device="cuda") * 0.5).requires_grad_(True)
+    w = (torch.randn(Kout, Cin, ks, ks, device="cuda") * 0.1).requires_grad_(True)
+    b = torch.randn(Kout, device="cuda", requires_grad=True)
+    y = _gen_conv(x.to(torch.bfloat16).contiguous(
+        memory_format=torch.channels_last), w, b, stride, pad, dil)
+    want = F.conv2d(x, w, b, stride, pad, dil)
+    assert y.shape == want.shape, (y.shape, want.shape)
+    assert frob_err(y.float(), want) < 2e-2
+    gy = torch.randn_like(want)
+    y.backward(gy.to(y.dtype))
+    gxr, gwr, gbr = torch.autograd.grad(want, [x, w, b], gy)
+    assert frob_err(w.grad.float(), gwr) < 2e-2
+    assert frob_err(b.grad.float(), gbr) < 2e-2
+    assert frob_err(x.grad.float(), gxr) < 2e-2
+
+
+@pytest.mark.parametrize("Cin,Kout,H", [(96, 48, 2), (48, 24, 4),
+                                        (24, 12, 8), (12, 3, 16)])
+def test_conv_transpose_vs_torch(Cin, Kout, H):
+    """FedConvTranspose2d (VAE decoder 4x4-s2, simple_models.py:262-265)."""
+    from fedkit.ops.conv import FedConvTranspose2d
+    torch.manual_seed(12)
+    mod = FedConvTranspose2d(Cin, Kout, 4, stride=2, padding=1).cuda()
+    x = (torch.randn(16, Cin, H, H, device="cuda") * 0.5).requires_grad_(True)
+    y = mod(x.to(torch.bfloat16).contiguous(
+        memory_format=torch.channels_last))
+    want = F.conv_transpose2d(x, mod.weight, mod.bias, 2, 1)
+    assert y.shape == want.shape
+    assert frob_err(y.float(), want) < 2e-2
+    gy = torch.randn_like(want)
+    y.backward(gy.to(y.dtype))
+    gxr, gwr = torch.autograd.grad(want, [x, mod.weight], gy,
+                                   retain_graph=False)
+    assert frob_err(x.grad.float(), gxr) < 2e-2
+    assert frob_err(mod.weight.grad.float(), gwr) < 2e-2
+
+
+def test_vae_model_gpu_step():
+    """AutoEncoderCNN forward+backward end-to-end on the generic conv path."""
+    from fedkit.models import AutoEncoderCNN
+    from fedkit.ops.losses import vae_loss
+    torch.manual_seed(13)
+    net = AutoEncoderCNN().cuda()
+    x = torch.rand(32, 3, 32, 32, device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out, mu, logvar = net(x)
+        loss = vae_loss(out.float(), x, mu.float(), logvar.float())
+    loss.backward()
+    assert torch.isfinite(loss)
+    for p in net.parameters():
+        assert p.grad is None or torch.isfinite(p.grad).all()
