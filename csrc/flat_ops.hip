@@ -121,3 +121,135 @@ void fedkit_add_flat_params(std::vector<at::Tensor> tensors, at::Tensor flat,
                             double alpha) {
   run_chunks(tensors, flat, Op::Axpy, (float)alpha);
 }
+
+// ------------------------------------------------- L-BFGS fused multi-ops
+// The two-loop recursion's 3n+1 separate torch dots each cost a launch AND
+// a device->host sync (lbfgsnew.py:588-659 equivalent).  fedkit instead
+// keeps the small Gram matrices (S^T Y, Y^T Y) on the host and refreshes
+// them with ONE fused pass per history update: multi_dot computes x . v_i
+// for up to 24 history vectors in a single read of x (partials slab +
+// wave-per-column finalize, same shape as the BN stage 2), and lincomb
+// materializes d = c_g * g + sum c_i v_i in one pass.  One sync per
+// step() iteration instead of ~21.
+
+namespace {
+
+constexpr int kMaxVecs = 24;
+
+struct VecPack {
+  const float* p[kMaxVecs];
+  int n;
+};
+
+struct CoefPack {
+  const float* p[kMaxVecs];
+  float c[kMaxVecs];
+  int n;
+};
+
+// partials[b][v] = block b's partial dot of x . p[v]
+__global__ void multi_dot_kernel(VecPack vp, const float* __restrict__ x,
+                                 long long total,
+                                 float* __restrict__ part /* [nb][nvec] */) {
+  __shared__ float red[256];
+  float acc[kMaxVecs] = {};
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    float xv = x[i];
+    for (int v = 0; v < vp.n; ++v) acc[v] += xv * vp.p[v][i];
+  }
+  for (int v = 0; v < vp.n; ++v) {
+    __syncthreads();
+    red[threadIdx.x] = acc[v];
+    __syncthreads();
+    for (int off = 128; off; off >>= 1) {
+      if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) part[(long long)blockIdx.x * vp.n + v] = red[0];
+  }
+}
+
+// out[v] = sum_b part[b][v] — one wave per column (nb <= 640 rows)
+__global__ void multi_dot_finalize_kernel(const float* __restrict__ part,
+                                          int nb, int nvec,
+                                          float* __restrict__ out) {
+  int v = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  int lane = threadIdx.x & 63;
+  if (v >= nvec) return;
+  float s = 0.f;
+  for (int b = lane; b < nb; b += 64) s += part[(long long)b * nvec + v];
+#pragma unroll
+  for (int off = 32; off; off >>= 1) s += __shfl_down(s, off, 64);
+  if (lane == 0) out[v] = s;
+}
+
+// out[i] = cg * g[i] + sum_v c[v] * p[v][i]
+__global__ void lincomb_kernel(CoefPack cp, const float* __restrict__ g,
+                               float cg, float* __restrict__ out,
+                               long long total) {
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    float s = cg * g[i];
+    for (int v = 0; v < cp.n; ++v) s += cp.c[v] * cp.p[v][i];
+    out[i] = s;
+  }
+}
+
+}  // namespace
+
+// returns a DEVICE fp32 tensor [len(vecs)] of dots x . vecs[i]; the caller
+// syncs once with .cpu() when it needs the scalars
+at::Tensor fedkit_multi_dot(std::vector<at::Tensor> vecs,
+                            const at::Tensor& x) {
+  TORCH_CHECK(!vecs.empty() && (int)vecs.size() <= kMaxVecs,
+              "multi_dot supports 1..", kMaxVecs, " vectors");
+  TORCH_CHECK(x.scalar_type() == at::kFloat && x.is_contiguous(),
+              "multi_dot: x must be contiguous fp32");
+  VecPack vp;
+  vp.n = (int)vecs.size();
+  long long total = x.numel();
+  for (int i = 0; i < vp.n; ++i) {
+    TORCH_CHECK(vecs[i].scalar_type() == at::kFloat &&
+                vecs[i].is_contiguous() && vecs[i].numel() == total,
+                "multi_dot: vecs must be contiguous fp32 of x's length");
+    vp.p[i] = vecs[i].data_ptr<float>();
+  }
+  auto stream = fedkit_stream();
+  int nb = grid_1d(total, 256, 512);
+  auto part = at::empty({nb, vp.n}, x.options());
+  auto out = at::empty({vp.n}, x.options());
+  hipLaunchKernelGGL(multi_dot_kernel, dim3(nb), dim3(256), 0, stream, vp,
+                     x.data_ptr<float>(), total, part.data_ptr<float>());
+  hipLaunchKernelGGL(multi_dot_finalize_kernel,
+                     dim3((vp.n + 3) / 4), dim3(256), 0, stream,
+                     part.data_ptr<float>(), nb, vp.n, out.data_ptr<float>());
+  return out;
+}
+
+// out = cg * g + sum coeffs[i] * vecs[i]   (fused direction build)
+at::Tensor fedkit_lincomb(const at::Tensor& g, double cg,
+                          std::vector<at::Tensor> vecs,
+                          std::vector<double> coeffs) {
+  TORCH_CHECK(vecs.size() == coeffs.size() && (int)vecs.size() <= kMaxVecs,
+              "lincomb: vecs/coeffs mismatch or too many");
+  TORCH_CHECK(g.scalar_type() == at::kFloat && g.is_contiguous(),
+              "lincomb: g must be contiguous fp32");
+  CoefPack cp;
+  cp.n = (int)vecs.size();
+  long long total = g.numel();
+  for (int i = 0; i < cp.n; ++i) {
+    TORCH_CHECK(vecs[i].scalar_type() == at::kFloat &&
+                vecs[i].is_contiguous() && vecs[i].numel() == total,
+                "lincomb: vecs must be contiguous fp32 of g's length");
+    cp.p[i] = vecs[i].data_ptr<float>();
+    cp.c[i] = (float)coeffs[i];
+  }
+  auto out = at::empty_like(g);
+  hipLaunchKernelGGL(lincomb_kernel, dim3(grid_1d(total, 256)), dim3(256), 0,
+                     fedkit_stream(), cp, g.data_ptr<float>(), (float)cg,
+                     out.data_ptr<float>(), total);
+  return out;
+}

@@ -12,6 +12,10 @@ void fedkit_pack_params(std::vector<at::Tensor> tensors, at::Tensor flat);
 void fedkit_unpack_params(at::Tensor flat, std::vector<at::Tensor> tensors);
 void fedkit_add_flat_params(std::vector<at::Tensor> tensors, at::Tensor flat,
                             double alpha);
+at::Tensor fedkit_multi_dot(std::vector<at::Tensor> vecs, const at::Tensor& x);
+at::Tensor fedkit_lincomb(const at::Tensor& g, double cg,
+                          std::vector<at::Tensor> vecs,
+                          std::vector<double> coeffs);
 
 std::vector<at::Tensor> fedkit_cross_entropy_fwd(const at::Tensor& logits,
                                                  const at::Tensor& labels);
@@ -62,6 +66,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("unpack_params", &fedkit_unpack_params, "flat fp32 -> multi-tensor");
   m.def("add_flat_params", &fedkit_add_flat_params,
         "t += alpha * flat slice (multi-tensor axpy)");
+  m.def("multi_dot", &fedkit_multi_dot,
+        "x . vecs[i] for up to 24 vectors in one pass -> device fp32 [n]");
+  m.def("lincomb", &fedkit_lincomb,
+        "cg*g + sum c[i]*vecs[i] in one pass (L-BFGS direction build)");
   m.def("cross_entropy_fwd", &fedkit_cross_entropy_fwd,
         "fused log-softmax + NLL (mean): returns (loss, lse)");
   m.def("cross_entropy_bwd", &fedkit_cross_entropy_bwd, "CE backward");

@@ -71,3 +71,33 @@ def add_flat(tensors: Sequence[torch.Tensor], flat: torch.Tensor, alpha: float) 
         require_ext().add_flat_params(list(tensors), flat, float(alpha))
         return
     torch._foreach_add_(list(tensors), _flat_views(flat, tensors), alpha=alpha)
+
+
+def fused_available(t: torch.Tensor) -> bool:
+    """True when the multi_dot/lincomb HIP kernels can run on t."""
+    return _use_native([t])
+
+
+def multi_dot(vecs: Sequence[torch.Tensor], x: torch.Tensor) -> torch.Tensor:
+    """x . vecs[i] for all i in ONE pass over x (device fp32 [len(vecs)]).
+
+    Replaces the reference two-loop's serial torch dots (lbfgsnew.py:645-659),
+    each of which is a separate launch plus a device->host sync.
+    """
+    if _use_native([x]):
+        from . import require_ext
+        return require_ext().multi_dot(list(vecs), x)
+    return torch.stack([torch.dot(v, x) for v in vecs])
+
+
+def lincomb(g: torch.Tensor, cg: float, vecs: Sequence[torch.Tensor],
+            coeffs: Sequence[float]) -> torch.Tensor:
+    """cg*g + sum coeffs[i]*vecs[i] in one fused pass (direction build)."""
+    if _use_native([g]):
+        from . import require_ext
+        return require_ext().lincomb(g, float(cg), list(vecs),
+                                     [float(c) for c in coeffs])
+    out = g.mul(cg)
+    for c, v in zip(coeffs, vecs):
+        out.add_(v, alpha=float(c))
+    return out

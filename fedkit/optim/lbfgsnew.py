@@ -311,6 +311,41 @@ class LBFGSNew(Optimizer):
             r.add_(hist_s[i], alpha=al[i] - be_i)
         return r
 
+    @torch.no_grad()
+    def _two_loop_fused(self, flat_grad, hist_y, hist_s, H_diag, SY, YY):
+        """Two-loop recursion via the compact (Gram-matrix) representation.
+
+        q and r stay in span{g, y_*, s_*}, so every s_i.q / y_i.r the
+        reference computes on the full N-vector (lbfgsnew.py:645-659)
+        reduces to algebra over SY[i][j] = s_i.y_j and YY[i][j] = y_i.y_j
+        (maintained incrementally on history updates) plus the 2n fresh
+        dots (s_i.g, y_i.g) from ONE fused multi_dot pass.  The direction
+        is then materialized by one fused lincomb kernel: ~4 launches and
+        one device sync per iteration instead of ~3n launches and ~2n syncs.
+        """
+        n = len(hist_y)
+        if n == 0:
+            return flat_ops.lincomb(flat_grad, -float(H_diag), [], [])
+        dots = flat_ops.multi_dot(hist_s + hist_y, flat_grad).tolist()
+        sg, yg = dots[:n], dots[n:]
+        ro = [1.0 / SY[i][i] for i in range(n)]
+        # backward pass: q = qg*g + sum qy[j]*y_j
+        qg, qy = -1.0, [0.0] * n
+        al = [0.0] * n
+        for i in range(n - 1, -1, -1):
+            siq = qg * sg[i] + sum(qy[j] * SY[i][j] for j in range(n))
+            al[i] = siq * ro[i]
+            qy[i] -= al[i]
+        # forward pass: r = rg*g + sum ry[j]*y_j + sum rs[j]*s_j
+        rg = H_diag * qg
+        ry = [H_diag * c for c in qy]
+        rs = [0.0] * n
+        for i in range(n):
+            yir = rg * yg[i] + sum(ry[j] * YY[i][j] for j in range(n)) \
+                + sum(rs[j] * SY[j][i] for j in range(n))
+            rs[i] += al[i] - yir * ro[i]
+        return flat_ops.lincomb(flat_grad, rg, hist_y + hist_s, ry + rs)
+
     def step(self, closure):
         assert len(self.param_groups) == 1
         group = self.param_groups[0]
@@ -359,6 +394,7 @@ class LBFGSNew(Optimizer):
             if state["n_iter"] == 1:
                 d = flat_grad.neg()
                 hist_y, hist_s = [], []
+                state["SY"], state["YY"] = [], []
                 H_diag = 1
                 if batch_mode:
                     running_avg = torch.zeros_like(flat_grad)
@@ -375,8 +411,25 @@ class LBFGSNew(Optimizer):
                 s = d.mul(t)
                 if batch_mode:
                     y.add_(s, alpha=_BATCH_LM0)      # trust-region damping
-                ys = float(y.dot(s))
-                sn = float(s.norm())
+
+                # fused path: every scalar the update and the two-loop need
+                # comes from two multi_dot passes and ONE host sync
+                fused = flat_ops.fused_available(flat_grad)
+                SY, YY = state.get("SY", []), state.get("YY", [])
+                n0 = len(hist_y)
+                if fused:
+                    d1 = flat_ops.multi_dot(hist_s + hist_y + [s, y], y)
+                    d2 = flat_ops.multi_dot(hist_y + [s], s)
+                    vals = torch.cat([d1, d2]).tolist()
+                    s_y = vals[:n0]                  # s_i . y_new
+                    y_y = vals[n0:2 * n0]            # y_i . y_new
+                    ys = vals[2 * n0]
+                    yy = vals[2 * n0 + 1]
+                    y_s = vals[2 * n0 + 2:3 * n0 + 2]  # y_j . s_new
+                    sn2 = vals[3 * n0 + 2]
+                else:
+                    ys = float(y.dot(s))
+                    sn2 = float(s.norm()) ** 2
 
                 # first iteration of a new step() call == new minibatch
                 batch_changed = batch_mode and (n_iter == 1 and state["n_iter"] > 1)
@@ -389,22 +442,43 @@ class LBFGSNew(Optimizer):
                     alphabar = 1.0 / (1.0 + float(running_avg_sq.sum())
                                       / ((state["n_iter"] - 1) * grad_nrm))
 
-                if ys > 1e-10 * sn * sn and not batch_changed:
+                if ys > 1e-10 * sn2 and not batch_changed:
                     if len(hist_y) == history_size:
                         hist_y.pop(0)
                         hist_s.pop(0)
+                        if fused:
+                            SY.pop(0)
+                            YY.pop(0)
+                            for row in SY:
+                                row.pop(0)
+                            for row in YY:
+                                row.pop(0)
+                            s_y, y_y, y_s = s_y[1:], y_y[1:], y_s[1:]
                     hist_y.append(y)
                     hist_s.append(s)
-                    H_diag = ys / float(y.dot(y))
+                    if fused:
+                        for i, row in enumerate(SY):
+                            row.append(s_y[i])
+                        SY.append(list(y_s) + [ys])
+                        for i, row in enumerate(YY):
+                            row.append(y_y[i])
+                        YY.append(list(y_y) + [yy])
+                        H_diag = ys / yy
+                    else:
+                        H_diag = ys / float(y.dot(y))
 
                 if isinstance(H_diag, float) and math.isnan(H_diag):
                     print("Warning H_diag nan")
 
-                if "ro" not in state:
-                    state["ro"] = [None] * history_size
-                    state["al"] = [None] * history_size
-                d = self._two_loop(flat_grad, hist_y, hist_s, H_diag,
-                                   state["ro"], state["al"])
+                if fused:
+                    d = self._two_loop_fused(flat_grad, hist_y, hist_s,
+                                             H_diag, SY, YY)
+                else:
+                    if "ro" not in state:
+                        state["ro"] = [None] * history_size
+                        state["al"] = [None] * history_size
+                    d = self._two_loop(flat_grad, hist_y, hist_s, H_diag,
+                                       state["ro"], state["al"])
 
             if prev_flat_grad is None:
                 prev_flat_grad = flat_grad.clone()
