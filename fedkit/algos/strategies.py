@@ -68,6 +68,18 @@ class Strategy:
         """Run the communication round; returns log info.  Updates state."""
         raise NotImplementedError
 
+    # -- split form: start() launches the collective (comm stream), finish()
+    #    consumes it.  The engine can slot independent work (test-set eval
+    #    for strategies that do not write z back) between the two.
+    def aggregate_start(self, comm: Communicator, state: dict,
+                        x: Dict[int, torch.Tensor]):
+        return None
+
+    def aggregate_finish(self, comm: Communicator, state: dict,
+                         x: Dict[int, torch.Tensor], pending,
+                         round_idx: int) -> dict:
+        return self.aggregate(comm, state, x, round_idx)
+
 
 class NoConsensus(Strategy):
     """K independent models, no communication ever (no_consensus_multi.py)."""
@@ -82,13 +94,22 @@ class FedAvg(Strategy):
     name = "fedavg"
     writeback_z = True
 
-    def aggregate(self, comm, state, x, round_idx):
+    def aggregate_start(self, comm, state, x):
+        return comm.sum_across_clients_async(
+            {k: v.clone() for k, v in x.items()})
+
+    def aggregate_finish(self, comm, state, x, pending, round_idx):
         z, N = state["z"], state["N"]
-        znew = comm.sum_across_clients({k: v.clone() for k, v in x.items()})
+        znew = pending()
         znew /= comm.K
         dual_residual = torch.norm(z - znew).item() / N
         state["z"] = znew
         return {"dual": dual_residual}
+
+    def aggregate(self, comm, state, x, round_idx):
+        return self.aggregate_finish(comm, state, x,
+                                     self.aggregate_start(comm, state, x),
+                                     round_idx)
 
 
 class FedProx(Strategy):
@@ -108,9 +129,18 @@ class FedProx(Strategy):
         xdelta = xvec - state["z"]
         return 0.5 * state["rho"] * (torch.norm(xdelta, 2) ** 2)
 
+    def aggregate_start(self, comm, state, x):
+        return comm.sum_across_clients_async(
+            {k: v.clone() for k, v in x.items()})
+
     def aggregate(self, comm, state, x, round_idx):
+        return self.aggregate_finish(comm, state, x,
+                                     self.aggregate_start(comm, state, x),
+                                     round_idx)
+
+    def aggregate_finish(self, comm, state, x, pending, round_idx):
         z, N, rho = state["z"], state["N"], state["rho"]
-        znew = comm.sum_across_clients({k: v.clone() for k, v in x.items()})
+        znew = pending()
         znew /= comm.K
         dual_residual = torch.norm(z - znew).item() / N
         state["z"] = znew

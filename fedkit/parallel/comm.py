@@ -42,6 +42,18 @@ class Communicator:
     def device(self) -> torch.device:
         raise NotImplementedError
 
+    def sum_across_clients_async(self, per_client: Dict[int, torch.Tensor]):
+        """Launch the cross-client sum; returns wait() -> summed tensor.
+
+        The MI355X overlap hook: DistComm runs the RCCL
+        all-reduce on a dedicated HIP stream so independent work (e.g. the
+        per-round test-set evaluation for non-writeback strategies) proceeds
+        on the compute stream; wait() fences the compute stream on the
+        collective, without blocking the host.  Default: synchronous.
+        """
+        out = self.sum_across_clients(per_client)
+        return lambda: out
+
     def sum_across_clients(self, per_client: Dict[int, torch.Tensor]) -> torch.Tensor:
         """Return sum over ALL K clients of a per-client vector.
 
@@ -130,6 +142,32 @@ class DistComm(Communicator):
         vec = per_client[self.rank].contiguous()
         dist.all_reduce(vec, op=dist.ReduceOp.SUM)
         return vec
+
+    def sum_across_clients_async(self, per_client):
+        assert list(per_client.keys()) == self.my_clients
+        vec = per_client[self.rank].contiguous()
+        if self.comm_stream is not None:
+            # comm stream waits for the producer (pack) on the compute
+            # stream, runs the collective, and wait() makes the compute
+            # stream wait on the result -- no host blocking on either side
+            ev = torch.cuda.Event()
+            ev.record()
+            with torch.cuda.stream(self.comm_stream):
+                self.comm_stream.wait_event(ev)
+                work = dist.all_reduce(vec, op=dist.ReduceOp.SUM,
+                                       async_op=True)
+            vec.record_stream(self.comm_stream)
+
+            def wait():
+                work.wait()      # fences the CURRENT (compute) stream
+                return vec
+            return wait
+        work = dist.all_reduce(vec, op=dist.ReduceOp.SUM, async_op=True)
+
+        def wait():
+            work.wait()
+            return vec
+        return wait
 
     def gather_scalar_rows(self, per_client):
         row = torch.tensor(per_client[self.rank], dtype=torch.float64)

@@ -302,7 +302,18 @@ class FederatedJob:
                     t0 = time.perf_counter()
                     x = {ck: get_trainable_values(self.nets[ck], self.device)
                          for ck in comm.my_clients}
-                    info = self.strategy.aggregate(comm, self._state, x, nadmm)
+                    # launch the all-reduce on the comm stream; for
+                    # strategies that do not write z back (FedProx/ADMM the
+                    # models keep their own x_k) the full test-set eval runs
+                    # on the compute stream WHILE the collective is in
+                    # flight over xGMI (SURVEY.md §5 overlap design)
+                    pending = self.strategy.aggregate_start(
+                        comm, self._state, x)
+                    accs = None
+                    if cfg.check_results and not self.strategy.writeback_z:
+                        accs = self.verification_error_check()
+                    info = self.strategy.aggregate_finish(
+                        comm, self._state, x, pending, nadmm)
                     if self.strategy.writeback_z:
                         for ck in comm.my_clients:
                             put_trainable_values(self.nets[ck], self._state["z"])
@@ -310,8 +321,7 @@ class FederatedJob:
 
                     self._print_round(nloop, ci, nadmm, info,
                                       epoch=cfg.Nepoch - 1)
-                    accs = None
-                    if cfg.check_results:
+                    if cfg.check_results and accs is None:
                         accs = self.verification_error_check()
                     self._log_round(nloop, ci, nadmm, N, info, t_local,
                                     t_comm, accs)
