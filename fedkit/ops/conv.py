@@ -5,11 +5,17 @@ reference's modules exactly (the federated block partitions index parameter
 tensors by position — simple_models.py:222-226).
 
 GPU path (ROCm, fedkit._C built): NHWC bf16 implicit-GEMM on MFMA
-(csrc/conv2d_mfma.hip) for the ResNet shapes (3x3 s1/s2 pad1 and 1x1 s1/s2,
-no bias), covering fwd, bwd-data and bwd-weight.
-fp32 accumulate; weights/activations bf16.  Unsupported shapes raise on GPU
-rather than silently falling back (bias convs in the non-flagship models use
-plain nn.Conv2d modules and are not routed here).
+(csrc/conv2d_mfma.hip), fp32 accumulate, covering fwd, bwd-data and
+bwd-weight:
+  * FedConv2d — the flagship ResNet shapes (3x3 s1/s2 pad1, 1x1 s1/s2,
+    no bias); unsupported shapes raise on GPU rather than silently
+    falling back;
+  * FedConvGeneric — any channel counts via zero channel padding
+    (in -> x8, out -> x64 with dense-at-true-K stores), bias, dilation:
+    the VAE encoder 4x4-s2 chain and the CPC dilated bank / 2x2 / 1x1
+    convs;
+  * FedConvTranspose2d — the VAE decoder deconvs as dilate-pad + the
+    stride-1 conv adjoint.
 
 CPU path: stock F.conv2d (tests compare the HIP kernels against this).
 """
@@ -72,6 +78,27 @@ def _pad_mult(n, m):
     return (n + m - 1) // m * m
 
 
+def _cached_frozen(mod, key, src_param, make):
+    """Per-module cache of a derived (bf16/padded) weight while the source
+    parameter is FROZEN.  In per-block federated training all but one
+    layer-block is frozen (simple_utils.py:34-45 semantics), so the
+    per-forward fp32->bf16 cast of every conv weight (~200 us/step on
+    ResNet18) is recomputed only when the parameter's in-place version
+    counter moves (optimizer step / put_trainable_values / load).
+    Trainable parameters are never cached: their cast must live in the
+    current autograd graph so gradients reach the fp32 master."""
+    if src_param.requires_grad:
+        return make()
+    ver = src_param._version
+    cache = getattr(mod, key, None)
+    if cache is not None and cache[0] == ver:
+        return cache[1]
+    with torch.no_grad():
+        out = make()
+    setattr(mod, key, (ver, out))
+    return out
+
+
 class _GenConvFn(torch.autograd.Function):
     """Generic conv on the MFMA kernels: any in/out channel count via zero
     channel padding (in -> mult of 8, out -> mult of 64, with the kernel
@@ -120,16 +147,25 @@ def _chanpad_nhwc(x, c_to):
     return x.contiguous(memory_format=torch.channels_last)
 
 
-def _gen_conv(x, weight, bias, stride, padding, dil):
-    """bf16 NHWC generic conv via _GenConvFn, channel-padding both sides."""
+def _prep_gen_weight(weight):
+    """fp32 [K][C][R][S] -> bf16 channels_last, channel-padded to
+    [pad64(K)][pad8(C)][R][S] (autograd slices the grads back through
+    F.pad / .to)."""
     K, C, R, S = weight.shape
     C8 = _pad_mult(C, 8)
     K64 = _pad_mult(K, 64)
-    xb = _chanpad_nhwc(x, C8)
     wb = weight.to(torch.bfloat16)
     if (K64, C8) != (K, C):
         wb = F.pad(wb, (0, 0, 0, 0, 0, C8 - C, 0, K64 - K))
-    wb = wb.contiguous(memory_format=torch.channels_last)
+    return wb.contiguous(memory_format=torch.channels_last)
+
+
+def _gen_conv(x, weight, bias, stride, padding, dil, wb=None):
+    """bf16 NHWC generic conv via _GenConvFn, channel-padding both sides."""
+    K, C = weight.shape[0], weight.shape[1]
+    xb = _chanpad_nhwc(x, _pad_mult(C, 8))
+    if wb is None:
+        wb = _prep_gen_weight(weight)
     y = _GenConvFn.apply(xb, wb, stride, padding, dil, K)
     if bias is not None:
         y = y + bias.to(y.dtype).view(1, -1, 1, 1)
@@ -159,8 +195,10 @@ class FedConvGeneric(nn.Conv2d):
             if x.dtype != torch.bfloat16 and torch.is_autocast_enabled():
                 x = x.to(torch.bfloat16)
             if x.dtype == torch.bfloat16:
+                wb = _cached_frozen(self, "_wgen", self.weight,
+                                    lambda: _prep_gen_weight(self.weight))
                 return _gen_conv(x, self.weight, self.bias, self.stride[0],
-                                 self.padding[0], self.dilation[0])
+                                 self.padding[0], self.dilation[0], wb=wb)
         return F.conv2d(x, self.weight.to(x.dtype),
                         self.bias.to(x.dtype) if self.bias is not None else None,
                         self.stride, self.padding, self.dilation, self.groups)
@@ -209,10 +247,12 @@ class FedConvTranspose2d(nn.ConvTranspose2d):
             return super().forward(x, output_size)
         # weight [Cin][Cout][R][S] -> conv weight [Cout][Cin][R][S], rot180
         wr = self.weight.flip(2, 3).transpose(0, 1)
+        wb = _cached_frozen(self, "_wgen", self.weight,
+                            lambda: _prep_gen_weight(wr))
         pt = k[0] - 1 - p[0]
         xb = _chanpad_nhwc(x, C8)
         xd = _DilatePadFn.apply(xb, s[0], pt, pt + op[0], pt, pt + op[1])
-        yb = _gen_conv(xd, wr, self.bias, 1, 0, 1)
+        yb = _gen_conv(xd, wr, self.bias, 1, 0, 1, wb=wb)
         return yb
 
 
@@ -227,8 +267,10 @@ class FedConv2d(nn.Conv2d):
                 # the fp32->bf16 weight cast is autograd-tracked, so bwd-weight
                 # gradients land on the fp32 master copy
                 x = x.contiguous(memory_format=torch.channels_last)
-                w = self.weight.to(torch.bfloat16).contiguous(
-                    memory_format=torch.channels_last)
+                w = _cached_frozen(
+                    self, "_w16", self.weight,
+                    lambda: self.weight.to(torch.bfloat16).contiguous(
+                        memory_format=torch.channels_last))
                 return _ConvFn.apply(x, w, self.stride[0], self.padding[0])
         return F.conv2d(x, self.weight.to(x.dtype),
                         self.bias.to(x.dtype) if self.bias is not None else None,
