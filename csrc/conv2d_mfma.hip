@@ -125,19 +125,37 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   auto bufB = [&](int b) -> char* { return smem + b * (AB + BB) + AB; };
 
   bf16x8 areg[2][A_SLOTS], brg[2][B_SLOTS];
+  const bool c64 = (C % 64) == 0;
   auto stage = [&](int buf, int kt, int rset) {
     // A tile: per-slot source (r,s,c) from k_global; 16-B LDS-DMA.
     // LDS dest for a glds is wave-uniform base + lane*16: slot d = pass*256
     // + wave*64 + lane matches d = pass*256 + tid exactly.
-#pragma unroll
-    for (int i = 0; i < A_SLOTS; ++i) {
-      int kg = kt * BK + a_k8[i] * 8;
-      int c = kg % C;
-      int rs = kg / C;
+    // When C % 64 == 0 (every flagship shape) a whole 64-k tile lies in ONE
+    // (r,s) filter tap, so the kg -> (r,s,c) split is SCALAR per stage —
+    // the per-lane divisions by runtime C/S otherwise cost a magic-number
+    // sequence per slot right in the hot loop.
+    long long tap_off = 0;
+    int c0 = 0;
+    if (c64) {
+      int rs = (kt * BK) / C;
+      c0 = kt * BK - rs * C;
       int s = rs % S;
       int r = rs / S;
-      const bf16* src =
-          xp + a_rowbase[i] + ((long long)r * dil * Wp + s * dil) * C + c;
+      tap_off = ((long long)r * dil * Wp + s * dil) * C;
+    }
+#pragma unroll
+    for (int i = 0; i < A_SLOTS; ++i) {
+      const bf16* src;
+      if (c64) {
+        src = xp + a_rowbase[i] + tap_off + c0 + a_k8[i] * 8;
+      } else {
+        int kg = kt * BK + a_k8[i] * 8;
+        int c = kg % C;
+        int rs = kg / C;
+        int s = rs % S;
+        int r = rs / S;
+        src = xp + a_rowbase[i] + ((long long)r * dil * Wp + s * dil) * C + c;
+      }
       if (REGSTAGE)
         __builtin_memcpy(&areg[rset][i], src, sizeof(bf16x8));
       else
@@ -609,6 +627,36 @@ __global__ void im2col_kernel(const bf16* __restrict__ xp,
   }
 }
 
+// 3-D-grid dilate/pad for C % 8 == 0: block.x covers Wp*Cv (Cv = C/8 is a
+// power of two -> shifts), grid.y = Hp, grid.z = N — no per-element integer
+// division (the flat-index form's div/mod by runtime Wp/Hp/Cv cost more
+// than the memory traffic at these sizes).
+__global__ void dilate_pad_out3d_kernel(const bf16* __restrict__ in,
+                                        bf16* __restrict__ out, int H, int W,
+                                        int cvshift, int Wp, int pt, int pl,
+                                        int str) {
+  typedef __attribute__((ext_vector_type(8))) __bf16 v8;
+  const int Cv = 1 << cvshift;
+  const int hp = blockIdx.y, n = blockIdx.z;
+  const int hs = hp - pt;
+  const bool hok = hs >= 0 && hs % str == 0 && hs / str < H;
+  const int h = hs / str;
+  const v8* inv = reinterpret_cast<const v8*>(in);
+  v8* outv = reinterpret_cast<v8*>(out);
+  long long orow = ((long long)n * gridDim.y + hp) * Wp << cvshift;
+  long long irow = ((long long)n * H + h) * W << cvshift;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < (Wp << cvshift);
+       i += gridDim.x * blockDim.x) {
+    int wp = i >> cvshift;
+    int cv = i - (wp << cvshift);
+    int ws = wp - pl;
+    v8 v = {};
+    if (hok && ws >= 0 && ws % str == 0 && ws / str < W)
+      v = inv[irow + ((long long)(ws / str) << cvshift) + cv];
+    outv[orow + i] = v;
+  }
+}
+
 at::Tensor dilate_pad_core(const at::Tensor& x, int pt, int pb, int pl,
                            int pr, int str) {
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
@@ -616,7 +664,14 @@ at::Tensor dilate_pad_core(const at::Tensor& x, int pt, int pb, int pl,
   auto xp = at::empty({N, C, Hp, Wp},
                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto stream = fedkit_stream();
-  if (C % 8 == 0) {
+  if (C % 8 == 0 && ((C / 8) & (C / 8 - 1)) == 0) {
+    int cvshift = __builtin_ctz((unsigned)(C / 8));
+    int wc = Wp << cvshift;
+    hipLaunchKernelGGL(dilate_pad_out3d_kernel,
+                       dim3((wc + 255) / 256, Hp, N), dim3(256), 0, stream,
+                       (const bf16*)x.data_ptr(), (bf16*)xp.data_ptr(), H, W,
+                       cvshift, Wp, pt, pl, str);
+  } else if (C % 8 == 0) {
     long long total = (long long)N * Hp * Wp * C / 8;
     hipLaunchKernelGGL(dilate_pad_out_kernel<8>, dim3(grid_1d(total, 256)),
                        dim3(256), 0, stream, (const bf16*)x.data_ptr(),
