@@ -40,6 +40,10 @@ def main():
     ap.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     ap.add_argument("--native", type=int, default=1,
                     help="0 = eager torch ops for A/B comparison")
+    ap.add_argument("--graph", type=int, default=0,
+                    help="1 = hipGraph-capture the local step (measured: no "
+                         "gain at this size — the GPU, not the host enqueue "
+                         "path, is saturated; kept for larger configs)")
     args = ap.parse_args()
 
     if not args.native:
@@ -72,10 +76,12 @@ def main():
     net = ResNet18().to(device)
     if use_cuda:
         net = net.to(memory_format=torch.channels_last)
-    # capturable=True keeps Adam's step counters on device so the whole
-    # local step records into one hipGraph (launch-bound otherwise:
-    # ~380 kernel launches/step of host enqueue)
-    opt = torch.optim.Adam(net.parameters(), lr=1e-3, capturable=use_cuda)
+    # fused=True runs the whole Adam update as one multi-tensor kernel
+    # sweep; capturable only when hipGraph capture is requested (device-side
+    # step counters cost a little eagerly)
+    opt = torch.optim.Adam(net.parameters(), lr=1e-3,
+                           capturable=bool(args.graph and use_cuda),
+                           fused=use_cuda and not args.graph)
     blocks = net.train_order_block_ids()
     params = list(net.parameters())
 
@@ -117,7 +123,7 @@ def main():
     # FedAvg all-reduce stays OUTSIDE the graph (collectives every
     # AGG_EVERY steps, eager).  Falls back to eager if capture fails.
     graph = None
-    if use_cuda:
+    if use_cuda and args.graph:
         try:
             static_x = xs[0].clone()
             static_y = ys[0].clone()
@@ -134,6 +140,8 @@ def main():
             print(f"[bench] hipGraph capture unavailable ({e}); eager path",
                   file=sys.stderr)
             graph = None
+        else:
+            print("[bench] hipGraph capture active", file=sys.stderr)
 
     def one_step():
         nonlocal step_i

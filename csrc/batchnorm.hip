@@ -369,7 +369,7 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
       TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
                   "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                   ") | 256, got C=", C);
-      int nb = grid_1d(M * C / VEC, 256, 1024);
+      int nb = grid_1d(M * C / VEC, 256, 512);
       auto part = at::empty({nb, 2, C}, fopts);
       hipLaunchKernelGGL((bn_partials_kernel<scalar_t, VEC>),
                          dim3(nb), dim3(256), 0,
@@ -456,7 +456,7 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
     TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,
                 "bn kernel needs C % ", VEC, " == 0 and (C/", VEC,
                 ") | 256, got C=", C);
-    int nb = grid_1d(M * C / VEC, 256, 1024);
+    int nb = grid_1d(M * C / VEC, 256, 512);
     auto part = at::empty({nb, 2, C}, fopts);
     auto launch1 = [&](auto ekind) {
       hipLaunchKernelGGL((bn_bwd_partials_kernel<scalar_t, VEC,
