@@ -78,13 +78,30 @@ _SYNTH_CACHE = {}
 
 
 def _synthetic_cifar(train: bool, seed: int = 1234):
-    """Deterministic CIFAR-shaped uint8 images + balanced labels."""
+    """Deterministic CIFAR-shaped uint8 images + labels.
+
+    Class-STRUCTURED (not pure noise): one random smooth prototype per
+    class plus heavy pixel noise, train/test drawn from the same
+    distribution.  This keeps accuracy curves meaningful without network
+    access to the real CIFAR10 — the federated-vs-standalone ORDERING the
+    reference's comparison.png demonstrates (README.md:28-31) is
+    reproducible on it (see profiles/acc_synthetic.md).
+    """
     key = (train, seed)
     if key not in _SYNTH_CACHE:
         n = CIFAR_TRAIN_N if train else CIFAR_TEST_N
-        g = torch.Generator().manual_seed(seed + (0 if train else 1))
-        x = torch.randint(0, 256, (n, 3, 32, 32), generator=g, dtype=torch.uint8)
-        y = torch.randint(0, 10, (n,), generator=g, dtype=torch.long)
+        g = torch.Generator().manual_seed(seed)  # prototypes shared
+        # smooth prototypes: upsampled 8x8 random fields, amplified
+        proto = torch.rand(10, 3, 8, 8, generator=g)
+        proto = torch.nn.functional.interpolate(proto, size=(32, 32),
+                                                mode="bilinear",
+                                                align_corners=False)
+        g2 = torch.Generator().manual_seed(seed + (0 if train else 1))
+        y = torch.arange(n, dtype=torch.long) % 10
+        perm = torch.randperm(n, generator=g2)
+        y = y[perm]
+        noise = torch.rand(n, 3, 32, 32, generator=g2)
+        x = (0.3 * proto[y] + 0.7 * noise).mul_(255).to(torch.uint8)
         _SYNTH_CACHE[key] = (x, y)
     return _SYNTH_CACHE[key]
 

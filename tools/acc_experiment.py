@@ -1,0 +1,96 @@
+#!/usr/bin/env python3
+"""Accuracy-ordering experiment — the reference's comparison.png protocol.
+
+The reference's only published result is an accuracy figure (README.md:28-31,
+K=10, `Net` CNN, CIFAR10): standalone K=1 (upper bound) >= FedAvg >=
+consensus >= standalone K=10 (lower bound).  There is no network access for
+the real CIFAR10, so this runs the SAME protocol on the deterministic
+class-structured synthetic set (fedkit.data.cifar._synthetic_cifar) and
+records the ordering.  Output: one JSON line per config on stdout; the
+curated result is committed as profiles/acc_synthetic.md.
+
+    python tools/acc_experiment.py [--nloop 4] [--nadmm 3] [--quick]
+"""
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from fedkit.parallel import FedConfig, FederatedJob
+from fedkit.parallel.runtime import run_standalone
+
+
+def common(K, args, **kw):
+    return FedConfig(
+        K=K, model="Net", default_batch=128,
+        Nloop=args.nloop, Nepoch=1, Nadmm=args.nadmm,
+        check_results=False, save_model=False, load_model=False,
+        init_model=True, biased_input=True, be_verbose=False,
+        use_cuda=True, max_steps_per_epoch=args.max_steps, **kw)
+
+
+def final_accs(job):
+    return [100.0 * float(a) for a in job.verification_error_check()]
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--nloop", type=int, default=4)
+    ap.add_argument("--nadmm", type=int, default=3)
+    ap.add_argument("--max-steps", type=int, default=0, dest="max_steps")
+    ap.add_argument("--quick", action="store_true",
+                    help="tiny run for smoke testing")
+    args = ap.parse_args()
+    if args.quick:
+        args.nloop, args.nadmm, args.max_steps = 1, 1, 3
+
+    results = {}
+
+    # standalone K=1, full data, Nepoch scaled to match total passes
+    t0 = time.time()
+    cfg = common(1, args)
+    cfg.strategy = "none"
+    cfg.Nepoch = args.nloop
+    job = run_standalone(cfg)
+    results["standalone_K1"] = final_accs(job)
+
+    # standalone K=10 (1/K data each, no communication)
+    cfg = common(10, args, strategy="none")
+    job = FederatedJob(cfg)
+    job.run()
+    results["standalone_K10"] = final_accs(job)
+
+    # FedAvg K=10
+    cfg = common(10, args, strategy="fedavg")
+    job = FederatedJob(cfg)
+    job.run()
+    results["fedavg_K10"] = final_accs(job)
+
+    # consensus ADMM K=10 (rho0 = 0.1, consensus_multi.py:29)
+    cfg = common(10, args, strategy="admm", admm_rho0=0.1)
+    job = FederatedJob(cfg)
+    job.run()
+    results["admm_K10"] = final_accs(job)
+
+    out = {name: {"mean_acc": round(statistics.mean(a), 2),
+                  "per_client": [round(v, 1) for v in a]}
+           for name, a in results.items()}
+    out["wall_s"] = round(time.time() - t0, 1)
+    out["protocol"] = (f"Net CNN, synthetic class-structured CIFAR-shaped "
+                       f"data, Nloop={args.nloop} Nadmm={args.nadmm} "
+                       f"Nepoch=1, Adam lr=1e-3, batch 128")
+    print(json.dumps(out))
+    ordering = (out["standalone_K1"]["mean_acc"],
+                out["fedavg_K10"]["mean_acc"],
+                out["admm_K10"]["mean_acc"],
+                out["standalone_K10"]["mean_acc"])
+    print("ordering K1 >= fedavg >= admm >= K10:", ordering, file=sys.stderr)
+
+
+if __name__ == "__main__":
+    main()
