@@ -90,18 +90,24 @@ def _synthetic_cifar(train: bool, seed: int = 1234):
     key = (train, seed)
     if key not in _SYNTH_CACHE:
         n = CIFAR_TRAIN_N if train else CIFAR_TEST_N
+        M = 150                      # prototype clusters per class
         g = torch.Generator().manual_seed(seed)  # prototypes shared
-        # smooth prototypes: upsampled 8x8 random fields, amplified
-        proto = torch.rand(10, 3, 8, 8, generator=g)
+        # smooth cluster prototypes: upsampled 8x8 random fields.  A sample
+        # is one of its class's M clusters plus heavy noise, so accuracy is
+        # limited by how many clusters the training shard has covered —
+        # i.e. by data VOLUME, like a real vision task's learning curve.
+        proto = torch.rand(10 * M, 3, 8, 8, generator=g)
         proto = torch.nn.functional.interpolate(proto, size=(32, 32),
                                                 mode="bilinear",
                                                 align_corners=False)
+        proto = proto.view(10, M, 3, 32, 32)
         g2 = torch.Generator().manual_seed(seed + (0 if train else 1))
         y = torch.arange(n, dtype=torch.long) % 10
         perm = torch.randperm(n, generator=g2)
         y = y[perm]
+        j = torch.randint(0, M, (n,), generator=g2)
         noise = torch.rand(n, 3, 32, 32, generator=g2)
-        x = (0.3 * proto[y] + 0.7 * noise).mul_(255).to(torch.uint8)
+        x = (0.4 * proto[y, j] + 0.6 * noise).mul_(255).to(torch.uint8)
         _SYNTH_CACHE[key] = (x, y)
     return _SYNTH_CACHE[key]
 
