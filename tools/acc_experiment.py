@@ -45,37 +45,54 @@ def main():
     ap.add_argument("--max-steps", type=int, default=0, dest="max_steps")
     ap.add_argument("--quick", action="store_true",
                     help="tiny run for smoke testing")
+    ap.add_argument("--only", type=str, default="",
+                    help="comma list of configs to run")
+    ap.add_argument("--rho", type=float, default=0.1)
+    ap.add_argument("--bb", type=int, default=0)
     args = ap.parse_args()
     if args.quick:
         args.nloop, args.nadmm, args.max_steps = 1, 1, 3
 
     results = {}
-
-    # standalone K=1, full data, Nepoch scaled to match total passes
     t0 = time.time()
-    cfg = common(1, args)
-    cfg.strategy = "none"
-    cfg.Nepoch = args.nloop
-    job = run_standalone(cfg)
-    results["standalone_K1"] = final_accs(job)
+    only = [s for s in args.only.split(",") if s]
 
-    # standalone K=10 (1/K data each, no communication)
-    cfg = common(10, args, strategy="none")
-    job = FederatedJob(cfg)
-    job.run()
-    results["standalone_K10"] = final_accs(job)
+    def want(name):
+        return not only or name in only
 
-    # FedAvg K=10
-    cfg = common(10, args, strategy="fedavg")
-    job = FederatedJob(cfg)
-    job.run()
-    results["fedavg_K10"] = final_accs(job)
+    if want("standalone_K1"):
+        # standalone K=1, full data, Nepoch scaled to match total passes
+        cfg = common(1, args)
+        cfg.strategy = "none"
+        cfg.Nepoch = args.nloop
+        job = run_standalone(cfg)
+        results["standalone_K1"] = final_accs(job)
 
-    # consensus ADMM K=10 (rho0 = 0.1, consensus_multi.py:29)
-    cfg = common(10, args, strategy="admm", admm_rho0=0.1)
-    job = FederatedJob(cfg)
-    job.run()
-    results["admm_K10"] = final_accs(job)
+    if want("standalone_K10"):
+        cfg = common(10, args, strategy="none")
+        job = FederatedJob(cfg)
+        job.run()
+        results["standalone_K10"] = final_accs(job)
+
+    if want("fedavg_K10"):
+        cfg = common(10, args, strategy="fedavg")
+        job = FederatedJob(cfg)
+        job.run()
+        results["fedavg_K10"] = final_accs(job)
+
+    if want("fedprox_K10"):
+        cfg = common(10, args, strategy="fedprox", admm_rho0=1.0)
+        job = FederatedJob(cfg)
+        job.run()
+        results["fedprox_K10"] = final_accs(job)
+
+    if want("admm_K10"):
+        # consensus ADMM K=10 (rho0, optionally BB-adaptive)
+        cfg = common(10, args, strategy="admm", admm_rho0=args.rho,
+                     bb_update=bool(args.bb))
+        job = FederatedJob(cfg)
+        job.run()
+        results["admm_K10"] = final_accs(job)
 
     out = {name: {"mean_acc": round(statistics.mean(a), 2),
                   "per_client": [round(v, 1) for v in a]}
@@ -85,11 +102,10 @@ def main():
                        f"data, Nloop={args.nloop} Nadmm={args.nadmm} "
                        f"Nepoch=1, Adam lr=1e-3, batch 128")
     print(json.dumps(out))
-    ordering = (out["standalone_K1"]["mean_acc"],
-                out["fedavg_K10"]["mean_acc"],
-                out["admm_K10"]["mean_acc"],
-                out["standalone_K10"]["mean_acc"])
-    print("ordering K1 >= fedavg >= admm >= K10:", ordering, file=sys.stderr)
+    keys = ["standalone_K1", "fedavg_K10", "fedprox_K10", "admm_K10",
+            "standalone_K10"]
+    print("ordering", [(k, out[k]["mean_acc"]) for k in keys if k in out],
+          file=sys.stderr)
 
 
 if __name__ == "__main__":
