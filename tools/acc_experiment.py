@@ -81,7 +81,7 @@ def main():
         results["fedavg_K10"] = final_accs(job)
 
     if want("fedprox_K10"):
-        cfg = common(10, args, strategy="fedprox", admm_rho0=1.0)
+        cfg = common(10, args, strategy="fedprox", admm_rho0=args.rho)
         job = FederatedJob(cfg)
         job.run()
         results["fedprox_K10"] = final_accs(job)
