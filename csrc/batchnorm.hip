@@ -174,6 +174,43 @@ __global__ void bn_colsum_finalize_kernel(
   }
 }
 
+// finalize from CONV-EPILOGUE partials ([Kout/64][mtiles][2][64], written
+// by conv_fwd_kernel<BNPART=true> while y was still in registers): one
+// wave per channel, same math as bn_colsum_finalize_kernel — BN's own
+// stage-1 pass over y never runs for conv-fed layers.
+__global__ void bn_conv_colsum_finalize_kernel(
+    const float* __restrict__ part, int mtiles, int C, long long count,
+    float eps, float momentum, bool track, float* __restrict__ running_mean,
+    float* __restrict__ running_var, float* __restrict__ save_mean,
+    float* __restrict__ save_invstd) {
+  int c = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  int lane = threadIdx.x & 63;
+  if (c >= C) return;
+  const float* base = part + (long long)(c >> 6) * mtiles * 128 + (c & 63);
+  float s = 0.f, sq = 0.f;
+  for (int b = lane; b < mtiles; b += 64) {
+    const float* row = base + (long long)b * 128;
+    s += row[0];
+    sq += row[64];
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) {
+    s += __shfl_down(s, off, 64);
+    sq += __shfl_down(sq, off, 64);
+  }
+  if (lane == 0) {
+    float mean = s / count;
+    float var = fmaxf(sq / count - mean * mean, 0.f);  // biased
+    if (track) {
+      float unbiased = count > 1 ? var * count / (count - 1) : var;
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    }
+    save_mean[c] = mean;
+    save_invstd[c] = rsqrtf(var + eps);
+  }
+}
+
 // normalize + scale/shift, with optional fused residual add (RES) and ELU
 // epilogue (the reference's `elu(bn(conv) [+ shortcut])` patterns,
 // simple_models.py:150-153) — one pass instead of bn/add/elu separate
@@ -350,7 +387,8 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
                                       at::Tensor running_var, bool training,
                                       double momentum, double eps,
                                       c10::optional<at::Tensor> residual,
-                                      bool elu) {
+                                      bool elu,
+                                      c10::optional<at::Tensor> conv_part) {
   check_nhwc(x);
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   long long M = (long long)N * H * W;
@@ -363,7 +401,23 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
   auto beta_f = beta.contiguous();
   TORCH_CHECK(gamma_f.scalar_type() == at::kFloat, "bn gamma must be fp32");
 
-  if (training) {
+  if (training && conv_part.has_value()) {
+    // stage 1 already happened inside the producing conv's epilogue
+    auto& cp = *conv_part;
+    TORCH_CHECK(cp.scalar_type() == at::kFloat && cp.dim() == 4 &&
+                cp.size(0) * 64 == C && cp.size(2) == 2 && cp.size(3) == 64,
+                "bad conv_part shape for C=", C);
+    bool track = running_mean.defined();
+    int nwaves = 4;
+    hipLaunchKernelGGL(bn_conv_colsum_finalize_kernel,
+                       dim3((C + nwaves - 1) / nwaves), dim3(256), 0, stream,
+                       cp.data_ptr<float>(), (int)cp.size(1), C, M,
+                       (float)eps, (float)momentum, track,
+                       track ? running_mean.data_ptr<float>() : nullptr,
+                       track ? running_var.data_ptr<float>() : nullptr,
+                       save_mean.data_ptr<float>(),
+                       save_invstd.data_ptr<float>());
+  } else if (training) {
     DISPATCH_F32_BF16(x, "bn_partials", {
       constexpr int VEC = 16 / sizeof(scalar_t);
       TORCH_CHECK(C % VEC == 0 && 256 % (C / VEC) == 0,

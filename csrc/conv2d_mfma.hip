@@ -49,13 +49,14 @@ __device__ __forceinline__ int lds_off(int row, int k) {
   return row * 128 + chunk * 16 + (k & 7) * 2;
 }
 
-// REGSTAGE: 0 = LDS-DMA (glds) staging, 1 = register staging (global
-// vector loads into VGPRs + ds_write_b128 committed right before the
-// barrier).  A glds instruction's ISSUE cost among MFMAs is ~60-185 cycles
-// (guide §LDS-DMA); at this kernel's small tiles (6 pieces per 16 MFMAs)
-// that is comparable to the MFMA work itself, so the cheap-issue register
-// path is worth an A/B (FEDKIT_CONV_REGSTAGE=0/1, default measured best).
-template <int BM, int STRIDE, int REGSTAGE>
+// (A register-staging variant was measured 2x SLOWER than glds staging
+// and removed — see profiles/r01_kernel_stats_final.md dead ends.)
+// BNPART: the conv feeds a BatchNorm, so the epilogue also emits this
+// workgroup's per-channel (sum, sumsq) partial row — BN's stage-1
+// reduction (a full extra read of y + a launch per layer) disappears.
+// Partials layout [Kout/64][mtiles][2][64] fp32, finalized by
+// bn_conv_colsum_finalize_kernel in batchnorm.hip.
+template <int BM, int STRIDE, bool BNPART>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      const bf16* __restrict__ w,   // [Kout][R*S*C]
@@ -64,7 +65,8 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      int R, int S, int P, int Q, int Kg, int dil,
                      int Ktrue /* <= Kout: K rows beyond are zero-padding
                                   (VAE/CPC channel counts), skipped on
-                                  store so y needs no unpad pass */) {
+                                  store so y needs no unpad pass */,
+                     float* __restrict__ bnpart) {
   constexpr int AB = BM * BK * 2;          // A tile bytes
   constexpr int BB = BN * BK * 2;          // B tile bytes
   __shared__ char smem[3 * (AB + BB)];     // 3-stage pipeline
@@ -124,9 +126,8 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   auto bufA = [&](int b) -> char* { return smem + b * (AB + BB); };
   auto bufB = [&](int b) -> char* { return smem + b * (AB + BB) + AB; };
 
-  bf16x8 areg[2][A_SLOTS], brg[2][B_SLOTS];
   const bool c64 = (C % 64) == 0;
-  auto stage = [&](int buf, int kt, int rset) {
+  auto stage = [&](int buf, int kt) {
     // A tile: per-slot source (r,s,c) from k_global; 16-B LDS-DMA.
     // LDS dest for a glds is wave-uniform base + lane*16: slot d = pass*256
     // + wave*64 + lane matches d = pass*256 + tid exactly.
@@ -156,29 +157,13 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
         int r = rs / S;
         src = xp + a_rowbase[i] + ((long long)r * dil * Wp + s * dil) * C + c;
       }
-      if (REGSTAGE)
-        __builtin_memcpy(&areg[rset][i], src, sizeof(bf16x8));
-      else
-        GLDS16(src, bufA(buf) + (i * 4 + wave) * 1024);
+      GLDS16(src, bufA(buf) + (i * 4 + wave) * 1024);
     }
 #pragma unroll
     for (int i = 0; i < B_SLOTS; ++i) {
       const bf16* src = w + b_rowbase[i] + kt * BK + b_k8[i] * 8;
-      if (REGSTAGE)
-        __builtin_memcpy(&brg[rset][i], src, sizeof(bf16x8));
-      else
-        GLDS16(src, bufB(buf) + (i * 4 + wave) * 1024);
+      GLDS16(src, bufB(buf) + (i * 4 + wave) * 1024);
     }
-  };
-  auto commit = [&](int buf, int rset) {
-#pragma unroll
-    for (int i = 0; i < A_SLOTS; ++i)
-      *(bf16x8*)(bufA(buf) + (i * 4 + wave) * 1024 + lane * 16) =
-          areg[rset][i];
-#pragma unroll
-    for (int i = 0; i < B_SLOTS; ++i)
-      *(bf16x8*)(bufB(buf) + (i * 4 + wave) * 1024 + lane * 16) =
-          brg[rset][i];
   };
 
   // ---- wave -> output sub-tile: 2x2 waves, wave tile (BM/2) x 32
@@ -201,20 +186,16 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   // single raw s_barrier does double duty: every wave's tile-kt loads have
   // landed, and every wave is done reading buffer (kt+2)%3 (used by kt-1).
   constexpr int LPS = A_SLOTS + B_SLOTS;
-  stage(0, 0, 0);
-  if (nkt > 1) stage(1, 1, 1);
+  stage(0, 0);
+  if (nkt > 1) stage(1, 1);
 
   for (int kt = 0; kt < nkt; ++kt) {
     if (kt + 1 < nkt)
       asm volatile("s_waitcnt vmcnt(%0)" ::"n"(LPS) : "memory");
     else
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    if (REGSTAGE) {
-      commit(kt % 3, kt % 2);
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    }
     __builtin_amdgcn_s_barrier();
-    if (kt + 2 < nkt) stage((kt + 2) % 3, kt + 2, kt % 2);
+    if (kt + 2 < nkt) stage((kt + 2) % 3, kt + 2);
     const char* A = bufA(kt % 3);
     const char* B = bufB(kt % 3);
 #pragma unroll
@@ -240,6 +221,7 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   // ---- epilogue: D fragment lane l holds col = l&15, rows (l>>4)*4 + v;
   // stores go at Ktrue stride and skip the zero-padded out channels
   const int col = bn * BN + wn + frag_row;
+  float bs[NR] = {}, bq[NR] = {};          // per-lane BN partials (2 cols)
 #pragma unroll
   for (int mfrag = 0; mfrag < MR; ++mfrag) {
 #pragma unroll
@@ -248,10 +230,53 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
       if (m < M) {
 #pragma unroll
         for (int nfrag = 0; nfrag < NR; ++nfrag)
-          if (col + nfrag * 16 < Ktrue)
-            y[m * Ktrue + col + nfrag * 16] =
-                __float2bfloat16(acc[mfrag][nfrag][v]);
+          if (col + nfrag * 16 < Ktrue) {
+            bf16 h = __float2bfloat16(acc[mfrag][nfrag][v]);
+            y[m * Ktrue + col + nfrag * 16] = h;
+            if (BNPART) {
+              float f = __bfloat162float(h);   // stats over the ROUNDED y
+              bs[nfrag] += f;
+              bq[nfrag] += f * f;
+            }
+          }
       }
+    }
+  }
+  if (BNPART) {
+    // channel totals: xor-reduce the 4 lane-groups sharing a column, then
+    // combine the wave pairs (wn 0: waves 0,2; wn 32: waves 1,3) via LDS
+#pragma unroll
+    for (int f = 0; f < NR; ++f) {
+#pragma unroll
+      for (int off = 16; off < 64; off <<= 1) {
+        bs[f] += __shfl_xor(bs[f], off, 64);
+        bq[f] += __shfl_xor(bq[f], off, 64);
+      }
+    }
+    float* red = (float*)smem;               // [4 waves][16 fr][4]
+    __syncthreads();                         // pipeline LDS reads done
+    if (lane < 16) {
+      float* r = red + (wave * 16 + frag_row) * 4;
+      r[0] = bs[0];
+      r[1] = bq[0];
+      r[2] = bs[1];
+      r[3] = bq[1];
+    }
+    __syncthreads();
+    // one thread per channel writes the workgroup's partial row
+    if (tid < 64) {
+      int grp = tid >> 5;                    // 0: wn=0 (waves 0,2), 1: wn=32
+      int nf = (tid >> 4) & 1;
+      int fr = tid & 15;
+      float s = red[((grp + 0) * 16 + fr) * 4 + 2 * nf]
+              + red[((grp + 2) * 16 + fr) * 4 + 2 * nf];
+      float q = red[((grp + 0) * 16 + fr) * 4 + 2 * nf + 1]
+              + red[((grp + 2) * 16 + fr) * 4 + 2 * nf + 1];
+      // layout [Kout/64 = bn][mtiles = gridDim.x][2][64]
+      float* out = bnpart +
+          (((long long)bn * gridDim.x + bm) * 2) * 64;
+      out[tid] = s;
+      out[64 + tid] = q;
     }
   }
 }
@@ -709,7 +734,8 @@ void check_conv_inputs(const at::Tensor& x, const at::Tensor& w) {
 // Ktrue < Kout means the trailing weight rows are channel zero-padding
 // (VAE/CPC shapes): y is allocated and stored dense at Ktrue channels.
 at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
-                     int stride, int P, int Q, int dil = 1, int Ktrue = -1) {
+                     int stride, int P, int Q, int dil = 1, int Ktrue = -1,
+                     float* bnpart = nullptr) {
   int N = xp.size(0), C = xp.size(1), Hp = xp.size(2), Wp = xp.size(3);
   int Kout = w_krs_c.size(0), R = w_krs_c.size(2), S = w_krs_c.size(3);
   int Kg = R * S * C;
@@ -725,27 +751,24 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
   bool bm64 = ((M + 127) / 128) * (Kout / BN) < 256;
   int BM = bm64 ? 64 : 128;
   dim3 grid((unsigned)((M + BM - 1) / BM), Kout / BN);
-  static const int regstage = []() {
-    const char* e = getenv("FEDKIT_CONV_REGSTAGE");
-    return e ? atoi(e) : 0;
-  }();
   auto L = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
                        (const bf16*)xp.data_ptr(),
                        (const bf16*)w_krs_c.data_ptr(), (bf16*)y.data_ptr(),
-                       N, Hp, Wp, C, Kout, R, S, P, Q, Kg, dil, Ktrue);
+                       N, Hp, Wp, C, Kout, R, S, P, Q, Kg, dil, Ktrue,
+                       bnpart);
   };
   TORCH_CHECK(stride == 1 || stride == 2, "conv kernel supports stride 1/2");
-  if (regstage) {
+  if (bnpart) {
     if (stride == 1)
-      bm64 ? L(conv_fwd_kernel<64, 1, 1>) : L(conv_fwd_kernel<128, 1, 1>);
+      bm64 ? L(conv_fwd_kernel<64, 1, true>) : L(conv_fwd_kernel<128, 1, true>);
     else
-      bm64 ? L(conv_fwd_kernel<64, 2, 1>) : L(conv_fwd_kernel<128, 2, 1>);
+      bm64 ? L(conv_fwd_kernel<64, 2, true>) : L(conv_fwd_kernel<128, 2, true>);
   } else {
     if (stride == 1)
-      bm64 ? L(conv_fwd_kernel<64, 1, 0>) : L(conv_fwd_kernel<128, 1, 0>);
+      bm64 ? L(conv_fwd_kernel<64, 1, false>) : L(conv_fwd_kernel<128, 1, false>);
     else
-      bm64 ? L(conv_fwd_kernel<64, 2, 0>) : L(conv_fwd_kernel<128, 2, 0>);
+      bm64 ? L(conv_fwd_kernel<64, 2, false>) : L(conv_fwd_kernel<128, 2, false>);
   }
   return y;
 }
@@ -785,6 +808,28 @@ at::Tensor fedkit_conv2d_fwd_prepadded(const at::Tensor& xp,
   int P = (Hp - Reff) / (int)stride + 1;
   int Q = (Wp - Seff) / (int)stride + 1;
   return conv_core(xp, w, (int)stride, P, Q, (int)dil, (int)ktrue);
+}
+
+// prepadded forward that ALSO emits the BatchNorm stage-1 partials
+// ([Kout/64][mtiles][2][64] fp32) from the epilogue registers — feeds
+// fedkit_bn_fwd's conv_part fast path (one launch and one full read of y
+// fewer per conv+BN pair).
+std::vector<at::Tensor> fedkit_conv2d_fwd_prepadded_bnstats(
+    const at::Tensor& xp, const at::Tensor& w, long stride) {
+  check_conv_inputs(xp, w);
+  int Hp = xp.size(2), Wp = xp.size(3);
+  int R = w.size(2), S = w.size(3);
+  int Kout = w.size(0);
+  int P = (Hp - R) / (int)stride + 1;
+  int Q = (Wp - S) / (int)stride + 1;
+  long long M = (long long)xp.size(0) * P * Q;
+  bool bm64 = ((M + 127) / 128) * (Kout / BN) < 256;
+  long long mtiles = (M + (bm64 ? 63 : 127)) / (bm64 ? 64 : 128);
+  auto part = at::empty({Kout / BN, mtiles, 2, 64},
+                        xp.options().dtype(at::kFloat));
+  auto y = conv_core(xp, w, (int)stride, P, Q, 1, -1,
+                     part.data_ptr<float>());
+  return {y, part};
 }
 
 at::Tensor fedkit_conv2d_fwd(const at::Tensor& x, const at::Tensor& w,

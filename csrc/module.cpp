@@ -31,7 +31,8 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
                                       at::Tensor running_var, bool training,
                                       double momentum, double eps,
                                       c10::optional<at::Tensor> residual,
-                                      bool elu);
+                                      bool elu,
+                                      c10::optional<at::Tensor> conv_part);
 std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& gamma,
                                       const at::Tensor& save_mean,
@@ -47,6 +48,8 @@ at::Tensor fedkit_dilate_pad(const at::Tensor& x, long pt, long pb, long pl,
 at::Tensor fedkit_conv2d_fwd_prepadded(const at::Tensor& xp,
                                        const at::Tensor& w, long stride,
                                        long dil, long ktrue);
+std::vector<at::Tensor> fedkit_conv2d_fwd_prepadded_bnstats(
+    const at::Tensor& xp, const at::Tensor& w, long stride);
 at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
                                               const at::Tensor& xp,
                                               long stride, long R, long S,
@@ -79,7 +82,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("gamma"), py::arg("beta"),
         py::arg("running_mean"), py::arg("running_var"), py::arg("training"),
         py::arg("momentum"), py::arg("eps"),
-        py::arg("residual") = c10::nullopt, py::arg("elu") = false);
+        py::arg("residual") = c10::nullopt, py::arg("elu") = false,
+        py::arg("conv_part") = c10::nullopt);
   m.def("bn_bwd", &fedkit_bn_bwd,
         "NHWC BatchNorm bwd (optionally fused with ELU backward from the "
         "saved output): returns (gx, gw, gb[, g])",
@@ -92,6 +96,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("w"), py::arg("stride"), py::arg("padding"),
         py::arg("dil") = 1, py::arg("ktrue") = -1);
   m.def("conv2d_pad_input", &fedkit_conv2d_pad_input, "zero-pad NHWC input");
+  m.def("conv2d_fwd_prepadded_bnstats", &fedkit_conv2d_fwd_prepadded_bnstats,
+        "conv fwd that also emits BatchNorm stage-1 partials from the "
+        "epilogue registers: returns (y, part)");
   m.def("dilate_pad", &fedkit_dilate_pad,
         "NHWC zero-insert dilation + border pad (transposed-conv input)");
   m.def("conv2d_fwd_prepadded", &fedkit_conv2d_fwd_prepadded,

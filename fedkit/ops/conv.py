@@ -41,23 +41,29 @@ class _ConvFn(torch.autograd.Function):
         # x: [N,C,H,W] logical, carried NHWC-contiguous; w: [K,C,R,S].
         # The padded input is computed ONCE and saved, so bwd-weight reuses
         # it instead of re-padding (and x itself need not be kept).
+        # On the prepadded path the epilogue also emits the BatchNorm
+        # stage-1 partials (every FedConv2d output feeds a BN in the
+        # flagship models); the downstream FedBatchNorm2d picks them up
+        # off the output tensor and skips its own reduction pass.
         ext = _ext()
         small_c = x.shape[1] % 8 != 0
         if small_c:
             y = ext.conv2d_fwd(x, w, stride, padding)
+            part = y.new_empty(0)
             ctx.save_for_backward(x, w)
         else:
             xp = ext.conv2d_pad_input(x, padding)
-            y = ext.conv2d_fwd_prepadded(xp, w, stride)
+            y, part = ext.conv2d_fwd_prepadded_bnstats(xp, w, stride)
             ctx.save_for_backward(xp, w)
         ctx.small_c = small_c
         ctx.stride = stride
         ctx.padding = padding
         ctx.hw = (x.shape[2], x.shape[3])
-        return y
+        ctx.mark_non_differentiable(part)
+        return y, part
 
     @staticmethod
-    def backward(ctx, gy):
+    def backward(ctx, gy, _gpart):
         xsaved, w = ctx.saved_tensors
         gy = gy.contiguous(memory_format=torch.channels_last)
         gx = gw = None
@@ -271,7 +277,10 @@ class FedConv2d(nn.Conv2d):
                     self, "_w16", self.weight,
                     lambda: self.weight.to(torch.bfloat16).contiguous(
                         memory_format=torch.channels_last))
-                return _ConvFn.apply(x, w, self.stride[0], self.padding[0])
+                y, part = _ConvFn.apply(x, w, self.stride[0], self.padding[0])
+                if self.training and part.numel():
+                    y._fedkit_bn_stats = part
+                return y
         return F.conv2d(x, self.weight.to(x.dtype),
                         self.bias.to(x.dtype) if self.bias is not None else None,
                         self.stride, self.padding, self.dilation, self.groups)

@@ -33,10 +33,11 @@ def _ext():
 class _BnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
-                training, momentum, eps):
+                training, momentum, eps, conv_part=None):
         y, save_mean, save_invstd = _ext().bn_fwd(
             x, weight, bias, running_mean, running_var,
-            bool(training), float(momentum), float(eps))
+            bool(training), float(momentum), float(eps),
+            conv_part=conv_part)
         ctx.save_for_backward(x, weight, save_mean, save_invstd)
         return y
 
@@ -45,7 +46,7 @@ class _BnFn(torch.autograd.Function):
         x, weight, save_mean, save_invstd = ctx.saved_tensors
         gy = gy.contiguous(memory_format=torch.channels_last)
         gx, gw, gb = _ext().bn_bwd(gy, x, weight, save_mean, save_invstd)
-        return gx, gw, gb, None, None, None, None, None
+        return gx, gw, gb, None, None, None, None, None, None
 
 
 class _BnActFn(torch.autograd.Function):
@@ -59,11 +60,11 @@ class _BnActFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
-                training, momentum, eps, residual):
+                training, momentum, eps, residual, conv_part=None):
         y, save_mean, save_invstd = _ext().bn_fwd(
             x, weight, bias, running_mean, running_var,
             bool(training), float(momentum), float(eps),
-            residual=residual, elu=True)
+            residual=residual, elu=True, conv_part=conv_part)
         ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
         ctx.has_res = residual is not None
         return y
@@ -78,17 +79,20 @@ class _BnActFn(torch.autograd.Function):
         out = _ext().bn_bwd(gy, x, weight, save_mean, save_invstd,
                             elu_y=y, want_g=ctx.has_res)
         gres = out[3] if ctx.has_res else None
-        return out[0], out[1], out[2], None, None, None, None, None, gres
+        return (out[0], out[1], out[2], None, None, None, None, None, gres,
+                None)
 
 
 class FedBatchNorm2d(nn.BatchNorm2d):
     def forward(self, x):
         if _native(x):
             self._prep(x)
+            part = getattr(x, "_fedkit_bn_stats", None) \
+                if self.training else None
             x = x.contiguous(memory_format=torch.channels_last)
             return _BnFn.apply(x, self.weight, self.bias,
                                self.running_mean, self.running_var,
-                               self.training, self.momentum, self.eps)
+                               self.training, self.momentum, self.eps, part)
         return super().forward(x)
 
     def _prep(self, x):
@@ -125,6 +129,7 @@ def bn_elu(bn: FedBatchNorm2d, x: torch.Tensor,
     """elu(bn(x) [+ residual]) — fused on GPU, composed on CPU."""
     if _native(x):
         bn._prep(x)
+        part = getattr(x, "_fedkit_bn_stats", None) if bn.training else None
         x = x.contiguous(memory_format=torch.channels_last)
         if residual is not None:
             residual = residual.contiguous(memory_format=torch.channels_last)
@@ -132,7 +137,8 @@ def bn_elu(bn: FedBatchNorm2d, x: torch.Tensor,
                 residual = residual.to(x.dtype)
         return _BnActFn.apply(x, bn.weight, bn.bias,
                               bn.running_mean, bn.running_var,
-                              bn.training, bn.momentum, bn.eps, residual)
+                              bn.training, bn.momentum, bn.eps, residual,
+                              part)
     y = nn.BatchNorm2d.forward(bn, x)
     if residual is not None:
         y = y + residual
