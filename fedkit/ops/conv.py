@@ -20,6 +20,8 @@ bwd-weight:
 CPU path: stock F.conv2d (tests compare the HIP kernels against this).
 """
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -41,19 +43,27 @@ class _ConvFn(torch.autograd.Function):
         # x: [N,C,H,W] logical, carried NHWC-contiguous; w: [K,C,R,S].
         # The padded input is computed ONCE and saved, so bwd-weight reuses
         # it instead of re-padding (and x itself need not be kept).
-        # On the prepadded path the epilogue also emits the BatchNorm
-        # stage-1 partials (every FedConv2d output feeds a BN in the
-        # flagship models); the downstream FedBatchNorm2d picks them up
-        # off the output tensor and skips its own reduction pass.
+        # FEDKIT_CONV_BNSTATS=1: the epilogue also emits the BatchNorm
+        # stage-1 partials and the downstream FedBatchNorm2d skips its own
+        # reduction pass.  MEASURED NET LOSS at CIFAR sizes (step 3.9 ->
+        # ~4.9 ms): the per-channel finalize must then read one partial
+        # row per conv workgroup with XCD-crossing strided loads, which
+        # costs more than the single streaming pass it replaces — kept as
+        # a documented experiment for larger-image workloads.
         ext = _ext()
         small_c = x.shape[1] % 8 != 0
+        want_stats = os.environ.get("FEDKIT_CONV_BNSTATS") == "1"
         if small_c:
             y = ext.conv2d_fwd(x, w, stride, padding)
             part = y.new_empty(0)
             ctx.save_for_backward(x, w)
         else:
             xp = ext.conv2d_pad_input(x, padding)
-            y, part = ext.conv2d_fwd_prepadded_bnstats(xp, w, stride)
+            if want_stats:
+                y, part = ext.conv2d_fwd_prepadded_bnstats(xp, w, stride)
+            else:
+                y = ext.conv2d_fwd_prepadded(xp, w, stride)
+                part = y.new_empty(0)
             ctx.save_for_backward(xp, w)
         ctx.small_c = small_c
         ctx.stride = stride
