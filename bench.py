@@ -111,8 +111,12 @@ def main():
 
     step_i = 0
 
-    def eager_step(x, y):
-        opt.zero_grad(set_to_none=False)
+    def eager_step(x, y, keep_grad_buffers=False):
+        # set_to_none=True matters eagerly: with persistent grad buffers
+        # every step pays one fill + one accumulate-add kernel PER PARAM
+        # (~120 launches, ~350 us on ResNet18); graph capture needs the
+        # buffers kept
+        opt.zero_grad(set_to_none=not keep_grad_buffers)
         with autocast():
             loss = L.cross_entropy(net(x), y)
         loss.backward()
@@ -131,11 +135,11 @@ def main():
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
                 for _ in range(3):
-                    eager_step(static_x, static_y)
+                    eager_step(static_x, static_y, keep_grad_buffers=True)
             torch.cuda.current_stream().wait_stream(side)
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
-                eager_step(static_x, static_y)
+                eager_step(static_x, static_y, keep_grad_buffers=True)
         except Exception as e:  # pragma: no cover - capture support varies
             print(f"[bench] hipGraph capture unavailable ({e}); eager path",
                   file=sys.stderr)
