@@ -6,6 +6,12 @@
 #include <vector>
 
 at::Tensor fedkit_elu_fwd(const at::Tensor& x);
+std::vector<at::Tensor> fedkit_max_pool2d_fwd(const at::Tensor& x, long k);
+at::Tensor fedkit_max_pool2d_bwd(const at::Tensor& gy, const at::Tensor& idx,
+                                 long k, long H, long W);
+at::Tensor fedkit_avg_pool2d_fwd(const at::Tensor& x, long k);
+at::Tensor fedkit_avg_pool2d_bwd(const at::Tensor& gy, long k, long H,
+                                 long W);
 at::Tensor fedkit_elu_bwd(const at::Tensor& gy, const at::Tensor& y);
 
 void fedkit_pack_params(std::vector<at::Tensor> tensors, at::Tensor flat);
@@ -64,6 +70,12 @@ at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "fedkit hand-written CDNA4 (gfx950 / MI355X) kernels";
   m.def("elu_fwd", &fedkit_elu_fwd, "ELU forward (vectorized)");
+  m.def("max_pool2d_fwd", &fedkit_max_pool2d_fwd,
+        "NHWC non-overlapping max pool: returns (y, argmax_idx)");
+  m.def("max_pool2d_bwd", &fedkit_max_pool2d_bwd, "max pool backward");
+  m.def("avg_pool2d_fwd", &fedkit_avg_pool2d_fwd,
+        "NHWC non-overlapping average pool");
+  m.def("avg_pool2d_bwd", &fedkit_avg_pool2d_bwd, "avg pool backward");
   m.def("elu_bwd", &fedkit_elu_bwd, "ELU backward from saved output");
   m.def("pack_params", &fedkit_pack_params, "multi-tensor -> flat fp32");
   m.def("unpack_params", &fedkit_unpack_params, "flat fp32 -> multi-tensor");

@@ -13,6 +13,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.elu import elu
+from ..ops.pool import FedMaxPool2d
 
 
 class Net(nn.Module):
@@ -21,7 +22,7 @@ class Net(nn.Module):
     def __init__(self):
         super().__init__()
         self.conv1 = nn.Conv2d(3, 6, 5)
-        self.pool = nn.MaxPool2d(2, 2)
+        self.pool = FedMaxPool2d(2, 2)
         self.conv2 = nn.Conv2d(6, 16, 5)
         self.fc1 = nn.Linear(16 * 5 * 5, 120)
         self.fc2 = nn.Linear(120, 84)
@@ -61,8 +62,8 @@ class Net1(nn.Module):
         self.conv2 = nn.Conv2d(32, 32, 3)
         self.conv3 = nn.Conv2d(32, 64, 3)
         self.conv4 = nn.Conv2d(64, 64, 3)
-        self.pool1 = nn.MaxPool2d(2, 2)
-        self.pool2 = nn.MaxPool2d(2, 2)
+        self.pool1 = FedMaxPool2d(2, 2)
+        self.pool2 = FedMaxPool2d(2, 2)
         self.fc1 = nn.Linear(64 * 5 * 5, 512)
         self.fc2 = nn.Linear(512, 10)
 
@@ -95,10 +96,10 @@ class Net2(nn.Module):
         self.conv2 = nn.Conv2d(64, 128, 3, padding=1)
         self.conv3 = nn.Conv2d(128, 256, 3, padding=1)
         self.conv4 = nn.Conv2d(256, 512, 3, padding=1)
-        self.pool1 = nn.MaxPool2d(2, 2)
-        self.pool2 = nn.MaxPool2d(2, 2)
-        self.pool3 = nn.MaxPool2d(2, 2)
-        self.pool4 = nn.MaxPool2d(2, 2)
+        self.pool1 = FedMaxPool2d(2, 2)
+        self.pool2 = FedMaxPool2d(2, 2)
+        self.pool3 = FedMaxPool2d(2, 2)
+        self.pool4 = FedMaxPool2d(2, 2)
         self.fc1 = nn.Linear(512 * 2 * 2, 128)
         self.fc2 = nn.Linear(128, 256)
         self.fc3 = nn.Linear(256, 512)

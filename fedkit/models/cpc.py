@@ -12,6 +12,7 @@ import torch
 import torch.nn as nn
 
 from ..ops.conv import FedConvGeneric
+from ..ops.pool import avg_pool2d
 import torch.nn.functional as F
 
 from ..ops.elu import elu
@@ -38,7 +39,7 @@ class EncoderCNN(nn.Module):
         h = elu(self.conv2(bank))
         h = elu(self.conv3(h))
         h = elu(self.conv4(h))
-        return F.avg_pool2d(h, 2).squeeze()
+        return avg_pool2d(h, 2).squeeze()
 
     def train_order_block_ids(self):
         return [[0, 9], [10, 15]]

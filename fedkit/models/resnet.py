@@ -18,6 +18,7 @@ partition) is identical to the reference's.
 import torch.nn as nn
 
 from ..ops.conv import FedConv2d
+from ..ops.pool import avg_pool2d
 from ..ops.norm import FedBatchNorm2d, bn_elu
 
 
@@ -102,7 +103,7 @@ class ResNet(nn.Module):
         out = self.layer2(out)
         out = self.layer3(out)
         out = self.layer4(out)
-        out = nn.functional.avg_pool2d(out, 4)
+        out = avg_pool2d(out, 4)
         out = out.reshape(out.size(0), -1)
         return self.linear(out)
 

@@ -75,10 +75,12 @@ def native_enabled(t: torch.Tensor) -> bool:
 from . import elu, flat, losses          # noqa: E402,F401
 from .conv import (FedConv2d, FedConvGeneric,  # noqa: E402
                    FedConvTranspose2d)
+from .pool import FedMaxPool2d, max_pool2d, avg_pool2d  # noqa: E402
 from .norm import FedBatchNorm2d         # noqa: E402
 
 __all__ = [
     "ext", "has_ext", "require_ext", "native_enabled",
     "elu", "flat", "losses", "FedConv2d", "FedConvGeneric",
+    "FedMaxPool2d", "max_pool2d", "avg_pool2d",
     "FedConvTranspose2d", "FedBatchNorm2d",
 ]

@@ -22,6 +22,7 @@ SOURCES = [
     "csrc/batchnorm.hip",
     "csrc/conv2d_mfma.hip",
     "csrc/conv_small.hip",
+    "csrc/pool.hip",
 ]
 
 setup(

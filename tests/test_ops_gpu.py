@@ -484,3 +484,88 @@ def test_lbfgs_gpu_quadratic_converges():
     xstar = torch.linalg.solve(Q, b)
     rel = (x.detach() - xstar).norm() / xstar.norm()
     assert rel < 1e-2, float(rel)
+
+
+# ------------------------------------------------------------------- pooling
+
+@pytest.mark.parametrize("C,H,k", [(64, 32, 2), (16, 8, 2), (512, 4, 4),
+                                   (96, 16, 2)])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_pool_fwd_bwd_vs_torch(C, H, k, dtype):
+    from fedkit.ops.pool import max_pool2d, avg_pool2d
+    torch.manual_seed(31)
+    x = torch.randn(8, C, H, H, device="cuda", dtype=dtype,
+                    requires_grad=True)
+    for fed, ref in ((max_pool2d, F.max_pool2d), (avg_pool2d, F.avg_pool2d)):
+        y = fed(x, k)
+        want = ref(x.float(), k)
+        assert y.shape == want.shape
+        assert frob_err(y.float(), want) < 1e-2
+        gy = torch.randn_like(want)
+        gx, = torch.autograd.grad(y, x, gy.to(y.dtype), retain_graph=False)
+        gxr, = torch.autograd.grad(want, x, gy, retain_graph=False)
+        assert frob_err(gx.float(), gxr.float()) < 1e-2
+
+
+# ------------------------------------------------- driver-level GPU smokes
+
+def _smoke_cfg(**kw):
+    from fedkit.parallel import FedConfig
+    return FedConfig(K=2, default_batch=32, Nloop=1, Nepoch=1, Nadmm=1,
+                     check_results=False, save_model=False, use_cuda=True,
+                     max_steps_per_epoch=2, be_verbose=False, **kw)
+
+
+def test_vae_job_gpu_smoke():
+    """federated_vae path: AutoEncoderCNN + vae_loss on the generic convs."""
+    from fedkit.parallel import FederatedJob
+    from fedkit.ops.losses import vae_loss
+
+    def loss_fn(net, images, _labels):
+        out, mu, logvar = net(images)
+        return vae_loss(out.float(), images.float(), mu.float(),
+                        logvar.float())
+
+    job = FederatedJob(_smoke_cfg(model="AutoEncoderCNN", strategy="fedavg",
+                                  per_layer=True), loss_fn=loss_fn)
+    job.run()
+    for ck in job.comm.my_clients:
+        for p in job.nets[ck].parameters():
+            assert torch.isfinite(p).all()
+
+
+def test_vae_cl_job_gpu_smoke():
+    """federated_vae_cl path: clustering VAE with the 4-term ELBO."""
+    from fedkit.parallel import FederatedJob
+    from fedkit.ops.losses import vaecl_loss
+
+    def loss_fn(net, images, _labels):
+        ekhat, mu_xi, sig2_xi, mu_b, sig2_b, mu_th, sig2_th = net(images)
+        return vaecl_loss(ekhat, mu_xi, sig2_xi, mu_b, sig2_b, mu_th,
+                          sig2_th, images)
+
+    job = FederatedJob(_smoke_cfg(model="AutoEncoderCNNCL",
+                                  strategy="fedavg"), loss_fn=loss_fn)
+    job.run()
+    for ck in job.comm.my_clients:
+        for p in job.nets[ck].parameters():
+            assert torch.isfinite(p).all()
+
+
+def test_cpc_models_gpu_smoke():
+    """CPC encoder/contextgen/predictor + InfoNCE end-to-end on GPU."""
+    from fedkit.models import EncoderCNN, ContextgenCNN, PredictorCNN
+    from fedkit.ops.losses import info_nce
+    torch.manual_seed(33)
+    enc = EncoderCNN().cuda()
+    ctx = ContextgenCNN().cuda()
+    prd = PredictorCNN().cuda()
+    ybatch = torch.randn(9, 8, 32, 32, device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        lat = enc(ybatch)                      # [9, latent]
+        grid = lat.t().reshape(1, -1, 3, 3)    # [1, latent, px, py]
+        c = ctx(grid)
+        zhat, z = prd(grid, c)
+        loss = info_nce(z.float(), zhat.float())
+    loss.backward()
+    assert torch.isfinite(loss)
