@@ -865,11 +865,10 @@ at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
   // round the wrot rows up to the 64-row tile with zeros and store dx
   // dense at ctrue via the kernel's Ktrue guard.
   int Cpad = (C + 63) / 64 * 64;
-  auto wrot = Cpad == C
-      ? at::empty({C, K, R, S},
-                  w.options().memory_format(at::MemoryFormat::ChannelsLast))
-      : at::zeros({Cpad, K, R, S},
-                  w.options().memory_format(at::MemoryFormat::ChannelsLast));
+  // (empty + zero_: at::zeros drops the memory_format in options)
+  auto wrot = at::empty({Cpad, K, R, S},
+                        w.options().memory_format(at::MemoryFormat::ChannelsLast));
+  if (Cpad != C) wrot.zero_();
   {
     auto stream = fedkit_stream();
     long long total = (long long)K * R * S * C;

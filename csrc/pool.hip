@@ -189,8 +189,11 @@ at::Tensor fedkit_max_pool2d_bwd(const at::Tensor& gy, const at::Tensor& idx,
                                  long k, long H, long W) {
   check_pool(gy, 1);
   int N = gy.size(0), C = gy.size(1), P = gy.size(2), Q = gy.size(3);
-  auto gx = at::zeros({N, C, (long)H, (long)W},
+  // at::zeros IGNORES memory_format in options (empty honors it) —
+  // empty + zero_ keeps the channels_last layout the kernel indexes
+  auto gx = at::empty({N, C, (long)H, (long)W},
                       gy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  gx.zero_();
   auto stream = fedkit_stream();
   POOL_DISPATCH(gy, "max_pool_bwd", {
     long long total = (long long)N * P * Q * C / VEC;
