@@ -51,12 +51,14 @@ __device__ __forceinline__ int lds_off(int row, int k) {
 
 // (A register-staging variant was measured 2x SLOWER than glds staging
 // and removed — see profiles/r01_kernel_stats_final.md dead ends.)
-// BNPART: the conv feeds a BatchNorm, so the epilogue also emits this
-// workgroup's per-channel (sum, sumsq) partial row — BN's stage-1
-// reduction (a full extra read of y + a launch per layer) disappears.
-// Partials layout [Kout/64][mtiles][2][64] fp32, finalized by
-// bn_conv_colsum_finalize_kernel in batchnorm.hip.
-template <int BM, int STRIDE, bool BNPART>
+// MODE 0: plain store.  MODE 1 (BN stats): the epilogue also emits this
+// workgroup's per-channel (sum, sumsq) partial row into aux
+// ([Kout/64][mtiles][2][64] fp32, finalized by
+// bn_conv_colsum_finalize_kernel).  MODE 2 (split-K): grid.z slices the
+// Kg loop and the epilogue writes fp32 partials aux[z][M][Kout] —
+// for shapes whose (M, Kout) grid alone cannot fill 256 CUs
+// (layer4: 2048x512 -> 256 workgroups, measured 37 us at 12% MFMA).
+template <int BM, int STRIDE, int MODE>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      const bf16* __restrict__ w,   // [Kout][R*S*C]
@@ -66,7 +68,7 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      int Ktrue /* <= Kout: K rows beyond are zero-padding
                                   (VAE/CPC channel counts), skipped on
                                   store so y needs no unpad pass */,
-                     float* __restrict__ bnpart) {
+                     float* __restrict__ aux) {
   constexpr int AB = BM * BK * 2;          // A tile bytes
   constexpr int BB = BN * BK * 2;          // B tile bytes
   __shared__ char smem[3 * (AB + BB)];     // 3-stage pipeline
@@ -177,7 +179,12 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   const int frag_row = lane & 15;          // fragment row/col within 16
   const int frag_k = (lane >> 4) * 8;      // 8 contiguous k per lane
 
-  const int nkt = Kg / BK;
+  int kt0 = 0, nkt = Kg / BK;
+  if (MODE == 2) {
+    int kps = (nkt + gridDim.z - 1) / gridDim.z;
+    kt0 = blockIdx.z * kps;
+    nkt = min(kps, Kg / BK - kt0);
+  }
   // 3-stage software pipeline: tiles kt and kt+1 are in flight on entry to
   // iteration kt; kt+2 is issued right after the barrier.  The wait is a
   // MANUAL vmcnt(LPS) (LPS = this thread's GLDS ops per stage) so only tile
@@ -186,8 +193,8 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   // single raw s_barrier does double duty: every wave's tile-kt loads have
   // landed, and every wave is done reading buffer (kt+2)%3 (used by kt-1).
   constexpr int LPS = A_SLOTS + B_SLOTS;
-  stage(0, 0);
-  if (nkt > 1) stage(1, 1);
+  if (nkt > 0) stage(0, kt0);
+  if (nkt > 1) stage(1, kt0 + 1);
 
   for (int kt = 0; kt < nkt; ++kt) {
     if (kt + 1 < nkt)
@@ -195,7 +202,7 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
     else
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    if (kt + 2 < nkt) stage((kt + 2) % 3, kt + 2);
+    if (kt + 2 < nkt) stage((kt + 2) % 3, kt0 + kt + 2);
     const char* A = bufA(kt % 3);
     const char* B = bufB(kt % 3);
 #pragma unroll
@@ -228,12 +235,19 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
     for (int v = 0; v < 4; ++v) {
       long long m = (long long)bm * BM + wm + mfrag * 16 + (lane >> 4) * 4 + v;
       if (m < M) {
+        if (MODE == 2) {
+          float* out = aux + ((long long)blockIdx.z * M + m) * Kout + col;
+#pragma unroll
+          for (int nfrag = 0; nfrag < NR; ++nfrag)
+            out[nfrag * 16] = acc[mfrag][nfrag][v];
+          continue;
+        }
 #pragma unroll
         for (int nfrag = 0; nfrag < NR; ++nfrag)
           if (col + nfrag * 16 < Ktrue) {
             bf16 h = __float2bfloat16(acc[mfrag][nfrag][v]);
             y[m * Ktrue + col + nfrag * 16] = h;
-            if (BNPART) {
+            if (MODE == 1) {
               float f = __bfloat162float(h);   // stats over the ROUNDED y
               bs[nfrag] += f;
               bq[nfrag] += f * f;
@@ -242,7 +256,7 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
       }
     }
   }
-  if (BNPART) {
+  if (MODE == 1) {
     // channel totals: xor-reduce the 4 lane-groups sharing a column, then
     // combine the wave pairs (wn 0: waves 0,2; wn 32: waves 1,3) via LDS
 #pragma unroll
@@ -273,7 +287,7 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
       float q = red[((grp + 0) * 16 + fr) * 4 + 2 * nf + 1]
               + red[((grp + 2) * 16 + fr) * 4 + 2 * nf + 1];
       // layout [Kout/64 = bn][mtiles = gridDim.x][2][64]
-      float* out = bnpart +
+      float* out = aux +
           (((long long)bn * gridDim.x + bm) * 2) * 64;
       out[tid] = s;
       out[64 + tid] = q;
@@ -751,24 +765,51 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
   bool bm64 = ((M + 127) / 128) * (Kout / BN) < 256;
   int BM = bm64 ? 64 : 128;
   dim3 grid((unsigned)((M + BM - 1) / BM), Kout / BN);
+  TORCH_CHECK(stride == 1 || stride == 2, "conv kernel supports stride 1/2");
+  // split-K when the (M, Kout) grid alone cannot fill the 256 CUs and the
+  // Kg loop is deep enough to slice (layer4-class shapes)
+  int gridxy = (int)grid.x * (int)grid.y;
+  int nkt_total = Kg / BK;
+  int splits = 1;
+  if (!bnpart && Ktrue == Kout && gridxy <= 256) {
+    while (splits < 8 && gridxy * splits < 512 &&
+           nkt_total / (splits * 2) >= 4)
+      splits *= 2;
+  }
+  at::Tensor part;
+  float* aux = bnpart;
+  if (splits > 1) {
+    part = at::empty({splits, M, (long long)Kout},
+                     xp.options().dtype(at::kFloat));
+    aux = part.data_ptr<float>();
+    grid.z = splits;
+  }
   auto L = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
                        (const bf16*)xp.data_ptr(),
                        (const bf16*)w_krs_c.data_ptr(), (bf16*)y.data_ptr(),
                        N, Hp, Wp, C, Kout, R, S, P, Q, Kg, dil, Ktrue,
-                       bnpart);
+                       aux);
   };
-  TORCH_CHECK(stride == 1 || stride == 2, "conv kernel supports stride 1/2");
-  if (bnpart) {
+  if (splits > 1) {
     if (stride == 1)
-      bm64 ? L(conv_fwd_kernel<64, 1, true>) : L(conv_fwd_kernel<128, 1, true>);
+      bm64 ? L(conv_fwd_kernel<64, 1, 2>) : L(conv_fwd_kernel<128, 1, 2>);
     else
-      bm64 ? L(conv_fwd_kernel<64, 2, true>) : L(conv_fwd_kernel<128, 2, true>);
+      bm64 ? L(conv_fwd_kernel<64, 2, 2>) : L(conv_fwd_kernel<128, 2, 2>);
+    long long Ly = M * Kout;
+    hipLaunchKernelGGL(colsum_to_bf16_kernel, dim3(grid_1d(Ly, 256)),
+                       dim3(256), 0, stream, part.data_ptr<float>(), splits,
+                       Ly, (bf16*)y.data_ptr());
+  } else if (bnpart) {
+    if (stride == 1)
+      bm64 ? L(conv_fwd_kernel<64, 1, 1>) : L(conv_fwd_kernel<128, 1, 1>);
+    else
+      bm64 ? L(conv_fwd_kernel<64, 2, 1>) : L(conv_fwd_kernel<128, 2, 1>);
   } else {
     if (stride == 1)
-      bm64 ? L(conv_fwd_kernel<64, 1, false>) : L(conv_fwd_kernel<128, 1, false>);
+      bm64 ? L(conv_fwd_kernel<64, 1, 0>) : L(conv_fwd_kernel<128, 1, 0>);
     else
-      bm64 ? L(conv_fwd_kernel<64, 2, false>) : L(conv_fwd_kernel<128, 2, false>);
+      bm64 ? L(conv_fwd_kernel<64, 2, 0>) : L(conv_fwd_kernel<128, 2, 0>);
   }
   return y;
 }
@@ -883,6 +924,17 @@ at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
                          (bf16*)wrot.data_ptr(), K, R, S, C);
   }
   if (ctrue < 0) ctrue = C;
+  if (R == 1 && S == 1 && stride == 2 && padding == 0) {
+    // 1x1 stride-2 shortcut dx: only every other pixel is nonzero, so the
+    // dilate-then-conv form runs the GEMM at 4x the useful M.  Instead:
+    // dense GEMM at the SMALL resolution, then one zero-interleave pass
+    // (dx[2p][2q] = sum_k gy[p][q][k] w[k][c]).
+    int P = gy.size(2), Q = gy.size(3);
+    auto dxs = conv_core(gy, wrot, 1, P, Q, 1, (int)ctrue);
+    int pb = (int)H - ((P - 1) * 2 + 1);
+    int pr = (int)W - ((Q - 1) * 2 + 1);
+    return dilate_pad_nhwc(dxs, 0, pb, 0, pr, 2);
+  }
   // dilate+pad gy: pl = (R-1)*dil - pad, pr = pl + a with
   // a = (H + 2p - Reff) % stride, Reff = (R-1)*dil + 1
   int Reff = (R - 1) * (int)dil + 1;
