@@ -22,6 +22,10 @@ at::Tensor fedkit_multi_dot(std::vector<at::Tensor> vecs, const at::Tensor& x);
 at::Tensor fedkit_lincomb(const at::Tensor& g, double cg,
                           std::vector<at::Tensor> vecs,
                           std::vector<double> coeffs);
+void fedkit_cast_f32_to_bf16(std::vector<at::Tensor> srcs,
+                             std::vector<at::Tensor> dsts);
+void fedkit_cast_bf16_to_f32(std::vector<at::Tensor> srcs,
+                             std::vector<at::Tensor> dsts);
 
 std::vector<at::Tensor> fedkit_cross_entropy_fwd(const at::Tensor& logits,
                                                  const at::Tensor& labels);
@@ -85,6 +89,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "x . vecs[i] for up to 24 vectors in one pass -> device fp32 [n]");
   m.def("lincomb", &fedkit_lincomb,
         "cg*g + sum c[i]*vecs[i] in one pass (L-BFGS direction build)");
+  m.def("cast_f32_to_bf16", &fedkit_cast_f32_to_bf16,
+        "batched fp32->bf16 tensor casts (one kernel for <=48 tensors)");
+  m.def("cast_bf16_to_f32", &fedkit_cast_bf16_to_f32,
+        "batched bf16->fp32 tensor casts");
   m.def("cross_entropy_fwd", &fedkit_cross_entropy_fwd,
         "fused log-softmax + NLL (mean): returns (loss, lse)");
   m.def("cross_entropy_bwd", &fedkit_cross_entropy_bwd, "CE backward");

@@ -115,6 +115,44 @@ def _cached_frozen(mod, key, src_param, make):
     return out
 
 
+class _BatchCastFn(torch.autograd.Function):
+    """Fused fp32->bf16 cast of MANY weight tensors in one kernel, grads
+    cast back in one kernel (csrc/flat_ops.hip cast tables).  A full-model
+    training step otherwise pays one cast launch per conv each way
+    (~40 launches, ~200 us/step on ResNet18)."""
+
+    @staticmethod
+    def forward(ctx, *ws):
+        outs = tuple(torch.empty_like(w, dtype=torch.bfloat16) for w in ws)
+        _ext().cast_f32_to_bf16(list(ws), list(outs))
+        return outs
+
+    @staticmethod
+    def backward(ctx, *gys):
+        outs, srcs, dsts = [], [], []
+        for g in gys:
+            if g is None:
+                outs.append(None)
+            else:
+                f = torch.empty_like(g, dtype=torch.float32)
+                srcs.append(g)
+                dsts.append(f)
+                outs.append(f)
+        if srcs:
+            _ext().cast_bf16_to_f32(srcs, dsts)
+        return tuple(outs)
+
+
+def batch_cast_weights(mods):
+    """One-kernel bf16 cast of the given modules' weights; each module's
+    next native forward consumes its copy (version-checked one-shot)."""
+    if not mods:
+        return
+    outs = _BatchCastFn.apply(*[m.weight for m in mods])
+    for m, o in zip(mods, outs):
+        m._w16_once = (m.weight._version, o)
+
+
 class _GenConvFn(torch.autograd.Function):
     """Generic conv on the MFMA kernels: any in/out channel count via zero
     channel padding (in -> mult of 8, out -> mult of 64, with the kernel
@@ -283,10 +321,14 @@ class FedConv2d(nn.Conv2d):
                 # the fp32->bf16 weight cast is autograd-tracked, so bwd-weight
                 # gradients land on the fp32 master copy
                 x = x.contiguous(memory_format=torch.channels_last)
-                w = _cached_frozen(
-                    self, "_w16", self.weight,
-                    lambda: self.weight.to(torch.bfloat16).contiguous(
-                        memory_format=torch.channels_last))
+                once = self.__dict__.pop("_w16_once", None)
+                if once is not None and once[0] == self.weight._version:
+                    w = once[1]
+                else:
+                    w = _cached_frozen(
+                        self, "_w16", self.weight,
+                        lambda: self.weight.to(torch.bfloat16).contiguous(
+                            memory_format=torch.channels_last))
                 y, part = _ConvFn.apply(x, w, self.stride[0], self.padding[0])
                 if self.training and part.numel():
                     y._fedkit_bn_stats = part
