@@ -569,3 +569,44 @@ def test_cpc_models_gpu_smoke():
         loss = info_nce(z.float(), zhat.float())
     loss.backward()
     assert torch.isfinite(loss)
+
+
+# ---------------------------------------- full-model composition numerics
+
+def test_resnet18_gpu_end_to_end_vs_eager():
+    """Whole ResNet18 fwd+bwd on the native kernels vs the same weights on
+    stock fp32 torch ops — catches composition bugs the per-op tests miss
+    (layout handoffs, fused BN/ELU boundaries, pooling, weight pre-cast)."""
+    import fedkit.ops as ops
+    from fedkit.models import ResNet18
+    torch.manual_seed(41)
+    net = ResNet18().cuda().to(memory_format=torch.channels_last)
+    x = torch.randn(16, 3, 32, 32, device="cuda").contiguous(
+        memory_format=torch.channels_last)
+    yt = torch.randint(0, 10, (16,), device="cuda")
+
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        logits = net(x)
+        loss = F.cross_entropy(logits.float(), yt)
+    loss.backward()
+    nat_logits = logits.float().detach().clone()
+    nat_grads = torch.cat([p.grad.detach().float().reshape(-1)
+                           for p in net.parameters() if p.grad is not None])
+    net.zero_grad(set_to_none=True)
+
+    old = ops._NATIVE_ENV
+    ops._NATIVE_ENV = False
+    try:
+        logits2 = net(x)                   # stock fp32 ops, same weights
+        loss2 = F.cross_entropy(logits2, yt)
+        loss2.backward()
+    finally:
+        ops._NATIVE_ENV = old
+    ref_logits = logits2.float().detach()
+    ref_grads = torch.cat([p.grad.detach().float().reshape(-1)
+                           for p in net.parameters() if p.grad is not None])
+
+    assert frob_err(nat_logits, ref_logits) < 3e-2, \
+        frob_err(nat_logits, ref_logits)
+    assert frob_err(nat_grads, ref_grads) < 6e-2, \
+        frob_err(nat_grads, ref_grads)
