@@ -395,33 +395,6 @@ void rot_weight64_kernel(const bf16* __restrict__ w,   // [K][R*S*C]
   }
 }
 
-// 2x2 parity interleave for the stride-2 bwd-data decomposition:
-// dx[n][2a+dh][2b+dw][c] = cls_{dh,dw}[n][a][b][c]
-__global__ void interleave2x2_kernel(const bf16* __restrict__ ee,
-                                     const bf16* __restrict__ eo,
-                                     const bf16* __restrict__ oe,
-                                     const bf16* __restrict__ oo,
-                                     bf16* __restrict__ dx, int Cv, int H,
-                                     int W, long long total) {
-  typedef __attribute__((ext_vector_type(8))) __bf16 v8;
-  const v8* srcs[4] = {(const v8*)ee, (const v8*)eo, (const v8*)oe,
-                       (const v8*)oo};
-  v8* out = (v8*)dx;
-  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
-       i < total; i += (long long)gridDim.x * blockDim.x) {
-    int cv = (int)(i % Cv);
-    long long t = i / Cv;
-    int w = (int)(t % W);
-    t /= W;
-    int h = (int)(t % H);
-    int n = (int)(t / H);
-    int cls = (h & 1) * 2 + (w & 1);
-    long long si =
-        (((long long)n * (H >> 1) + (h >> 1)) * (W >> 1) + (w >> 1)) * Cv + cv;
-    out[i] = srcs[cls][si];
-  }
-}
-
 // dw GEMM on the transposed operands: D[kout][rsc] += A[kout][m] B[rsc][m]
 // with A = dyT (plain 2-D) and B = implicit transposed im2col of xpT.
 // Same 256-thread / 4-wave / 3-stage-glds structure as conv_fwd_kernel;
@@ -961,37 +934,6 @@ at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
     int pb = (int)H - ((P - 1) * 2 + 1);
     int pr = (int)W - ((Q - 1) * 2 + 1);
     return dilate_pad_nhwc(dxs, 0, pb, 0, pr, 2);
-  }
-  if (R == 3 && S == 3 && stride == 2 && padding == 1 && dil == 1 &&
-      H % 2 == 0 && W % 2 == 0 && C % 8 == 0) {
-    // Parity decomposition: the dilate-then-conv form runs the GEMM on an
-    // input that is 3/4 zeros.  Each (h%2, w%2) class of dx touches a
-    // FIXED subset of filter taps, so dx = four small dense convs of gy
-    // (padded +1 bottom/right) with tap-subsets of the rotated weight,
-    // interleaved back — exactly the useful FLOPs, no zero MFMA work.
-    int P = gy.size(2), Q = gy.size(3);
-    auto gyp = pad_nhwc(gy, 0, 1, 0, 1);
-    auto cl = at::MemoryFormat::ChannelsLast;
-    // wrot logical [Cpad][K][3][3]; r-hat/s-hat 0<->orig tap 2, 2<->0
-    auto wee = wrot.slice(2, 1, 2).slice(3, 1, 2).contiguous(cl);
-    auto weo = wrot.slice(2, 1, 2).slice(3, 0, 3, 2).contiguous(cl);
-    auto woe = wrot.slice(2, 0, 3, 2).slice(3, 1, 2).contiguous(cl);
-    auto woo = wrot.slice(2, 0, 3, 2).slice(3, 0, 3, 2).contiguous(cl);
-    auto dee = conv_core(gyp, wee, 1, P, Q, 1, (int)ctrue);
-    auto deo = conv_core(gyp, weo, 1, P, Q, 1, (int)ctrue);
-    auto doe = conv_core(gyp, woe, 1, P, Q, 1, (int)ctrue);
-    auto doo = conv_core(gyp, woo, 1, P, Q, 1, (int)ctrue);
-    auto dx = at::empty({gy.size(0), ctrue, H, W}, gy.options()
-                            .memory_format(at::MemoryFormat::ChannelsLast));
-    auto stream = fedkit_stream();
-    long long totalv = (long long)gy.size(0) * H * W * ctrue / 8;
-    TORCH_CHECK(ctrue % 8 == 0, "parity dx needs C % 8 == 0");
-    hipLaunchKernelGGL(interleave2x2_kernel, dim3(grid_1d(totalv, 256)),
-                       dim3(256), 0, stream, (const bf16*)dee.data_ptr(),
-                       (const bf16*)deo.data_ptr(), (const bf16*)doe.data_ptr(),
-                       (const bf16*)doo.data_ptr(), (bf16*)dx.data_ptr(),
-                       (int)ctrue / 8, (int)H, (int)W, totalv);
-    return dx;
   }
   // dilate+pad gy: pl = (R-1)*dil - pad, pr = pl + a with
   // a = (H + 2p - Reff) % stride, Reff = (R-1)*dil + 1
