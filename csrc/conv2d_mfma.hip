@@ -58,7 +58,12 @@ __device__ __forceinline__ int lds_off(int row, int k) {
 // Kg loop and the epilogue writes fp32 partials aux[z][M][Kout] —
 // for shapes whose (M, Kout) grid alone cannot fill 256 CUs
 // (layer4: 2048x512 -> 256 workgroups, measured 37 us at 12% MFMA).
-template <int BM, int STRIDE, int MODE>
+// STAGES: 3 = double-lookahead pipeline (72 KB LDS at BM=128 -> 2 wg/CU);
+// 2 = single-lookahead (48 KB -> 3 wg/CU: PMC shows 43% wave-wait at
+// occupancy 2 with zero LDS conflicts, so more resident waves may hide
+// more latency than the deeper pipeline) — picked per measurement via
+// FEDKIT_CONV_STAGES.
+template <int BM, int STRIDE, int MODE, int STAGES>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      const bf16* __restrict__ w,   // [Kout][R*S*C]
@@ -71,7 +76,7 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      float* __restrict__ aux) {
   constexpr int AB = BM * BK * 2;          // A tile bytes
   constexpr int BB = BN * BK * 2;          // B tile bytes
-  __shared__ char smem[3 * (AB + BB)];     // 3-stage pipeline
+  __shared__ char smem[STAGES * (AB + BB)];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -194,17 +199,26 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   // landed, and every wave is done reading buffer (kt+2)%3 (used by kt-1).
   constexpr int LPS = A_SLOTS + B_SLOTS;
   if (nkt > 0) stage(0, kt0);
-  if (nkt > 1) stage(1, kt0 + 1);
+  if (STAGES == 3 && nkt > 1) stage(1, kt0 + 1);
 
   for (int kt = 0; kt < nkt; ++kt) {
-    if (kt + 1 < nkt)
-      asm volatile("s_waitcnt vmcnt(%0)" ::"n"(LPS) : "memory");
-    else
+    if (STAGES == 3) {
+      // double lookahead: tile kt+1 stays in flight across the wait
+      if (kt + 1 < nkt)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(LPS) : "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      if (kt + 2 < nkt) stage((kt + 2) % 3, kt0 + kt + 2);
+    } else {
+      // single lookahead: drain kt, fence the buffer kt+1 overwrites
+      // (consumed in compute kt-1), issue kt+1, compute kt
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
-    if (kt + 2 < nkt) stage((kt + 2) % 3, kt0 + kt + 2);
-    const char* A = bufA(kt % 3);
-    const char* B = bufB(kt % 3);
+      __builtin_amdgcn_s_barrier();
+      if (kt + 1 < nkt) stage((kt + 1) % 2, kt0 + kt + 1);
+    }
+    const char* A = bufA(kt % STAGES);
+    const char* B = bufB(kt % STAGES);
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 32) {
       bf16x8 a[MR], b[NR];
@@ -791,25 +805,34 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
                        N, Hp, Wp, C, Kout, R, S, P, Q, Kg, dil, Ktrue,
                        aux);
   };
+  static const int stages2 = []() {
+    const char* e = getenv("FEDKIT_CONV_STAGES");
+    return e && atoi(e) == 2;
+  }();
   if (splits > 1) {
     if (stride == 1)
-      bm64 ? L(conv_fwd_kernel<64, 1, 2>) : L(conv_fwd_kernel<128, 1, 2>);
+      bm64 ? L(conv_fwd_kernel<64, 1, 2, 3>) : L(conv_fwd_kernel<128, 1, 2, 3>);
     else
-      bm64 ? L(conv_fwd_kernel<64, 2, 2>) : L(conv_fwd_kernel<128, 2, 2>);
+      bm64 ? L(conv_fwd_kernel<64, 2, 2, 3>) : L(conv_fwd_kernel<128, 2, 2, 3>);
     long long Ly = M * Kout;
     hipLaunchKernelGGL(colsum_to_bf16_kernel, dim3(grid_1d(Ly, 256)),
                        dim3(256), 0, stream, part.data_ptr<float>(), splits,
                        Ly, (bf16*)y.data_ptr());
   } else if (bnpart) {
     if (stride == 1)
-      bm64 ? L(conv_fwd_kernel<64, 1, 1>) : L(conv_fwd_kernel<128, 1, 1>);
+      bm64 ? L(conv_fwd_kernel<64, 1, 1, 3>) : L(conv_fwd_kernel<128, 1, 1, 3>);
     else
-      bm64 ? L(conv_fwd_kernel<64, 2, 1>) : L(conv_fwd_kernel<128, 2, 1>);
+      bm64 ? L(conv_fwd_kernel<64, 2, 1, 3>) : L(conv_fwd_kernel<128, 2, 1, 3>);
+  } else if (stages2) {
+    if (stride == 1)
+      bm64 ? L(conv_fwd_kernel<64, 1, 0, 2>) : L(conv_fwd_kernel<128, 1, 0, 2>);
+    else
+      bm64 ? L(conv_fwd_kernel<64, 2, 0, 2>) : L(conv_fwd_kernel<128, 2, 0, 2>);
   } else {
     if (stride == 1)
-      bm64 ? L(conv_fwd_kernel<64, 1, 0>) : L(conv_fwd_kernel<128, 1, 0>);
+      bm64 ? L(conv_fwd_kernel<64, 1, 0, 3>) : L(conv_fwd_kernel<128, 1, 0, 3>);
     else
-      bm64 ? L(conv_fwd_kernel<64, 2, 0>) : L(conv_fwd_kernel<128, 2, 0>);
+      bm64 ? L(conv_fwd_kernel<64, 2, 0, 3>) : L(conv_fwd_kernel<128, 2, 0, 3>);
   }
   return y;
 }
