@@ -610,3 +610,21 @@ def test_resnet18_gpu_end_to_end_vs_eager():
         frob_err(nat_logits, ref_logits)
     assert frob_err(nat_grads, ref_grads) < 6e-2, \
         frob_err(nat_grads, ref_grads)
+
+
+@pytest.mark.parametrize("name", ["Net", "Net1", "Net2"])
+def test_classifier_models_gpu_smoke(name):
+    """Net/Net1/Net2 fwd+bwd on GPU (5x5/valid-3x3 convs take the stock
+    path by design — R*S*pad8(C) % 64 != 0; pools take the NHWC kernels)."""
+    import fedkit.models as M
+    torch.manual_seed(51)
+    net = getattr(M, name)().cuda()
+    x = torch.randn(32, 3, 32, 32, device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out = net(x)
+        loss = F.cross_entropy(out.float(), torch.randint(0, 10, (32,),
+                                                          device="cuda"))
+    loss.backward()
+    assert torch.isfinite(loss)
+    for p in net.parameters():
+        assert p.grad is None or torch.isfinite(p.grad).all()
