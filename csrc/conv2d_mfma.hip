@@ -341,10 +341,13 @@ __device__ __forceinline__ int tr_off(int row, int k) {
   return row * 128 + chunk * 16 + (k & 7) * 2;
 }
 
+// mscale/moff subsample the input rows (out[k][m'] = in[m'*mscale+moff][k])
+// — used to build the even/odd-column planes for the stride-2 dw path.
 __global__ __launch_bounds__(256)
 void transpose_mk_kernel(const bf16* __restrict__ in,  // [M][K]
                          bf16* __restrict__ out,       // [K][M]
-                         long long M, int K) {
+                         long long M, int K, long long mscale,
+                         long long moff) {
   __shared__ char smem[64 * 128];
   const int tid = threadIdx.x;
   const long long mt = (long long)blockIdx.x * 64;
@@ -354,7 +357,7 @@ void transpose_mk_kernel(const bf16* __restrict__ in,  // [M][K]
     int d = i * 256 + tid;
     int row = d >> 3, c = d & 7;
     *(bf16x8*)(smem + tr_off(row, c * 8)) =
-        *(const bf16x8*)(in + (mt + row) * K + kt + c * 8);
+        *(const bf16x8*)(in + ((mt + row) * mscale + moff) * K + kt + c * 8);
   }
   __syncthreads();
 #pragma unroll
@@ -420,10 +423,15 @@ void rot_weight64_kernel(const bf16* __restrict__ w,   // [K][R*S*C]
 // numerics tests pass) and 34-40 % faster than register staging + ds_write
 // (layer3 dw 64.6 -> 38.8 us) — the default; FEDKIT_DW_GLDSB=0 falls back
 // to the alignment-safe register path.
-template <int BMK, int BGLDS>
+// STRIDE=2 reads from TWO half-width transposed planes (even/odd input
+// columns): 8 consecutive output q stay contiguous within the plane of
+// the filter column's parity, so the same glds staging applies; Wp is the
+// PLANE width and xpT2 the odd plane.
+template <int BMK, int BGLDS, int STRIDE>
 __global__ __launch_bounds__(256)
 void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
-                    const bf16* __restrict__ xpT,  // [C][N][Hp][Wp]
+                    const bf16* __restrict__ xpT,  // [C][N][Hp][Wp(lane)]
+                    const bf16* __restrict__ xpT2, // odd plane (STRIDE=2)
                     float* __restrict__ part,      // [SPLITS][K][RSC]
                     int K, int C, int N, int Hp, int Wp, int S,
                     long long M, int RSC, int mtiles_per_split,
@@ -455,7 +463,8 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
     a_k8[i] = (d & 7) ^ (row & 7);
     a_base[i] = (long long)(krow0 + row) * M;
   }
-  long long b_base[2];                     // xpT (c, r-row, s-col) offset
+  long long b_base[2];                     // plane (c, r-row, s-col) offset
+  const bf16* b_plane[2];
   int b_k8[2];
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
@@ -467,7 +476,13 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
     int rs = rsc / C;
     int s = rs % S;
     int r = rs / S;
-    b_base[i] = ((long long)c * N * Hp + r) * Wp + s;  // + n*Hp*Wp + p*Wp + q
+    if (STRIDE == 2) {
+      b_plane[i] = (s & 1) ? xpT2 : xpT;
+      b_base[i] = ((long long)c * N * Hp + r) * Wp + (s >> 1);
+    } else {
+      b_plane[i] = xpT;
+      b_base[i] = ((long long)c * N * Hp + r) * Wp + s;
+    }
   }
 
   auto bufA = [&](int b) -> char* { return smem + b * (AB + BB); };
@@ -492,7 +507,8 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
       int q = (int)(mm & qmask);
       int p = (int)((mm >> qshift) & pmask);
       int n = (int)(mm >> pshift);
-      const bf16* src = xpT + b_base[i] + ((long long)n * Hp + p) * Wp + q;
+      const bf16* src =
+          b_plane[i] + b_base[i] + ((long long)n * Hp + p * STRIDE) * Wp + q;
       if (BGLDS)
         GLDS16(src, bufB(buf) + (i * 4 + wave) * 1024);
       else
@@ -980,24 +996,45 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
   long long RSC = (long long)R * S * C;
 
   // transpose-then-implicit-GEMM MFMA path (no col buffer, no library GEMM):
-  // see the dw block comment above conv kernels.  stride-1 only (stride-2 m
-  // chunks are not q-contiguous in xpT), Q >= 8 so an m-chunk stays within
-  // one q row — covers 13/17 ResNet18 convs, ~80 % of dw time.
+  // see the dw block comment above conv kernels.  Q >= 8 so an m-chunk
+  // stays within one q row.  Stride 1 uses one transposed plane; stride 2
+  // uses EVEN/ODD-column planes (each filter column's parity is fixed, so
+  // 8 consecutive q remain contiguous in its plane) — together 15/17
+  // ResNet18 convs.
   int Hp = xp.size(2), Wp = xp.size(3);
   long long NHW = (long long)N * Hp * Wp;
   bool pow2 = P > 0 && Q > 0 && (P & (P - 1)) == 0 && (Q & (Q - 1)) == 0;
-  if (stride == 1 && dil == 1 && pow2 && Q % 8 == 0 && K % 64 == 0 &&
-      RSC % 64 == 0 && C % 64 == 0 && M % 64 == 0 && NHW % 64 == 0) {
+  bool common = dil == 1 && pow2 && Q % 8 == 0 && K % 64 == 0 &&
+      RSC % 64 == 0 && C % 64 == 0 && M % 64 == 0;
+  if ((stride == 1 && common && NHW % 64 == 0) ||
+      (stride == 2 && common && Wp % 2 == 0 && (NHW / 2) % 64 == 0)) {
     auto stream = fedkit_stream();
     auto dyT = at::empty({(long long)K, M}, gy.options());
-    auto xpT = at::empty({(long long)C, NHW}, xp.options());
     hipLaunchKernelGGL(transpose_mk_kernel, dim3((unsigned)(M / 64), K / 64),
                        dim3(256), 0, stream, (const bf16*)gy.data_ptr(),
-                       (bf16*)dyT.data_ptr(), M, K);
-    hipLaunchKernelGGL(transpose_mk_kernel,
-                       dim3((unsigned)(NHW / 64), C / 64), dim3(256), 0,
-                       stream, (const bf16*)xp.data_ptr(),
-                       (bf16*)xpT.data_ptr(), NHW, C);
+                       (bf16*)dyT.data_ptr(), M, K, 1LL, 0LL);
+    at::Tensor xpT, xpT2;
+    int Wlane = Wp;
+    if (stride == 1) {
+      xpT = at::empty({(long long)C, NHW}, xp.options());
+      hipLaunchKernelGGL(transpose_mk_kernel,
+                         dim3((unsigned)(NHW / 64), C / 64), dim3(256), 0,
+                         stream, (const bf16*)xp.data_ptr(),
+                         (bf16*)xpT.data_ptr(), NHW, C, 1LL, 0LL);
+    } else {
+      Wlane = Wp / 2;
+      long long NHW2 = NHW / 2;
+      xpT = at::empty({(long long)C, NHW2}, xp.options());
+      xpT2 = at::empty({(long long)C, NHW2}, xp.options());
+      hipLaunchKernelGGL(transpose_mk_kernel,
+                         dim3((unsigned)(NHW2 / 64), C / 64), dim3(256), 0,
+                         stream, (const bf16*)xp.data_ptr(),
+                         (bf16*)xpT.data_ptr(), NHW2, C, 2LL, 0LL);
+      hipLaunchKernelGGL(transpose_mk_kernel,
+                         dim3((unsigned)(NHW2 / 64), C / 64), dim3(256), 0,
+                         stream, (const bf16*)xp.data_ptr(),
+                         (bf16*)xpT2.data_ptr(), NHW2, C, 2LL, 1LL);
+    }
     int qshift = __builtin_ctz((unsigned)Q);
     int pshift = qshift + __builtin_ctz((unsigned)P);
     int BMK = K % 128 == 0 ? 128 : 64;     // kout tile (128 ~1.7x faster)
@@ -1020,14 +1057,26 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
     auto LD = [&](auto kern) {
       hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
                          (const bf16*)dyT.data_ptr(),
-                         (const bf16*)xpT.data_ptr(), part.data_ptr<float>(),
-                         K, C, N, Hp, Wp, S, M, (int)RSC, mps,
+                         (const bf16*)xpT.data_ptr(),
+                         stride == 2 ? (const bf16*)xpT2.data_ptr() : nullptr,
+                         part.data_ptr<float>(),
+                         K, C, N, Hp, Wlane, S, M, (int)RSC, mps,
                          qshift, Q - 1, pshift, P - 1);
     };
-    if (bglds)
-      BMK == 128 ? LD(dw_gemm_kernel<128, 1>) : LD(dw_gemm_kernel<64, 1>);
-    else
-      BMK == 128 ? LD(dw_gemm_kernel<128, 0>) : LD(dw_gemm_kernel<64, 0>);
+    if (stride == 2) {
+      if (bglds)
+        BMK == 128 ? LD(dw_gemm_kernel<128, 1, 2>)
+                   : LD(dw_gemm_kernel<64, 1, 2>);
+      else
+        BMK == 128 ? LD(dw_gemm_kernel<128, 0, 2>)
+                   : LD(dw_gemm_kernel<64, 0, 2>);
+    } else if (bglds) {
+      BMK == 128 ? LD(dw_gemm_kernel<128, 1, 1>)
+                 : LD(dw_gemm_kernel<64, 1, 1>);
+    } else {
+      BMK == 128 ? LD(dw_gemm_kernel<128, 0, 1>)
+                 : LD(dw_gemm_kernel<64, 0, 1>);
+    }
     long long L = (long long)K * RSC;
     if (splits > 8) {
       auto part2 = at::empty({8, L}, xp.options().dtype(at::kFloat));

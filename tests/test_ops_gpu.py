@@ -223,6 +223,10 @@ RESNET_SHAPES = [
     (16, 256, 8, 512, 3, 2),
     (16, 512, 4, 512, 3, 1),
     (16, 64, 32, 128, 1, 2),   # shortcut projections
+    # N=64 so the stride-2 even/odd-plane dw fast path qualifies
+    # ((N*Hp*Wp/2) % 64 == 0 — N=16 falls back to the library path)
+    (64, 64, 32, 128, 3, 2),
+    (64, 128, 16, 256, 3, 2),
     (16, 128, 16, 256, 1, 2),
     (16, 256, 8, 512, 1, 2),
 ]
