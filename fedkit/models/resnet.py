@@ -46,24 +46,7 @@ class BasicBlock(nn.Module):
                 FedBatchNorm2d(self.expansion * planes),
             )
 
-    def _precast(self, x):
-        # one fused kernel casts every TRAINABLE conv weight to bf16 for
-        # this step's graph (per-conv casts cost a launch each; frozen
-        # weights keep their version-cached copies)
-        from ..ops import native_enabled
-        if not native_enabled(x):
-            return
-        mods = self.__dict__.get("_fedconvs")
-        if mods is None:
-            mods = [m for m in self.modules() if isinstance(m, FedConv2d)]
-            self.__dict__["_fedconvs"] = mods
-        from ..ops.conv import batch_cast_weights
-        train_mods = [m for m in mods if m.weight.requires_grad]
-        if len(train_mods) > 1:
-            batch_cast_weights(train_mods)
-
     def forward(self, x):
-        self._precast(x)
         out = bn_elu(self.bn1, self.conv1(x))
         return bn_elu(self.bn2, self.conv2(out), residual=self.shortcut(x))
 
@@ -114,7 +97,24 @@ class ResNet(nn.Module):
             self.in_planes = planes * block.expansion
         return nn.Sequential(*layers)
 
+    def _precast(self, x):
+        # one fused kernel casts every TRAINABLE conv weight to bf16 for
+        # this step's graph (per-conv casts cost a launch each; frozen
+        # weights keep their version-cached copies)
+        from ..ops import native_enabled
+        if not native_enabled(x):
+            return
+        mods = self.__dict__.get("_fedconvs")
+        if mods is None:
+            mods = [m for m in self.modules() if isinstance(m, FedConv2d)]
+            self.__dict__["_fedconvs"] = mods
+        from ..ops.conv import batch_cast_weights
+        train_mods = [m for m in mods if m.weight.requires_grad]
+        if len(train_mods) > 1:
+            batch_cast_weights(train_mods)
+
     def forward(self, x):
+        self._precast(x)
         out = bn_elu(self.bn1, self.conv1(x))
         out = self.layer1(out)
         out = self.layer2(out)
