@@ -427,7 +427,10 @@ void rot_weight64_kernel(const bf16* __restrict__ w,   // [K][R*S*C]
 // columns): 8 consecutive output q stay contiguous within the plane of
 // the filter column's parity, so the same glds staging applies; Wp is the
 // PLANE width and xpT2 the odd plane.
-template <int BMK, int BGLDS, int STRIDE>
+// NT=2 doubles the rsc tile (two B tiles share one A stage: 8 MFMAs per
+// 6 LDS reads instead of 4 per 4) — used for layer1 (K=64 caps the kout
+// tile at 64 and its 48.8 us was the largest single kernel).
+template <int BMK, int BGLDS, int STRIDE, int NT = 1>
 __global__ __launch_bounds__(256)
 void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
                     const bf16* __restrict__ xpT,  // [C][N][Hp][Wp(lane)]
@@ -437,14 +440,15 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
                     long long M, int RSC, int mtiles_per_split,
                     int qshift /* log2 Q */, int qmask,
                     int pshift /* log2 (P*Q) */, int pmask) {
+  constexpr int BNT = BN * NT;             // rsc tile rows
   constexpr int AB = BMK * BK * 2;         // A tile bytes
-  constexpr int BB = BN * BK * 2;          // B tile bytes
+  constexpr int BB = BNT * BK * 2;         // B tile bytes
   __shared__ char smem[3 * (AB + BB)];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
-  const int rrow0 = blockIdx.x * BN;       // rsc tile base
+  const int rrow0 = blockIdx.x * BNT;      // rsc tile base
   const int krow0 = blockIdx.y * BMK;      // kout tile base
   const long long m0 =
       (long long)blockIdx.z * mtiles_per_split * BK;
@@ -463,15 +467,17 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
     a_k8[i] = (d & 7) ^ (row & 7);
     a_base[i] = (long long)(krow0 + row) * M;
   }
-  long long b_base[2];                     // plane (c, r-row, s-col) offset
-  const bf16* b_plane[2];
-  int b_k8[2];
+  constexpr int BS = 2 * NT;               // B slots per thread
+  long long b_base[BS];                    // plane (c, r-row, s-col) offset
+  const bf16* b_plane[BS];
+  int b_k8[BS];
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
+  for (int i = 0; i < BS; ++i) {
     int d = i * 256 + tid;
     int row = d >> 3;
     b_k8[i] = (d & 7) ^ (row & 7);
     int rsc = rrow0 + row;
+    if (rsc >= RSC) rsc = RSC - 1;         // tail tile: junk rows, stores guarded
     int c = rsc % C;
     int rs = rsc / C;
     int s = rs % S;
@@ -493,7 +499,7 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
   // unaligned-capable global vector load into VGPRs, committed to LDS with
   // ds_write_b128 right before the barrier (guide T14 — ties glds within a
   // few %, and sidesteps the LDS-DMA alignment question entirely).
-  bf16x8 breg[2][2];
+  bf16x8 breg[2][BS];
   auto stage = [&](int buf, int kt, bf16x8* br) {
     const long long mt = m0 + (long long)kt * BK;
 #pragma unroll
@@ -502,7 +508,7 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
       GLDS16(src, bufA(buf) + (i * 4 + wave) * 1024);
     }
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
+    for (int i = 0; i < BS; ++i) {
       long long mm = mt + b_k8[i] * 8;     // 8 consecutive m = 8 consecutive q
       int q = (int)(mm & qmask);
       int p = (int)((mm >> qshift) & pmask);
@@ -517,25 +523,26 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
   };
   auto commitB = [&](int buf, const bf16x8* br) {
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < BS; ++i)
       *(bf16x8*)(bufB(buf) + (i * 4 + wave) * 1024 + lane * 16) = br[i];
   };
 
   constexpr int MR = BMK / 2 / 16;         // kout frags per wave (2 or 4)
+  constexpr int NRW = 2 * NT;              // rsc frags per wave
   const int frag_row = lane & 15;
   const int frag_k = (lane >> 4) * 8;
-  f32x4 acc[MR][2] = {};                   // wave tile (BMK/2) x 32(rsc)
+  f32x4 acc[MR][NRW] = {};                 // wave tile (BMK/2) x (BNT/2)
   const int wk = (wave >> 1) * (BMK / 2);
-  const int wr = (wave & 1) * 32;
+  const int wr = (wave & 1) * (BNT / 2);
 
   if (nkt > 0) {
     stage(0, 0, breg[0]);
     if (nkt > 1) stage(1, 1, breg[1]);
     for (int kt = 0; kt < nkt; ++kt) {
-      // drain stage kt's VMEM ops (A_SLOTS glds + 2 B reg loads); stage
+      // drain stage kt's VMEM ops (A_SLOTS glds + BS B loads); stage
       // kt+1's stay in flight
       if (kt + 1 < nkt)
-        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(A_SLOTS + 2) : "memory");
+        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(A_SLOTS + BS) : "memory");
       else
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       if (!BGLDS) {
@@ -548,19 +555,19 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
       const char* B = bufB(kt % 3);
 #pragma unroll
       for (int kk = 0; kk < BK; kk += 32) {
-        bf16x8 a[MR], b[2];
+        bf16x8 a[MR], b[NRW];
 #pragma unroll
         for (int f = 0; f < MR; ++f)
           a[f] = *(const bf16x8*)(A + lds_off(wk + f * 16 + frag_row,
                                               kk + frag_k));
 #pragma unroll
-        for (int f = 0; f < 2; ++f)
+        for (int f = 0; f < NRW; ++f)
           b[f] = *(const bf16x8*)(B + lds_off(wr + f * 16 + frag_row,
                                               kk + frag_k));
 #pragma unroll
         for (int fa = 0; fa < MR; ++fa)
 #pragma unroll
-          for (int fb = 0; fb < 2; ++fb)
+          for (int fb = 0; fb < NRW; ++fb)
             acc[fa][fb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 a[fa], b[fb], acc[fa][fb], 0, 0, 0);
       }
@@ -576,8 +583,9 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
     for (int v = 0; v < 4; ++v) {
       int k = krow0 + wk + fa * 16 + (lane >> 4) * 4 + v;
 #pragma unroll
-      for (int fb = 0; fb < 2; ++fb)
-        out[(long long)k * RSC + col + fb * 16] = acc[fa][fb][v];
+      for (int fb = 0; fb < NRW; ++fb)
+        if (NT == 1 || col + fb * 16 < RSC)
+          out[(long long)k * RSC + col + fb * 16] = acc[fa][fb][v];
     }
   }
 }
@@ -1038,10 +1046,14 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
     int qshift = __builtin_ctz((unsigned)Q);
     int pshift = qshift + __builtin_ctz((unsigned)P);
     int BMK = K % 128 == 0 ? 128 : 64;     // kout tile (128 ~1.7x faster)
+    // K=64 caps the kout tile; widen the rsc tile instead (NT=2)
+    bool nt2 = BMK == 64 && stride == 1 && RSC >= 128;
+    int rsc_tiles = nt2 ? (int)((RSC + 127) / 128) : (int)(RSC / 64);
     long long mtiles = M / 64;
-    long long tiles_xy = (RSC / 64) * (K / BMK);
+    long long tiles_xy = (long long)rsc_tiles * (K / BMK);
+    int maxsplit = nt2 ? 128 : 64;
     int splits = 1;
-    while (splits < 64 && tiles_xy * splits < 512 &&
+    while (splits < maxsplit && tiles_xy * splits < 512 &&
            (long long)splits * 2 <= mtiles)
       splits *= 2;
     int mps = (int)((mtiles + splits - 1) / splits);
@@ -1053,7 +1065,7 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
       const char* e = getenv("FEDKIT_DW_GLDSB");
       return e ? atoi(e) : 1;
     }();
-    dim3 grid((unsigned)(RSC / 64), K / BMK, splits);
+    dim3 grid((unsigned)rsc_tiles, K / BMK, splits);
     auto LD = [&](auto kern) {
       hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
                          (const bf16*)dyT.data_ptr(),
@@ -1070,6 +1082,9 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
       else
         BMK == 128 ? LD(dw_gemm_kernel<128, 0, 2>)
                    : LD(dw_gemm_kernel<64, 0, 2>);
+    } else if (nt2) {
+      bglds ? LD(dw_gemm_kernel<64, 1, 1, 2>)
+            : LD(dw_gemm_kernel<64, 0, 1, 2>);
     } else if (bglds) {
       BMK == 128 ? LD(dw_gemm_kernel<128, 1, 1>)
                  : LD(dw_gemm_kernel<64, 1, 1>);
