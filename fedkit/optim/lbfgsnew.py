@@ -81,6 +81,10 @@ class LBFGSNew(Optimizer):
         return self._numel_cache
 
     def _flat_grad(self):
+        # layouts preserved: the pack kernel traverses PHYSICAL storage
+        # order, matching _move_along's add_flat over p.data — a standard
+        # .contiguous() here would reorder channels_last conv-weight grads
+        # and permute the applied direction
         grads = []
         for p in self._params:
             if p.grad is None:
@@ -89,7 +93,7 @@ class LBFGSNew(Optimizer):
                 grads.append(p.grad.to_dense())
             else:
                 grads.append(p.grad)
-        return flat_ops.pack([g.data.contiguous().view(-1).float() for g in grads])
+        return flat_ops.pack([g.data.float() for g in grads])
 
     def _move_along(self, alpha, direction):
         """params += alpha * direction (flat)."""

@@ -41,8 +41,8 @@ from ..data import make_client_datasets
 from ..models import MODEL_FACTORIES
 from ..optim import LBFGSNew
 from ..ops import losses as loss_ops
-from ..utils import (freeze_all_layers, get_trainable_values, init_weights,
-                     number_of_blocks, put_trainable_values,
+from ..utils import (flat_trainable, freeze_all_layers, get_trainable_values,
+                     init_weights, number_of_blocks, put_trainable_values,
                      save_client_checkpoint, load_client_checkpoint,
                      trainable_params, unfreeze_one_block, unfreeze_one_layer)
 from .comm import Communicator, make_comm
@@ -238,7 +238,10 @@ class FederatedJob:
                 loss = loss.float()
                 reg_on = (ci in lin_ids) or cfg.l2_all_blocks
                 if self.strategy.uses_penalty or reg_on:
-                    vec = torch.cat([p.view(-1) for p in trainable_params(net)])
+                    # PHYSICAL element order so the FedProx/ADMM penalty's
+                    # (x - z) pairs elements the way the packed z was built
+                    # (channels_last weights reorder under reshape(-1))
+                    vec = flat_trainable(net)
                     pen = self.strategy.penalty(self._state, ck, vec)
                     if pen is not None:
                         loss = loss + pen

@@ -4,7 +4,7 @@ from .paramvec import (
     unfreeze_all_layers,
     unfreeze_one_layer,
     unfreeze_one_block,
-    get_trainable_values,
+    flat_physical, flat_trainable, get_trainable_values,
     put_trainable_values,
     number_of_layers,
     number_of_blocks,

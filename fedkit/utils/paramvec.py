@@ -66,6 +66,24 @@ def trainable_params(net: nn.Module) -> List[torch.Tensor]:
     return [p for p in net.parameters() if p.requires_grad]
 
 
+def flat_physical(p: torch.Tensor) -> torch.Tensor:
+    """Differentiable flat view of p in PHYSICAL (storage) element order —
+    the order the fused pack/unpack/axpy kernels traverse.  Penalty terms
+    built from parameters (FedProx/ADMM closures) must use this so
+    (x - z) pairs elements consistently with the packed federation
+    vectors; channels_last conv weights differ from .reshape(-1)'s
+    logical order.  Zero-copy for standard and channels_last layouts."""
+    if p.dim() == 4 and p.is_contiguous(memory_format=torch.channels_last):
+        return p.permute(0, 2, 3, 1).reshape(-1)
+    return p.reshape(-1)
+
+
+def flat_trainable(net: nn.Module) -> torch.Tensor:
+    """Differentiable physical-order flat vector of the trainable params
+    (the closure-side counterpart of get_trainable_values)."""
+    return torch.cat([flat_physical(p) for p in trainable_params(net)])
+
+
 def get_trainable_values(net: nn.Module, mydevice=None) -> torch.Tensor:
     """Flatten trainable parameters into one contiguous fp32 vector
     (simple_utils.py:47-66) — one fused kernel on GPU."""

@@ -632,3 +632,28 @@ def test_classifier_models_gpu_smoke(name):
     assert torch.isfinite(loss)
     for p in net.parameters():
         assert p.grad is None or torch.isfinite(p.grad).all()
+
+
+def test_flat_order_consistency_channels_last():
+    """The penalty-side flat vector (flat_trainable) and the LBFGS flat
+    grad must traverse elements in the SAME physical order the fused
+    pack/unpack/axpy kernels use — channels_last weights made the old
+    reshape(-1)/contiguous() forms silently permute (x - z) pairs."""
+    from fedkit.models import ResNet18
+    from fedkit.utils import (flat_trainable, get_trainable_values,
+                              unfreeze_one_block)
+    from fedkit.optim.lbfgsnew import LBFGSNew
+    torch.manual_seed(61)
+    net = ResNet18().cuda().to(memory_format=torch.channels_last)
+    unfreeze_one_block(net, 3)
+    v_pack = get_trainable_values(net, torch.device("cuda"))
+    v_cat = flat_trainable(net).detach()
+    assert torch.equal(v_pack, v_cat)
+
+    # grad order: set grad = value, flat grad must equal packed values
+    params = [p for p in net.parameters() if p.requires_grad]
+    for p in params:
+        p.grad = p.data.clone()
+    opt = LBFGSNew(params)
+    g = opt._flat_grad()
+    assert torch.equal(g, v_pack)
