@@ -657,3 +657,23 @@ def test_flat_order_consistency_channels_last():
     opt = LBFGSNew(params)
     g = opt._flat_grad()
     assert torch.equal(g, v_pack)
+
+
+def test_frozen_weight_cache_invalidation():
+    """The per-module bf16 weight cache keys on the parameter's in-place
+    version counter: put_trainable_values / optimizer writes must be
+    picked up on the next forward."""
+    from fedkit.ops.conv import FedConv2d
+    torch.manual_seed(71)
+    m = FedConv2d(64, 64, 3, padding=1, bias=False).cuda().to(
+        memory_format=torch.channels_last)
+    m.weight.requires_grad_(False)
+    x = torch.randn(8, 64, 16, 16, device="cuda", dtype=torch.bfloat16
+                    ).contiguous(memory_format=torch.channels_last)
+    y1 = m(x)
+    y2 = m(x)                               # served from cache
+    assert torch.equal(y1, y2)
+    with torch.no_grad():
+        m.weight.mul_(2.0)                  # in-place: version bump
+    y3 = m(x)
+    assert frob_err(y3.float(), 2.0 * y1.float()) < 1e-2
