@@ -1,4 +1,5 @@
-// Direct NHWC conv for tiny input-channel counts (ResNet conv1: C=3).
+// Direct NHWC conv for tiny input-channel counts (ResNet conv1: C=3,
+// simple_models.py:191).
 //
 // Two paths:
 //  * conv_c3_mfma_kernel — the flagship conv1 shape (C=3, 3x3, stride 1,
@@ -164,7 +165,9 @@ void conv_c3_mfma_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][4]
           *(const uint64_t*)src;
     }
   }
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  // drain the B-tile DMA (vmcnt) AND this wave's A-tap ds_writes
+  // (lgkmcnt) before the raw barrier publishes the tiles
+  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
   constexpr int MR = 4;                        // (BM/2)/16 frags per wave
