@@ -515,10 +515,16 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
       int n = (int)(mm >> pshift);
       const bf16* src =
           b_plane[i] + b_base[i] + ((long long)n * Hp + p * STRIDE) * Wp + q;
-      if (BGLDS)
+      if (QH) {
+        // q == 0 here (chunks are 8-aligned, Q == 4): rows p and p+1
+        __builtin_memcpy(&br[i], src, 8);
+        __builtin_memcpy(reinterpret_cast<char*>(&br[i]) + 8,
+                         src + (long long)STRIDE * Wp, 8);
+      } else if (BGLDS) {
         GLDS16(src, bufB(buf) + (i * 4 + wave) * 1024);
-      else
+      } else {
         __builtin_memcpy(&br[i], src, sizeof(bf16x8));
+      }
     }
   };
   auto commitB = [&](int buf, const bf16x8* br) {
@@ -539,10 +545,11 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
     stage(0, 0, breg[0]);
     if (nkt > 1) stage(1, 1, breg[1]);
     for (int kt = 0; kt < nkt; ++kt) {
-      // drain stage kt's VMEM ops (A_SLOTS glds + BS B loads); stage
-      // kt+1's stay in flight
+      // drain stage kt's VMEM ops (A_SLOTS glds + the B loads — two
+      // 8-byte pieces per slot under QH); stage kt+1's stay in flight
+      constexpr int BOPS = (QH ? 2 : 1) * BS;
       if (kt + 1 < nkt)
-        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(A_SLOTS + BS) : "memory");
+        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(A_SLOTS + BOPS) : "memory");
       else
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       if (!BGLDS) {
@@ -1012,7 +1019,8 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
   int Hp = xp.size(2), Wp = xp.size(3);
   long long NHW = (long long)N * Hp * Wp;
   bool pow2 = P > 0 && Q > 0 && (P & (P - 1)) == 0 && (Q & (Q - 1)) == 0;
-  bool common = dil == 1 && pow2 && Q % 8 == 0 && K % 64 == 0 &&
+  bool qh = Q == 4 && P >= 2;
+  bool common = dil == 1 && pow2 && (Q % 8 == 0 || qh) && K % 64 == 0 &&
       RSC % 64 == 0 && C % 64 == 0 && M % 64 == 0;
   if ((stride == 1 && common && NHW % 64 == 0) ||
       (stride == 2 && common && Wp % 2 == 0 && (NHW / 2) % 64 == 0)) {
@@ -1047,7 +1055,7 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
     int pshift = qshift + __builtin_ctz((unsigned)P);
     int BMK = K % 128 == 0 ? 128 : 64;     // kout tile (128 ~1.7x faster)
     // K=64 caps the kout tile; widen the rsc tile instead (NT=2)
-    bool nt2 = BMK == 64 && stride == 1 && RSC >= 128;
+    bool nt2 = BMK == 64 && stride == 1 && RSC >= 128 && !qh;
     int rsc_tiles = nt2 ? (int)((RSC + 127) / 128) : (int)(RSC / 64);
     long long mtiles = M / 64;
     long long tiles_xy = (long long)rsc_tiles * (K / BMK);
@@ -1075,7 +1083,15 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
                          K, C, N, Hp, Wlane, S, M, (int)RSC, mps,
                          qshift, Q - 1, pshift, P - 1);
     };
-    if (stride == 2) {
+    if (qh) {
+      // Q == 4 shapes take the two-piece register commit (see kernel doc)
+      if (stride == 2)
+        BMK == 128 ? LD(dw_gemm_kernel<128, 0, 2, 1, 1>)
+                   : LD(dw_gemm_kernel<64, 0, 2, 1, 1>);
+      else
+        BMK == 128 ? LD(dw_gemm_kernel<128, 0, 1, 1, 1>)
+                   : LD(dw_gemm_kernel<64, 0, 1, 1, 1>);
+    } else if (stride == 2) {
       if (bglds)
         BMK == 128 ? LD(dw_gemm_kernel<128, 1, 2>)
                    : LD(dw_gemm_kernel<64, 1, 2>);
