@@ -430,7 +430,11 @@ void rot_weight64_kernel(const bf16* __restrict__ w,   // [K][R*S*C]
 // NT=2 doubles the rsc tile (two B tiles share one A stage: 8 MFMAs per
 // 6 LDS reads instead of 4 per 4) — used for layer1 (K=64 caps the kout
 // tile at 64 and its 48.8 us was the largest single kernel).
-template <int BMK, int BGLDS, int STRIDE, int NT = 1>
+// QH=1 handles Q == 4 (layer4-class shapes): an 8-m chunk spans TWO
+// output rows (q 0..3 of p and of p+1), staged as two 8-byte register
+// loads — LDS-DMA cannot compose two pieces into one 16-B/lane image, so
+// these shapes force the register-staging commit path.
+template <int BMK, int BGLDS, int STRIDE, int NT = 1, int QH = 0>
 __global__ __launch_bounds__(256)
 void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
                     const bf16* __restrict__ xpT,  // [C][N][Hp][Wp(lane)]
