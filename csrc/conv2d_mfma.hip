@@ -1023,7 +1023,11 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
   int Hp = xp.size(2), Wp = xp.size(3);
   long long NHW = (long long)N * Hp * Wp;
   bool pow2 = P > 0 && Q > 0 && (P & (P - 1)) == 0 && (Q & (Q - 1)) == 0;
-  bool qh = Q == 4 && P >= 2;
+  // Q==4 via two-piece register staging MEASURED SLOWER than the im2col
+  // + rocBLAS split-K path it would replace (layer4 dw 36 -> 67 us,
+  // C256-s2 30 -> 44): at M = 2048 the library's tiling wins; the
+  // kernel-side QH support stays for reference, disabled here.
+  bool qh = false;
   bool common = dil == 1 && pow2 && (Q % 8 == 0 || qh) && K % 64 == 0 &&
       RSC % 64 == 0 && C % 64 == 0 && M % 64 == 0;
   if ((stride == 1 && common && NHW % 64 == 0) ||
