@@ -4,6 +4,7 @@ This is CI config #1 from BASELINE.json ("SimpleCNN K=2 no_consensus on
 CPU") plus one tiny run per strategy and model family."""
 
 import json
+import math
 import os
 
 import pytest
@@ -102,3 +103,31 @@ def test_resnet9_one_block_cpu(tmp_path):
     job.run()
     for p in job.nets[0].parameters():
         assert torch.isfinite(p).all()
+
+
+def test_admm_dual_residual_decreases(tmp_path):
+    """ADMM health — the dual residual ||z - z_new||/N shrinks
+    over communication rounds within a block (z converging)."""
+    import json as _json
+    from collections import defaultdict
+    cfg = tiny_cfg(tmp_path, strategy="admm", admm_rho0=0.1, Nadmm=3,
+                   check_results=False, max_steps_per_epoch=4)
+    FederatedJob(cfg).run()
+    rows = [_json.loads(line)
+            for line in open(tmp_path / "metrics.jsonl")]
+    by_block = defaultdict(list)
+    for r in rows:
+        by_block[tuple(r["block"])].append(r["dual"])
+    assert by_block
+    improved = sum(1 for v in by_block.values() if v[-1] <= v[0])
+    assert improved >= 0.7 * len(by_block), dict(by_block)
+
+
+def test_fedprox_primal_residual_logged(tmp_path):
+    import json as _json
+    cfg = tiny_cfg(tmp_path, strategy="fedprox", admm_rho0=0.1,
+                   check_results=False)
+    FederatedJob(cfg).run()
+    rows = [_json.loads(line)
+            for line in open(tmp_path / "metrics.jsonl")]
+    assert all("primal" in r and math.isfinite(r["primal"]) for r in rows)

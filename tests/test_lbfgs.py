@@ -143,3 +143,25 @@ def test_multi_tensor_params():
     _run(opt, closure, 5)
     assert torch.allclose(a.detach(), target_a, atol=1e-2)
     assert torch.allclose(b.detach(), target_b, atol=1e-2)
+
+
+def test_two_loop_fused_matches_serial_cpu():
+    """The Gram-matrix two-loop (_two_loop_fused) must reproduce the
+    serial reference recursion — CPU version of the GPU test, exercising
+    the host-side algebra with the fallback multi_dot/lincomb."""
+    import torch
+    from fedkit.optim.lbfgsnew import LBFGSNew
+    torch.manual_seed(5)
+    n = 4099
+    p = torch.nn.Parameter(torch.zeros(n))
+    opt = LBFGSNew([p], history_size=7)
+    g = torch.randn(n)
+    hist_s = [torch.randn(n) for _ in range(6)]
+    hist_y = [2.0 * s + 0.05 * torch.randn(n) for s in hist_s]
+    SY = [[float(s.dot(y)) for y in hist_y] for s in hist_s]
+    YY = [[float(a.dot(b)) for b in hist_y] for a in hist_y]
+    H = 0.41
+    d_ref = opt._two_loop(g, list(hist_y), list(hist_s), H,
+                          [None] * 7, [None] * 7)
+    d_fus = opt._two_loop_fused(g, hist_y, hist_s, H, SY, YY)
+    assert torch.allclose(d_fus, d_ref, rtol=1e-5, atol=1e-5)
