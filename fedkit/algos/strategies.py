@@ -55,7 +55,8 @@ class Strategy:
     uses_penalty = False    # does the closure add a penalty term?
 
     def init_block(self, comm: Communicator, N: int, device,
-                   x0: Optional[Dict[int, torch.Tensor]] = None) -> dict:
+                   x0: Optional[Dict[int, torch.Tensor]] = None,
+                   block_idx: int = 0) -> dict:
         """Fresh per-block state; z starts at 0 (federated_multi.py:151-154)."""
         return {"z": torch.zeros(N, dtype=torch.float32, device=device), "N": N}
 
@@ -119,15 +120,30 @@ class FedProx(Strategy):
 
     def __init__(self, rho0: float = 1.0):
         self.rho0 = rho0
+        # per-(block, client) proximal weight table, lazily filled at rho0.
+        # The reference allocates rho = torch.ones(L,3)*admm_rho0
+        # (fedprox_multi.py:142, "per layer, per slave") but only ever READS
+        # rho[ci,0] and never writes any entry, so every value stays rho0;
+        # keeping the full table here implements the evident intent
+        # (per-block, per-client weights) while remaining numerically
+        # identical to the reference.
+        self.rho_table: Dict[tuple, float] = {}
 
-    def init_block(self, comm, N, device, x0=None):
+    def _rho(self, block_idx: int, ck: int) -> float:
+        return self.rho_table.setdefault((block_idx, ck), self.rho0)
+
+    def init_block(self, comm, N, device, x0=None, block_idx=0):
         st = super().init_block(comm, N, device)
-        st["rho"] = self.rho0
+        for ck in comm.my_clients:
+            self._rho(block_idx, ck)
+        st["block_idx"] = block_idx
+        st["rho"] = self._rho(block_idx, comm.my_clients[0])
         return st
 
     def penalty(self, state, ck, xvec):
         xdelta = xvec - state["z"]
-        return 0.5 * state["rho"] * (torch.norm(xdelta, 2) ** 2)
+        rho = self._rho(state.get("block_idx", 0), ck)
+        return 0.5 * rho * (torch.norm(xdelta, 2) ** 2)
 
     def aggregate_start(self, comm, state, x):
         return comm.sum_across_clients_async(
@@ -161,7 +177,7 @@ class ConsensusADMM(Strategy):
         self.rho0 = rho0
         self.bb = bb or BBConfig()
 
-    def init_block(self, comm, N, device, x0=None):
+    def init_block(self, comm, N, device, x0=None, block_idx=0):
         st = super().init_block(comm, N, device)
         st["rho"] = self.rho0
         st["y"] = {ck: torch.zeros(N, dtype=torch.float32, device=device)

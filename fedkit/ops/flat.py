@@ -29,15 +29,39 @@ def total_numel(tensors: Sequence[torch.Tensor]) -> int:
 
 
 def _flat_views(flat: torch.Tensor, tensors: Sequence[torch.Tensor]) -> List[torch.Tensor]:
+    """Views of `flat` shaped like each tensor, laid out in each tensor's
+    PHYSICAL (storage) element order — matching what the native
+    pack/unpack/axpy kernels traverse (they walk storage linearly).  For a
+    channels_last 4D tensor the slice is viewed [N,H,W,C] then permuted to
+    the logical [N,C,H,W] shape, so elementwise copies between the view and
+    the tensor land in storage order.  Keeps the CPU fallback bit-consistent
+    with GPU-produced federation vectors."""
     views = []
     off = 0
     for t in tensors:
         n = t.numel()
-        views.append(flat.narrow(0, off, n).view_as(t))
+        sl = flat.narrow(0, off, n)
+        if t.dim() == 4 and t.is_contiguous(memory_format=torch.channels_last) \
+                and not t.is_contiguous():
+            N, C, H, W = t.shape
+            views.append(sl.view(N, H, W, C).permute(0, 3, 1, 2))
+        else:
+            views.append(sl.view_as(t))
         off += n
     if off != flat.numel():
         raise ValueError(f"flat vector has {flat.numel()} elements, tensors need {off}")
     return views
+
+
+def _bump_versions(tensors: Sequence[torch.Tensor]) -> None:
+    """The native kernels write through raw data pointers, bypassing the
+    dispatcher's in-place version bump; derived-weight caches key on
+    `param._version` (fedkit.ops.conv._cached_frozen), so a silent write
+    would serve stale frozen bf16 weights after put_trainable_values /
+    LBFGS moves.  `.data`/`.detach()` share the counter
+    with the parameter, so bumping here covers the param itself."""
+    for t in tensors:
+        torch.autograd.graph.increment_version(t)
 
 
 def pack_into(tensors: Sequence[torch.Tensor], flat: torch.Tensor) -> torch.Tensor:
@@ -60,6 +84,7 @@ def unpack(flat: torch.Tensor, tensors: Sequence[torch.Tensor]) -> None:
     if _use_native(tensors):
         from . import require_ext
         require_ext().unpack_params(flat, list(tensors))
+        _bump_versions(tensors)
         return
     torch._foreach_copy_(list(tensors), _flat_views(flat, tensors))
 
@@ -69,6 +94,7 @@ def add_flat(tensors: Sequence[torch.Tensor], flat: torch.Tensor, alpha: float) 
     if _use_native(tensors):
         from . import require_ext
         require_ext().add_flat_params(list(tensors), flat, float(alpha))
+        _bump_versions(tensors)
         return
     torch._foreach_add_(list(tensors), _flat_views(flat, tensors), alpha=alpha)
 

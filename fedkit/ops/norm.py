@@ -110,10 +110,23 @@ class FedBatchNorm2d(nn.BatchNorm2d):
                 self._nbt_pending = getattr(self, "_nbt_pending", 0) + 1
 
     def _flush_nbt(self):
+        """Materialize host-accumulated step counts into the
+        num_batches_tracked buffer.  INVARIANT: the buffer may lag by
+        `_nbt_pending` steps between flush points; flushes happen on
+        state_dict save, state_dict load (reset), and train()/eval()
+        transitions — any code reading the buffer directly mid-epoch
+        (cross-rank BN sync, external tooling) must call this first
+        (ADVICE r1)."""
         pending = getattr(self, "_nbt_pending", 0)
         if pending and self.num_batches_tracked is not None:
             self.num_batches_tracked.add_(pending)
         self._nbt_pending = 0
+
+    def train(self, mode: bool = True):
+        # flush at mode transitions so eval-time readers see the true count
+        if getattr(self, "_nbt_pending", 0):
+            self._flush_nbt()
+        return super().train(mode)
 
     def _save_to_state_dict(self, destination, prefix, keep_vars):
         self._flush_nbt()

@@ -40,6 +40,8 @@ _WOLFE_T2 = 0.1
 _WOLFE_T3 = 0.5
 # trust-region damping coefficient in batch mode (lbfgsnew.py:559)
 _BATCH_LM0 = 1e-6
+# Max vectors per fused multi_dot call (csrc/flat_ops.hip kMaxVecs).
+_FUSED_MAX_VECS = 24
 
 
 class LBFGSNew(Optimizer):
@@ -368,7 +370,8 @@ class LBFGSNew(Optimizer):
         state.setdefault("n_iter", 0)
 
         orig_loss = closure()
-        loss = float(orig_loss)
+        loss = float(orig_loss.detach() if torch.is_tensor(orig_loss)
+                     else orig_loss)
         current_evals = 1
         state["func_evals"] += 1
 
@@ -417,8 +420,14 @@ class LBFGSNew(Optimizer):
                     y.add_(s, alpha=_BATCH_LM0)      # trust-region damping
 
                 # fused path: every scalar the update and the two-loop need
-                # comes from two multi_dot passes and ONE host sync
-                fused = flat_ops.fused_available(flat_grad)
+                # comes from two multi_dot passes and ONE host sync.
+                # multi_dot takes at most _FUSED_MAX_VECS vectors per call
+                # (csrc/flat_ops.hip kMaxVecs); the largest call below is
+                # hist_s + hist_y + [s, y] = 2*history+2, so history sizes
+                # past (kMaxVecs-2)//2 fall back to the unfused two-loop
+                # instead of aborting mid-training.
+                fused = flat_ops.fused_available(flat_grad) \
+                    and history_size <= (_FUSED_MAX_VECS - 2) // 2
                 SY, YY = state.get("SY", []), state.get("YY", [])
                 n0 = len(hist_y)
                 if fused:
@@ -516,7 +525,8 @@ class LBFGSNew(Optimizer):
 
             if n_iter != max_iter:
                 # re-evaluate (new grads for the next inner iteration)
-                loss = float(closure())
+                _l = closure()
+                loss = float(_l.detach() if torch.is_tensor(_l) else _l)
                 flat_grad = self._flat_grad()
                 abs_grad_sum = float(flat_grad.abs().sum())
                 if math.isnan(abs_grad_sum):

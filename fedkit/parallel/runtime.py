@@ -289,7 +289,8 @@ class FederatedJob:
                 if cfg.bb_update:
                     x0 = {ck: get_trainable_values(self.nets[ck], self.device)
                           for ck in comm.my_clients}
-                self._state = self.strategy.init_block(comm, N, self.device, x0)
+                self._state = self.strategy.init_block(comm, N, self.device,
+                                                       x0, block_idx=ci)
                 opts = {ck: self._make_optimizer(self.nets[ck], ci)
                         for ck in comm.my_clients}
                 self.last_opts = opts

@@ -677,3 +677,52 @@ def test_frozen_weight_cache_invalidation():
         m.weight.mul_(2.0)                  # in-place: version bump
     y3 = m(x)
     assert frob_err(y3.float(), 2.0 * y1.float()) < 1e-2
+
+
+def test_frozen_cache_invalidation_native_unpack():
+    """The hazard ADVICE r1 flagged: put_trainable_values writes through the
+    native unpack kernel (raw data_ptr, no dispatcher version bump) — the
+    frozen bf16 weight cache must still see the new values on the next
+    forward (fedkit.ops.flat bumps the version explicitly)."""
+    from fedkit.ops.conv import FedConv2d
+    from fedkit.utils.paramvec import (get_trainable_values,
+                                       put_trainable_values)
+    torch.manual_seed(73)
+    m = FedConv2d(64, 64, 3, padding=1, bias=False).cuda().to(
+        memory_format=torch.channels_last)
+    m.weight.requires_grad_(False)
+    x = torch.randn(8, 64, 16, 16, device="cuda", dtype=torch.bfloat16
+                    ).contiguous(memory_format=torch.channels_last)
+    y1 = m(x)
+    _ = m(x)                                # cache now populated
+    m.weight.requires_grad_(True)           # "unfreeze block"
+    vec = get_trainable_values(m)
+    put_trainable_values(m, 2.0 * vec)      # native unpack write
+    m.weight.requires_grad_(False)          # "re-freeze"
+    y3 = m(x)
+    assert frob_err(y3.float(), 2.0 * y1.float()) < 1e-2
+
+
+def test_lbfgs_history_past_fused_limit_falls_back():
+    """history_size > (kMaxVecs-2)//2 must run via the unfused two-loop
+    instead of aborting mid-training."""
+    from fedkit.optim import LBFGSNew
+    torch.manual_seed(3)
+    A = torch.randn(40, 40, device="cuda")
+    A = A @ A.t() / 40 + torch.eye(40, device="cuda")
+    b = torch.randn(40, device="cuda")
+    x = torch.zeros(40, device="cuda", requires_grad=True)
+    opt = LBFGSNew([x], history_size=14, max_iter=6,
+                   line_search_fn=True, batch_mode=False)
+
+    def closure():
+        opt.zero_grad()
+        f = 0.5 * x @ A @ x - b @ x
+        f.backward()
+        return f
+
+    f0 = float(closure())
+    for _ in range(12):
+        opt.step(closure)
+    f1 = float(closure())
+    assert f1 < f0 - 1.0  # made real progress, no abort

@@ -76,3 +76,37 @@ def test_layer_block_counts():
     assert number_of_blocks(m) == 10
     freeze_all_layers(m)
     assert len(trainable_params(m)) == 0
+
+
+def test_cpu_fallback_physical_order_channels_last():
+    """The CPU fallback pack/unpack/add_flat must traverse PHYSICAL
+    (storage) order for channels_last tensors, matching flat_physical and
+    the native kernels."""
+    from fedkit.utils.paramvec import flat_physical
+    torch.manual_seed(5)
+    w = torch.randn(4, 6, 3, 3).contiguous(memory_format=torch.channels_last)
+    b = torch.randn(6)
+    flat = flat_ops.pack([w, b])
+    expect = torch.cat([flat_physical(w), b])
+    assert torch.equal(flat, expect)
+    # unpack round trip lands values back in the right places
+    w2 = torch.empty_like(w).contiguous(memory_format=torch.channels_last)
+    b2 = torch.empty_like(b)
+    flat_ops.unpack(flat, [w2, b2])
+    assert torch.equal(w2, w) and torch.equal(b2, b)
+    # add_flat pairs elements physically too
+    w3 = w.clone()
+    flat_ops.add_flat([w3, b2], flat, 1.0)
+    assert torch.allclose(w3, 2 * w)
+
+
+def test_unpack_bumps_version_counter():
+    """Derived-weight caches key on param._version; unpack/add_flat must
+    advance it on every write path."""
+    t = torch.randn(8)
+    v0 = t._version
+    flat_ops.unpack(torch.randn(8), [t])
+    assert t._version > v0
+    v1 = t._version
+    flat_ops.add_flat([t], torch.randn(8), 0.5)
+    assert t._version > v1
