@@ -56,7 +56,10 @@ def main():
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
-    distributed = world > 1
+    # any torchrun launch (even nproc=1) goes through the full distributed
+    # path — rendezvous, RCCL init, barriers, the FedAvg all-reduce — so a
+    # 1-GPU lease exercises exactly the code the 8-GPU run will use
+    distributed = world > 1 or "TORCHELASTIC_RUN_ID" in os.environ
     use_cuda = torch.cuda.is_available()
     if distributed:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")

@@ -121,6 +121,10 @@ class DistComm(Communicator):
         self.my_clients = [self.rank]
         if torch.cuda.is_available():
             local_rank = int(os.environ.get("LOCAL_RANK", self.rank))
+            if os.environ.get("FEDKIT_FORCE_DEV0") == "1":
+                # co-locate every rank on device 0 (test rig: exercises the
+                # RCCL collectives with world>1 inside a 1-GPU lease)
+                local_rank = 0
             torch.cuda.set_device(local_rank)
             self._device = torch.device("cuda", local_rank)
             # dedicated stream for collectives so the engine can overlap

@@ -107,3 +107,35 @@ def test_admm_gloo_world2_z_rank_invariant(tmp_path):
     assert s0["rho"] == s1["rho"]          # BB decision replicated exactly
     recs = [json.loads(l) for l in open(tmp_path / "admm.jsonl")]
     assert all("primal" in r and "dual" in r for r in recs)
+
+
+FEDPROX_W4_WORKER = r"""
+import json, os, sys
+sys.path.insert(0, %r)
+import torch
+from fedkit.parallel import FedConfig, FederatedJob
+from fedkit.parallel.comm import DistComm
+
+out_dir = sys.argv[1]
+cfg = FedConfig(K=4, default_batch=16, Nloop=1, Nepoch=1, Nadmm=2,
+                use_cuda=False, check_results=False, max_steps_per_epoch=1,
+                save_model=False, strategy="fedprox", admm_rho0=0.5,
+                jsonl_path=os.path.join(out_dir, "prox.jsonl"))
+comm = DistComm(cfg.K, backend="gloo")
+job = FederatedJob(cfg, comm=comm)
+job.run()
+torch.save({"z": job._state["z"]},
+           os.path.join(out_dir, f"prox_state{comm.rank}.pt"))
+"""
+
+
+def test_fedprox_gloo_world4_z_rank_invariant(tmp_path):
+    """K=4 ranks (cover K>2): z and the logged
+    residuals must be identical on every rank."""
+    run_torchrun(FEDPROX_W4_WORKER % REPO, str(tmp_path), nproc=4)
+    states = [torch.load(tmp_path / f"prox_state{r}.pt", weights_only=False)
+              for r in range(4)]
+    for s in states[1:]:
+        assert torch.allclose(states[0]["z"], s["z"], atol=1e-6)
+    recs = [json.loads(l) for l in open(tmp_path / "prox.jsonl")]
+    assert recs and all("primal" in r and "rho" in r for r in recs)
