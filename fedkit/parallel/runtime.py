@@ -29,6 +29,7 @@ parameters, aggregation, optimizer state and BN statistics stay fp32
 import contextlib
 import json
 import math
+import os
 import time
 from dataclasses import dataclass, field
 from typing import Callable, Dict, Optional
@@ -89,6 +90,11 @@ class FedConfig:
     jsonl_path: Optional[str] = None   # structured per-round metrics
     max_steps_per_epoch: int = 0       # 0 = full shard (tests/bench shrink it)
     max_eval_batches: int = 0          # 0 = full test set
+    round_checkpoint: bool = False     # save s{k}.model + RNG sidecar at each
+                                       # (nloop, ci) boundary; load_model=True
+                                       # then RESUMES at the recorded round
+                                       # instead of replaying (SURVEY §5
+                                       # failure-recovery minimum)
     l2_all_blocks: bool = False        # VAE-CL: L2 reg on every block
                                        # (federated_vae_cl.py:230)
 
@@ -269,13 +275,67 @@ class FederatedJob:
                          self._state["N"], i, epoch, loss1))
         return running_loss
 
+    # -------------------------------------------------- per-round checkpoints
+
+    def _round_state_path(self, ck):
+        return f"{self.cfg.ckpt_prefix}{ck}.round"
+
+    def _save_round_state(self, nloop, ci, opts):
+        """Checkpoint at a (nloop, ci) boundary: the reference-contract
+        s{k}.model file (federated_multi.py:226-233 layout, unchanged) plus
+        a sidecar with everything needed to continue EXACTLY where the
+        uninterrupted run would be — loader shuffle generators and global
+        RNG (VAE reparametrization noise)."""
+        for ck in self.comm.my_clients:
+            save_client_checkpoint(
+                self.nets[ck], opts.get(ck), self.cfg.Nepoch - 1,
+                self.last_running_loss.get(ck, 0.0), ck, self.cfg.ckpt_prefix)
+            side = {
+                "nloop": nloop, "ci": ci,
+                "loader_gen": self.train_loaders[ck].generator.get_state()
+                if self.train_loaders[ck].generator is not None else None,
+                "torch_rng": torch.get_rng_state(),
+                "cuda_rng": torch.cuda.get_rng_state(self.device)
+                if self.device.type == "cuda" else None,
+                "running_loss": self.last_running_loss.get(ck, 0.0),
+            }
+            torch.save(side, self._round_state_path(ck))
+
+    def _load_round_state(self):
+        """Restore sidecar state; returns the last COMPLETED (nloop, ci)
+        or None.  Model weights were already loaded by _build_clients
+        (load_model=True path)."""
+        marks = []
+        for ck in self.comm.my_clients:
+            path = self._round_state_path(ck)
+            if not os.path.exists(path):
+                return None
+            side = torch.load(path, map_location="cpu", weights_only=False)
+            marks.append((side["nloop"], side["ci"]))
+            if side.get("loader_gen") is not None and \
+                    self.train_loaders[ck].generator is not None:
+                self.train_loaders[ck].generator.set_state(side["loader_gen"])
+            torch.set_rng_state(side["torch_rng"])
+            if side.get("cuda_rng") is not None and self.device.type == "cuda":
+                torch.cuda.set_rng_state(side["cuda_rng"], self.device)
+            self.last_running_loss[ck] = side.get("running_loss", 0.0)
+        assert len(set(marks)) == 1, "clients disagree on resume round"
+        return marks[0]
+
     def run(self):
         cfg = self.cfg
         comm = self.comm
         self.last_running_loss = {ck: 0.0 for ck in comm.my_clients}
         self.last_opts = {}
+        done_upto = None
+        if cfg.load_model and cfg.round_checkpoint:
+            done_upto = self._load_round_state()
+            if done_upto is not None and comm.is_primary:
+                print('Resuming after loop=%d block=%d' % done_upto)
         for nloop in range(cfg.Nloop):
             for ci in range(self.L):
+                if done_upto is not None and (nloop, ci) <= done_upto:
+                    continue
                 for ck in comm.my_clients:
                     if cfg.per_layer:
                         unfreeze_one_layer(self.nets[ck], ci)
@@ -329,6 +389,8 @@ class FederatedJob:
                         accs = self.verification_error_check()
                     self._log_round(nloop, ci, nadmm, N, info, t_local,
                                     t_comm, accs)
+                if cfg.round_checkpoint:
+                    self._save_round_state(nloop, ci, opts)
         if comm.is_primary:
             print('Finished Training')
         if cfg.save_model:
