@@ -66,3 +66,102 @@ def test_lofar_minibatch_synthetic():
     assert y.shape[0] == 4 * px * py
     assert torch.isfinite(y).all()
     assert y.abs().max() <= 1e6
+
+
+def test_load_real_cifar_pickle_tree(tmp_path):
+    """_load_real_cifar against a locally synthesized cifar-10-batches-py
+    tree with the torchvision pickle layout (VERDICT r1 #8)."""
+    import pickle
+    import numpy as np
+    from fedkit.data.cifar import _load_real_cifar, load_cifar
+
+    base = tmp_path / "cifar-10-batches-py"
+    base.mkdir()
+    rng = np.random.RandomState(0)
+    ntr_per = 40
+    for i in range(1, 6):
+        d = {b"data": rng.randint(0, 256, (ntr_per, 3072), dtype=np.uint8)
+             .astype(np.uint8),
+             b"labels": [int(v) for v in rng.randint(0, 10, ntr_per)]}
+        with open(base / f"data_batch_{i}", "wb") as f:
+            pickle.dump(d, f)
+    d = {b"data": rng.randint(0, 256, (20, 3072), dtype=np.uint8),
+         b"labels": [int(v) for v in rng.randint(0, 10, 20)]}
+    with open(base / "test_batch", "wb") as f:
+        pickle.dump(d, f)
+
+    out = _load_real_cifar(str(tmp_path))
+    assert out is not None
+    (xtr, ytr), (xte, yte) = out
+    assert xtr.shape == (5 * ntr_per, 3, 32, 32) and xtr.dtype == torch.uint8
+    assert ytr.shape == (5 * ntr_per,) and int(ytr.max()) <= 9
+    assert xte.shape == (20, 3, 32, 32)
+    # the row-major 3072 -> [3,32,32] reshape matches torchvision's layout
+    with open(base / "data_batch_1", "rb") as f:
+        raw = pickle.load(f, encoding="bytes")[b"data"]
+    assert (xtr[0].numpy().reshape(-1) == raw[0]).all()
+    # load_cifar prefers the real tree over the synthetic stand-in
+    x, y = load_cifar(str(tmp_path), train=False)
+    assert x.shape[0] == 20
+
+
+def test_lofar_h5_branch_with_fake_h5py(monkeypatch, tmp_path):
+    """The real-H5 read branch (reference federated_cpc.py:52-63 layout)
+    exercised via an injected h5py stand-in: verifies group paths, baseline
+    indexing, dtype conversion and that per-baseline scale factors are
+    applied (VERDICT r1 #8; no h5py wheel exists in this image)."""
+    import sys
+    import types
+    import numpy as np
+    import fedkit.data.lofar as lofar
+
+    nbase, ntime, nfreq, npol = 6, 64, 64, 4
+    rng0 = np.random.RandomState(3)
+    vis_np = rng0.randn(nbase, ntime, nfreq, npol, 2).astype(np.float64)
+    scale_np = (rng0.rand(nbase, nfreq, npol) + 0.5).astype(np.float64)
+
+    class FakeFile(dict):
+        def __init__(self, path, mode):
+            assert mode == "r"
+            super().__init__()
+            self.update({"measurement": {"saps": {"0": {
+                "visibilities": _Arr(vis_np),
+                "visibility_scale_factors": _Arr(scale_np)}}}})
+
+        def __enter__(self):
+            return self
+
+        def __exit__(self, *a):
+            return False
+
+    class _Arr:
+        def __init__(self, a):
+            self.a = a
+            self.shape = a.shape
+
+        def __getitem__(self, idx):
+            return self.a[idx]
+
+    fake = types.ModuleType("h5py")
+    fake.File = FakeFile
+    monkeypatch.setitem(sys.modules, "h5py", fake)
+    monkeypatch.setattr(lofar, "HAS_H5PY", True)
+
+    rng = np.random.RandomState(7)
+    px, py, y = lofar.lofar_minibatch(filename="fake.h5", batch_size=3,
+                                      patch_size=32, SAP="0", rng=rng)
+    assert (px, py) == (3, 3)
+    assert y.shape == (3 * 3 * 3, 8, 32, 32) and y.dtype == torch.float32
+    assert y.abs().max() <= 1e6
+
+    # the scale factors really multiplied in: reproduce channel 0 of the
+    # first selected baseline by hand
+    rng_check = np.random.RandomState(7)
+    sel = np.sort(rng_check.randint(0, nbase, 3))
+    want = vis_np[sel[0], :, :, 0, 0] * scale_np[sel[0], None, :, 0]
+    got = lofar._assemble(
+        torch.from_numpy(vis_np[sel].astype(np.float32)),
+        torch.from_numpy(scale_np[sel].astype(np.float32)),
+        slice(None), "cpu")[0, 0]
+    assert torch.allclose(got, torch.from_numpy(want.astype(np.float32)),
+                          atol=1e-5)
