@@ -103,7 +103,12 @@ class FedAvg(Strategy):
         z, N = state["z"], state["N"]
         znew = pending()
         znew /= comm.K
-        dual_residual = torch.norm(z - znew).item() / N
+        # device-side residual, NO host sync here: the engine enqueues the
+        # writeback unpack (and the eval forwards behind it) immediately;
+        # the float() happens at print/log time after everything is queued
+        # (VERDICT r1 #10 — removes the host bubble between collective and
+        # writeback on the FedAvg path)
+        dual_residual = torch.norm(z - znew) / N
         state["z"] = znew
         return {"dual": dual_residual}
 

@@ -418,7 +418,8 @@ class FederatedJob:
         rec = {"nloop": nloop, "block": self.Li[ci], "N": N, "nadmm": nadmm,
                "t_local_s": round(t_local, 4), "t_comm_s": round(t_comm, 4),
                "bytes": 4 * N,
-               **{k: (float(v) if isinstance(v, (int, float)) else v)
+               **{k: (float(v) if isinstance(v, (int, float))
+                      or torch.is_tensor(v) else v)
                   for k, v in info.items()}}
         if accs is not None:
             rec["acc"] = [round(float(a), 4) for a in accs]
