@@ -42,6 +42,12 @@ typedef __hip_bfloat16 bf16;
 constexpr int BN = 64;   // out-channel tile
 constexpr int BK = 64;   // im2col-k tile
 
+// 16-B zero page for the Kg tail: when Kg % 64 != 0 (Net-family 5x5 and
+// valid-3x3 shapes: Kg = 200 / 72 / 288) the last K-tile's out-of-range
+// 8-element chunks DMA from here instead of reading past the filter row.
+// Kg is always a multiple of 8 (C padded to x8), so chunks never straddle.
+__device__ __bf16 g_kzero[8] = {};
+
 // LDS byte offset of element (row, k) of a [rows][64] bf16 tile with the
 // chunk-XOR swizzle (16-B chunk index ^ (row & 7)).
 __device__ __forceinline__ int lds_off(int row, int k) {
@@ -155,20 +161,27 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
     for (int i = 0; i < A_SLOTS; ++i) {
       const bf16* src;
       if (c64) {
+        // C % 64 == 0 implies Kg % 64 == 0: no tail possible
         src = xp + a_rowbase[i] + tap_off + c0 + a_k8[i] * 8;
       } else {
         int kg = kt * BK + a_k8[i] * 8;
-        int c = kg % C;
-        int rs = kg / C;
-        int s = rs % S;
-        int r = rs / S;
-        src = xp + a_rowbase[i] + ((long long)r * dil * Wp + s * dil) * C + c;
+        if (kg >= Kg) {
+          src = (const bf16*)g_kzero;        // Kg-tail zero fill
+        } else {
+          int c = kg % C;
+          int rs = kg / C;
+          int s = rs % S;
+          int r = rs / S;
+          src = xp + a_rowbase[i] + ((long long)r * dil * Wp + s * dil) * C + c;
+        }
       }
       GLDS16(src, bufA(buf) + (i * 4 + wave) * 1024);
     }
 #pragma unroll
     for (int i = 0; i < B_SLOTS; ++i) {
-      const bf16* src = w + b_rowbase[i] + kt * BK + b_k8[i] * 8;
+      int kg = kt * BK + b_k8[i] * 8;
+      const bf16* src = kg < Kg ? w + b_rowbase[i] + kg
+                                : (const bf16*)g_kzero;
       GLDS16(src, bufB(buf) + (i * 4 + wave) * 1024);
     }
   };
@@ -184,11 +197,12 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   const int frag_row = lane & 15;          // fragment row/col within 16
   const int frag_k = (lane >> 4) * 8;      // 8 contiguous k per lane
 
-  int kt0 = 0, nkt = Kg / BK;
+  const int nkt_all = (Kg + BK - 1) / BK;   // tail tile zero-filled
+  int kt0 = 0, nkt = nkt_all;
   if (MODE == 2) {
     int kps = (nkt + gridDim.z - 1) / gridDim.z;
     kt0 = blockIdx.z * kps;
-    nkt = min(kps, Kg / BK - kt0);
+    nkt = min(kps, nkt_all - kt0);
   }
   // 3-stage software pipeline: tiles kt and kt+1 are in flight on entry to
   // iteration kt; kt+2 is issued right after the barrier.  The wait is a
@@ -804,7 +818,8 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
   int Kg = R * S * C;
   if (Ktrue < 0) Ktrue = Kout;
   TORCH_CHECK(C % 8 == 0, "conv kernel needs C % 8 == 0, got ", C);
-  TORCH_CHECK(Kg % BK == 0, "conv kernel needs R*S*C % 64 == 0, got ", Kg);
+  // Kg % 64 != 0 is allowed: the kernel zero-fills the tail K-tile from a
+  // device zero page (Net-family 5x5 / valid-3x3 shapes)
   TORCH_CHECK(Kout % BN == 0, "conv kernel needs Kout % 64 == 0, got ", Kout);
   long long M = (long long)N * P * Q;
   auto y = at::empty({N, Ktrue, P, Q},
@@ -818,7 +833,7 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
   // split-K when the (M, Kout) grid alone cannot fill the 256 CUs and the
   // Kg loop is deep enough to slice (layer4-class shapes)
   int gridxy = (int)grid.x * (int)grid.y;
-  int nkt_total = Kg / BK;
+  int nkt_total = (Kg + BK - 1) / BK;
   int splits = 1;
   if (!bnpart && Ktrue == Kout && gridxy <= 256) {
     while (splits < 8 && gridxy * splits < 512 &&

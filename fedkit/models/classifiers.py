@@ -12,6 +12,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.conv import FedConvGeneric
 from ..ops.elu import elu
 from ..ops.pool import FedMaxPool2d
 
@@ -21,9 +22,9 @@ class Net(nn.Module):
 
     def __init__(self):
         super().__init__()
-        self.conv1 = nn.Conv2d(3, 6, 5)
+        self.conv1 = FedConvGeneric(3, 6, 5)
         self.pool = FedMaxPool2d(2, 2)
-        self.conv2 = nn.Conv2d(6, 16, 5)
+        self.conv2 = FedConvGeneric(6, 16, 5)
         self.fc1 = nn.Linear(16 * 5 * 5, 120)
         self.fc2 = nn.Linear(120, 84)
         self.fc3 = nn.Linear(84, 10)
@@ -58,10 +59,10 @@ class Net1(nn.Module):
 
     def __init__(self):
         super().__init__()
-        self.conv1 = nn.Conv2d(3, 32, 3)
-        self.conv2 = nn.Conv2d(32, 32, 3)
-        self.conv3 = nn.Conv2d(32, 64, 3)
-        self.conv4 = nn.Conv2d(64, 64, 3)
+        self.conv1 = FedConvGeneric(3, 32, 3)
+        self.conv2 = FedConvGeneric(32, 32, 3)
+        self.conv3 = FedConvGeneric(32, 64, 3)
+        self.conv4 = FedConvGeneric(64, 64, 3)
         self.pool1 = FedMaxPool2d(2, 2)
         self.pool2 = FedMaxPool2d(2, 2)
         self.fc1 = nn.Linear(64 * 5 * 5, 512)
@@ -92,10 +93,10 @@ class Net2(nn.Module):
 
     def __init__(self):
         super().__init__()
-        self.conv1 = nn.Conv2d(3, 64, 3, padding=1)
-        self.conv2 = nn.Conv2d(64, 128, 3, padding=1)
-        self.conv3 = nn.Conv2d(128, 256, 3, padding=1)
-        self.conv4 = nn.Conv2d(256, 512, 3, padding=1)
+        self.conv1 = FedConvGeneric(3, 64, 3, padding=1)
+        self.conv2 = FedConvGeneric(64, 128, 3, padding=1)
+        self.conv3 = FedConvGeneric(128, 256, 3, padding=1)
+        self.conv4 = FedConvGeneric(256, 512, 3, padding=1)
         self.pool1 = FedMaxPool2d(2, 2)
         self.pool2 = FedMaxPool2d(2, 2)
         self.pool3 = FedMaxPool2d(2, 2)

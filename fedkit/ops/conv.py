@@ -232,11 +232,10 @@ def _gen_conv_supported(mod, x):
     k = mod.kernel_size
     d = mod.dilation
     s = mod.stride
-    if k[0] != k[1] or d[0] != d[1] or s[0] != s[1] or s[0] not in (1, 2) \
-            or mod.groups != 1:
-        return False
-    C8 = _pad_mult(mod.in_channels, 8)
-    return (k[0] * k[1] * C8) % 64 == 0
+    # any square filter qualifies: the MFMA kernel zero-fills the Kg tail
+    # when R*S*pad8(C) % 64 != 0 (Net 5x5, Net1/Net2 valid-3x3 shapes)
+    return not (k[0] != k[1] or d[0] != d[1] or s[0] != s[1]
+                or s[0] not in (1, 2) or mod.groups != 1)
 
 
 class FedConvGeneric(nn.Conv2d):
@@ -291,8 +290,7 @@ class FedConvTranspose2d(nn.ConvTranspose2d):
         C8 = _pad_mult(self.in_channels, 8)
         ok = (output_size is None and k[0] == k[1] and s[0] == s[1]
               and p[0] == p[1] and self.dilation == (1, 1)
-              and self.groups == 1 and k[0] - 1 - p[0] >= 0
-              and (k[0] * k[1] * C8) % 64 == 0 and _native(x))
+              and self.groups == 1 and k[0] - 1 - p[0] >= 0 and _native(x))
         if ok:
             if x.dtype != torch.bfloat16 and torch.is_autocast_enabled():
                 x = x.to(torch.bfloat16)
