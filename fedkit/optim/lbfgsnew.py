@@ -99,13 +99,13 @@ class LBFGSNew(Optimizer):
 
     def _move_along(self, alpha, direction):
         """params += alpha * direction (flat)."""
-        flat_ops.add_flat([p.data for p in self._params], direction, alpha)
+        flat_ops.add_flat([p.detach() for p in self._params], direction, alpha)
 
     def _snapshot(self):
-        return flat_ops.pack([p.data for p in self._params])
+        return flat_ops.pack([p.detach() for p in self._params])
 
     def _restore(self, snap):
-        flat_ops.unpack(snap, [p.data for p in self._params])
+        flat_ops.unpack(snap, [p.detach() for p in self._params])
 
     # ----------------------------------------------------------- line searches
 

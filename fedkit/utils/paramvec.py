@@ -89,7 +89,7 @@ def get_trainable_values(net: nn.Module, mydevice=None) -> torch.Tensor:
     (simple_utils.py:47-66) — one fused kernel on GPU."""
     params = trainable_params(net)
     with torch.no_grad():
-        vec = flat_ops.pack([p.data for p in params])
+        vec = flat_ops.pack([p.detach() for p in params])
     if mydevice is not None and vec.device != torch.device(mydevice):
         vec = vec.to(mydevice)
     return vec
@@ -100,7 +100,7 @@ def put_trainable_values(net: nn.Module, x: torch.Tensor) -> None:
     (simple_utils.py:68-77) — one fused kernel on GPU."""
     params = trainable_params(net)
     with torch.no_grad():
-        flat_ops.unpack(x, [p.data for p in params])
+        flat_ops.unpack(x, [p.detach() for p in params])
 
 
 def number_of_layers(net: nn.Module) -> int:

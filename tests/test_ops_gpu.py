@@ -771,13 +771,15 @@ def test_lbfgs_history_past_fused_limit_falls_back():
                    line_search_fn=True, batch_mode=False)
 
     def closure():
-        opt.zero_grad()
+        if torch.is_grad_enabled():
+            opt.zero_grad()
         f = 0.5 * x @ A @ x - b @ x
-        f.backward()
+        if f.requires_grad:
+            f.backward()
         return f
 
-    f0 = float(closure())
+    f0 = float(closure().detach())
     for _ in range(12):
         opt.step(closure)
-    f1 = float(closure())
+    f1 = float(closure().detach())
     assert f1 < f0 - 1.0  # made real progress, no abort
