@@ -71,6 +71,30 @@ at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
                                     long stride, long padding, long R, long S,
                                     long dil);
 
+at::Tensor fedkit_vae_elbo_fwd(const at::Tensor& recon, const at::Tensor& x,
+                               const at::Tensor& mu, const at::Tensor& logvar);
+std::vector<at::Tensor> fedkit_vae_elbo_bwd(const at::Tensor& recon,
+                                            const at::Tensor& x,
+                                            const at::Tensor& mu,
+                                            const at::Tensor& logvar,
+                                            const at::Tensor& gloss);
+std::vector<at::Tensor> fedkit_vaecl_terms_fwd(
+    const at::Tensor& x, const at::Tensor& mu_th, const at::Tensor& s_th,
+    const at::Tensor& mu_q, const at::Tensor& s_q, const at::Tensor& mu_p,
+    const at::Tensor& s_p, long B);
+std::vector<at::Tensor> fedkit_vaecl_terms_bwd(
+    const at::Tensor& x, const at::Tensor& mu_th, const at::Tensor& s_th,
+    const at::Tensor& mu_q, const at::Tensor& s_q, const at::Tensor& mu_p,
+    const at::Tensor& s_p, const at::Tensor& gR1, const at::Tensor& gR3,
+    long B);
+std::vector<at::Tensor> fedkit_info_nce_fwd(const at::Tensor& Z,
+                                            const at::Tensor& Zhat);
+std::vector<at::Tensor> fedkit_info_nce_bwd(const at::Tensor& Z,
+                                            const at::Tensor& Zhat,
+                                            const at::Tensor& dzz,
+                                            const at::Tensor& tu,
+                                            const at::Tensor& norms);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "fedkit hand-written CDNA4 (gfx950 / MI355X) kernels";
   m.def("elu_fwd", &fedkit_elu_fwd, "ELU forward (vectorized)");
@@ -136,4 +160,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_weight", &fedkit_conv2d_bwd_weight, "conv bwd-weight",
         py::arg("gy"), py::arg("x"), py::arg("stride"), py::arg("padding"),
         py::arg("R"), py::arg("S"), py::arg("dil") = 1);
+  m.def("vae_elbo_fwd", &fedkit_vae_elbo_fwd,
+        "MSE(sum) + analytic KLD in one reduction pass");
+  m.def("vae_elbo_bwd", &fedkit_vae_elbo_bwd, "VAE ELBO backward");
+  m.def("vaecl_terms_fwd", &fedkit_vaecl_terms_fwd,
+        "VAE-CL per-(cluster,sample) cost1/cost3 reductions: (R1, R3)");
+  m.def("vaecl_terms_bwd", &fedkit_vaecl_terms_bwd,
+        "VAE-CL terms backward (elementwise)");
+  m.def("info_nce_fwd", &fedkit_info_nce_fwd,
+        "fused InfoNCE: returns (loss, zz, softmax, norms)");
+  m.def("info_nce_bwd", &fedkit_info_nce_bwd, "InfoNCE backward");
 }

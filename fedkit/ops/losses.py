@@ -58,8 +58,32 @@ class CrossEntropyLoss(torch.nn.Module):
 
 # ---------------------------------------------------------------- VAE losses
 
+class _VaeElboFn(torch.autograd.Function):
+    """MSE(sum) + analytic KLD as ONE fused reduction kernel
+    (csrc/losses_extra.hip; SURVEY.md §2a 'single fused reduction kernel')."""
+
+    @staticmethod
+    def forward(ctx, recon, x, mu, logvar):
+        from . import require_ext
+        loss = require_ext().vae_elbo_fwd(recon, x, mu, logvar)
+        ctx.save_for_backward(recon, x, mu, logvar)
+        return loss
+
+    @staticmethod
+    def backward(ctx, gloss):
+        from . import require_ext
+        recon, x, mu, logvar = ctx.saved_tensors
+        grecon, gmu, glogvar = require_ext().vae_elbo_bwd(
+            recon, x, mu, logvar, gloss)
+        return grecon, None, gmu, glogvar
+
+
 def vae_loss(recon_x, x, mu, logvar):
     """MSE(sum) reconstruction + analytic KL (federated_vae.py:97-108)."""
+    if _native(recon_x) and recon_x.dtype == x.dtype \
+            and mu.dtype == logvar.dtype == recon_x.dtype:
+        return _VaeElboFn.apply(recon_x.contiguous(), x.contiguous(),
+                                mu.contiguous(), logvar.contiguous())
     mse = F.mse_loss(recon_x, x, reduction="sum")
     kld = -0.5 * torch.sum(1 + logvar - mu.pow(2) - logvar.exp())
     return mse + kld
@@ -97,9 +121,65 @@ def cost3(pk, q_z_mu, q_z_sig2, p_z_mu, p_z_sig2):
     return 0.5 * (pk * per_sample).sum() / b
 
 
+class _VaeClTermsFn(torch.autograd.Function):
+    """Per-(cluster, sample) cost1 / cost3 reductions in one kernel over
+    [Kc*B] rows; backward is one elementwise pass (csrc/losses_extra.hip).
+    R1[ci,b] is cost1's per-sample sum, R3[ci,b] is 2x cost3's."""
+
+    @staticmethod
+    def forward(ctx, x, mu_th, s_th, mu_q, s_q, mu_p, s_p, B):
+        from . import require_ext
+        R1, R3 = require_ext().vaecl_terms_fwd(x, mu_th, s_th, mu_q, s_q,
+                                               mu_p, s_p, B)
+        ctx.save_for_backward(x, mu_th, s_th, mu_q, s_q, mu_p, s_p)
+        ctx.B = B
+        return R1, R3
+
+    @staticmethod
+    def backward(ctx, gR1, gR3):
+        from . import require_ext
+        x, mu_th, s_th, mu_q, s_q, mu_p, s_p = ctx.saved_tensors
+        outs = require_ext().vaecl_terms_bwd(
+            x, mu_th, s_th, mu_q, s_q, mu_p, s_p,
+            gR1.contiguous(), gR3.contiguous(), ctx.B)
+        return (None, *outs, None)
+
+
 def vaecl_loss(ekhat, mu_xi, sig2_xi, mu_b, sig2_b, mu_th, sig2_th, x,
                Kc=10, alpha=10.0, beta=1.0):
-    """4-term clustering ELBO (federated_vae_cl.py:142-162)."""
+    """4-term clustering ELBO (federated_vae_cl.py:142-162).
+
+    GPU path: the heavy cost1/cost3 per-sample reductions run as ONE HIP
+    kernel over all Kc clusters (the reference loops over the batch in
+    Python per cluster, 101-140); the tiny pk-weighted combines and
+    cost2/cost21 stay in torch so ekhat's autograd is untouched.
+    """
+    if _native(x):
+        B = x.shape[0]
+        mu_th_s = torch.stack([mu_th[ci] for ci in range(Kc)]).reshape(Kc * B, -1)
+        s_th_s = torch.stack([sig2_th[ci] for ci in range(Kc)]).reshape(Kc * B, -1)
+        mu_q_s = torch.stack([mu_xi[ci] for ci in range(Kc)]).reshape(Kc * B, -1)
+        s_q_s = torch.stack([sig2_xi[ci] for ci in range(Kc)]).reshape(Kc * B, -1)
+        mu_p_s = torch.stack([mu_b[ci] for ci in range(Kc)]).reshape(Kc * B, -1)
+        s_p_s = torch.stack([sig2_b[ci] for ci in range(Kc)]).reshape(Kc * B, -1)
+        xf = x.reshape(B, -1)
+        if not all(t.dtype == xf.dtype for t in
+                   (mu_th_s, s_th_s, mu_q_s, s_q_s, mu_p_s, s_p_s)):
+            xf = xf.float()
+            mu_th_s, s_th_s, mu_q_s, s_q_s, mu_p_s, s_p_s = (
+                t.float() for t in (mu_th_s, s_th_s, mu_q_s, s_q_s,
+                                    mu_p_s, s_p_s))
+        R1, R3 = _VaeClTermsFn.apply(xf.contiguous(), mu_th_s.contiguous(),
+                                     s_th_s.contiguous(), mu_q_s.contiguous(),
+                                     s_q_s.contiguous(), mu_p_s.contiguous(),
+                                     s_p_s.contiguous(), B)
+        pkT = ekhat.t().reshape(-1)                     # [Kc*B]
+        c1_tot = (pkT * R1).sum() / B
+        c3_tot = 0.5 * (pkT * R3).sum() / B
+        c2_tot = -(ekhat * torch.log(ekhat + 1e-9)).sum() / B
+        pbar = ekhat.mean(0)                            # [Kc]
+        c21_tot = (1.0 / (-pbar * torch.log(pbar + 1e-9) + 1e-9)).sum()
+        return c1_tot + alpha * (c2_tot + c3_tot) + beta * c21_tot
     loss = 0
     for ci in range(Kc):
         pk = ekhat[:, ci]
@@ -113,18 +193,51 @@ def vaecl_loss(ekhat, mu_xi, sig2_xi, mu_b, sig2_b, mu_th, sig2_th, x,
 
 # ------------------------------------------------------------------- InfoNCE
 
+class _InfoNceFn(torch.autograd.Function):
+    """Fused InfoNCE: column norms + similarity matrix + row softmax +
+    -sum log(diag) in ONE kernel launch; backward one elementwise pass
+    using t_i = sum_j dzz[i,j] zz[i,j] for the normalization chain
+    (csrc/losses_extra.hip)."""
+
+    @staticmethod
+    def forward(ctx, Z, Zhat):
+        from . import require_ext
+        loss, zz, soft, norms = require_ext().info_nce_fwd(Z, Zhat)
+        ctx.save_for_backward(Z, Zhat, zz, soft, norms)
+        return loss
+
+    @staticmethod
+    def backward(ctx, gloss):
+        from . import require_ext
+        Z, Zhat, zz, soft, norms = ctx.saved_tensors
+        P = zz.shape[0]
+        # dloss/dzz[i,j] = -g * s_ii (delta_ij - s_ij) / (s_ii + 1e-6)
+        diag = soft.diagonal()                          # [P]
+        w = (diag / (diag + 1e-6)) * gloss              # [P]
+        dzz = w.unsqueeze(1) * soft
+        dzz.diagonal().sub_(w)                          # -(I - S) row-scaled
+        t = (dzz * zz).sum(dim=1)                       # [P]
+        u = (dzz * zz).sum(dim=0)                       # [P]
+        tu = torch.stack([t, u])
+        gZ, gZhat = require_ext().info_nce_bwd(Z, Zhat, dzz, tu, norms)
+        return gZ, gZhat
+
+
 def info_nce(z: torch.Tensor, zhat: torch.Tensor) -> torch.Tensor:
     """InfoNCE over the patch grid (federated_cpc.py:149-180).
 
     z, zhat: [batch, channel, px, py].  Positive sample of patch (i) is the
     diagonal of the (px*py)^2 normalized inner-product matrix; negatives are
-    the rest of its row.  One GEMM replaces the reference's O(p^4) loop.
+    the rest of its row.  GPU: one fused kernel; CPU: one GEMM + row softmax
+    (both replace the reference's O(p^4) Python loop, identical values).
     """
     assert z.shape == zhat.shape
     px, py = z.shape[2], z.shape[3]
     P = px * py
     Z = z.reshape(-1, P)
     Zhat = zhat.reshape(-1, P)
+    if _native(z) and P <= 64:
+        return _InfoNceFn.apply(Z.contiguous(), Zhat.contiguous())
     Zn = Z / Z.norm(dim=0, keepdim=True)
     Zhatn = Zhat / Zhat.norm(dim=0, keepdim=True)
     zz = Zn.t() @ Zhatn                     # [P, P]

@@ -19,6 +19,7 @@ SOURCES = [
     "csrc/elementwise.hip",
     "csrc/flat_ops.hip",
     "csrc/loss.hip",
+    "csrc/losses_extra.hip",
     "csrc/batchnorm.hip",
     "csrc/conv2d_mfma.hip",
     "csrc/conv_small.hip",

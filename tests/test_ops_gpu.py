@@ -783,3 +783,100 @@ def test_lbfgs_history_past_fused_limit_falls_back():
         opt.step(closure)
     f1 = float(closure().detach())
     assert f1 < f0 - 1.0  # made real progress, no abort
+
+
+def test_vae_elbo_fused_vs_torch():
+    """Fused MSE(sum)+KLD kernel vs the fp32 torch composition."""
+    from fedkit.ops.losses import vae_loss
+    import fedkit.ops as ops
+    torch.manual_seed(7)
+    recon = torch.rand(64, 3, 32, 32, device="cuda", requires_grad=True)
+    x = torch.rand(64, 3, 32, 32, device="cuda")
+    mu = torch.randn(64, 10, device="cuda", requires_grad=True)
+    logvar = torch.randn(64, 10, device="cuda", requires_grad=True)
+
+    loss = vae_loss(recon, x, mu, logvar)
+    loss.backward()
+    g = [recon.grad.clone(), mu.grad.clone(), logvar.grad.clone()]
+    for t in (recon, mu, logvar):
+        t.grad = None
+
+    old = ops._NATIVE_ENV
+    ops._NATIVE_ENV = False
+    try:
+        loss2 = vae_loss(recon, x, mu, logvar)
+        loss2.backward()
+    finally:
+        ops._NATIVE_ENV = old
+    assert abs(float(loss) - float(loss2)) / abs(float(loss2)) < 1e-5
+    for got, p in zip(g, (recon, mu, logvar)):
+        assert frob_err(got, p.grad) < 1e-5
+
+
+def test_vaecl_terms_fused_vs_torch():
+    """Fused cost1/cost3 reduction kernel (whole vaecl_loss) vs the torch
+    per-cluster composition — values and all input grads."""
+    from fedkit.ops.losses import vaecl_loss
+    import fedkit.ops as ops
+    torch.manual_seed(11)
+    Kc, B, L = 10, 16, 32
+    ekhat = torch.softmax(torch.randn(B, Kc, device="cuda"), 1).requires_grad_(True)
+    x = torch.rand(B, 3, 32, 32, device="cuda")
+
+    def mk(shape):
+        return {ci: torch.rand(*shape, device="cuda").add_(0.1)
+                .requires_grad_(True) for ci in range(Kc)}
+    mu_xi, sig2_xi = mk((B, L)), mk((B, L))
+    mu_b, sig2_b = mk((B, L)), mk((B, L))
+    mu_th, sig2_th = mk((B, 3, 32, 32)), mk((B, 3, 32, 32))
+
+    loss = vaecl_loss(ekhat, mu_xi, sig2_xi, mu_b, sig2_b, mu_th, sig2_th, x)
+    loss.backward()
+    grads = {}
+    leaves = [("ekhat", [ekhat])] + [
+        (nm, list(d.values())) for nm, d in
+        [("mu_xi", mu_xi), ("sig2_xi", sig2_xi), ("mu_b", mu_b),
+         ("sig2_b", sig2_b), ("mu_th", mu_th), ("sig2_th", sig2_th)]]
+    for nm, ts in leaves:
+        grads[nm] = [t.grad.clone() for t in ts]
+        for t in ts:
+            t.grad = None
+
+    old = ops._NATIVE_ENV
+    ops._NATIVE_ENV = False
+    try:
+        loss2 = vaecl_loss(ekhat, mu_xi, sig2_xi, mu_b, sig2_b,
+                           mu_th, sig2_th, x)
+        loss2.backward()
+    finally:
+        ops._NATIVE_ENV = old
+    assert abs(float(loss) - float(loss2)) / abs(float(loss2)) < 1e-4, \
+        (float(loss), float(loss2))
+    for nm, ts in leaves:
+        for got, t in zip(grads[nm], ts):
+            assert frob_err(got, t.grad) < 1e-4, nm
+
+
+def test_info_nce_fused_vs_torch():
+    """Fused InfoNCE kernel vs the GEMM+softmax torch form."""
+    from fedkit.ops.losses import info_nce
+    import fedkit.ops as ops
+    torch.manual_seed(13)
+    z = torch.randn(4, 64, 3, 3, device="cuda", requires_grad=True)
+    zhat = torch.randn(4, 64, 3, 3, device="cuda", requires_grad=True)
+
+    loss = info_nce(z, zhat)
+    loss.backward()
+    gz, gzh = z.grad.clone(), zhat.grad.clone()
+    z.grad = zhat.grad = None
+
+    old = ops._NATIVE_ENV
+    ops._NATIVE_ENV = False
+    try:
+        loss2 = info_nce(z, zhat)
+        loss2.backward()
+    finally:
+        ops._NATIVE_ENV = old
+    assert abs(float(loss) - float(loss2)) / abs(float(loss2)) < 1e-5
+    assert frob_err(gz, z.grad) < 1e-4
+    assert frob_err(gzh, zhat.grad) < 1e-4
