@@ -48,6 +48,19 @@ constexpr int BK = 64;   // im2col-k tile
 // Kg is always a multiple of 8 (C padded to x8), so chunks never straddle.
 __device__ __bf16 g_kzero[8] = {};
 
+// Multi-tap descriptor (CPC dilated bank, SURVEY §2a: "fuse dilation bank
+// (5 convs, same input) into one multi-tap kernel").  The GEMM K dim is
+// the concatenation of each tap's R*S*C block (kg_per each, % 64 == 0);
+// the combined weight is block-diagonal over (tap out-channels, tap kg
+// range), so one MFMA pass computes all taps' outputs.  off[t] is the
+// origin shift (shared_pad - pad_t) into the once-padded input.
+struct TapDesc {
+  int n;        // taps
+  int kg_per;   // R*S*C per tap (multiple of BK)
+  int dil[8];
+  int off[8];
+};
+
 // LDS byte offset of element (row, k) of a [rows][64] bf16 tile with the
 // chunk-XOR swizzle (16-B chunk index ^ (row & 7)).
 __device__ __forceinline__ int lds_off(int row, int k) {
@@ -69,7 +82,7 @@ __device__ __forceinline__ int lds_off(int row, int k) {
 // occupancy 2 with zero LDS conflicts, so more resident waves may hide
 // more latency than the deeper pipeline) — picked per measurement via
 // FEDKIT_CONV_STAGES.
-template <int BM, int STRIDE, int MODE, int STAGES>
+template <int BM, int STRIDE, int MODE, int STAGES, bool MT = false>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      const bf16* __restrict__ w,   // [Kout][R*S*C]
@@ -79,7 +92,7 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      int Ktrue /* <= Kout: K rows beyond are zero-padding
                                   (VAE/CPC channel counts), skipped on
                                   store so y needs no unpad pass */,
-                     float* __restrict__ aux) {
+                     float* __restrict__ aux, TapDesc td = {}) {
   constexpr int AB = BM * BK * 2;          // A tile bytes
   constexpr int BB = BN * BK * 2;          // B tile bytes
   __shared__ char smem[STAGES * (AB + BB)];
@@ -139,7 +152,7 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   auto bufA = [&](int b) -> char* { return smem + b * (AB + BB); };
   auto bufB = [&](int b) -> char* { return smem + b * (AB + BB) + AB; };
 
-  const bool c64 = (C % 64) == 0;
+  const bool c64 = !MT && (C % 64) == 0;
   auto stage = [&](int buf, int kt) {
     // A tile: per-slot source (r,s,c) from k_global; 16-B LDS-DMA.
     // LDS dest for a glds is wave-uniform base + lane*16: slot d = pass*256
@@ -157,6 +170,15 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
       int r = rs / S;
       tap_off = ((long long)r * dil * Wp + s * dil) * C;
     }
+    // multi-tap: the whole 64-k tile lies in ONE tap (kg_per % 64 == 0),
+    // so the tap lookup is SCALAR per stage
+    int mt_dil = 0, mt_off = 0, mt_base = 0;
+    if (MT) {
+      int tap = (kt * BK) / td.kg_per;
+      mt_dil = td.dil[tap];
+      mt_off = td.off[tap];
+      mt_base = tap * td.kg_per;
+    }
 #pragma unroll
     for (int i = 0; i < A_SLOTS; ++i) {
       const bf16* src;
@@ -167,6 +189,15 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
         int kg = kt * BK + a_k8[i] * 8;
         if (kg >= Kg) {
           src = (const bf16*)g_kzero;        // Kg-tail zero fill
+        } else if (MT) {
+          int inner = kg - mt_base;
+          int c = inner % C;
+          int rs = inner / C;
+          int s = rs % S;
+          int r = rs / S;
+          src = xp + a_rowbase[i] +
+                ((long long)(r * mt_dil + mt_off) * Wp + s * mt_dil + mt_off)
+                    * C + c;
         } else {
           int c = kg % C;
           int rs = kg / C;
@@ -853,7 +884,7 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
                        (const bf16*)xp.data_ptr(),
                        (const bf16*)w_krs_c.data_ptr(), (bf16*)y.data_ptr(),
                        N, Hp, Wp, C, Kout, R, S, P, Q, Kg, dil, Ktrue,
-                       aux);
+                       aux, TapDesc{});
   };
   static const int stages2 = []() {
     const char* e = getenv("FEDKIT_CONV_STAGES");
@@ -888,6 +919,70 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
 }
 
 }  // namespace
+
+// Fused multi-dilation conv bank (CPC EncoderCNN, simple_models.py:441-460):
+// n_tap convs share one input and one launch.  x [N,C,H,W] NHWC bf16;
+// w2d [64, n_tap*R*R*C] bf16 contiguous BLOCK-DIAGONAL combined weight
+// (rows t*Kt .. per tap t nonzero only in its kg block, Kt true out
+// channels per tap); dils/pads per tap; all taps must produce the same
+// P x Q.  Returns [N, ktrue, P, Q] channels_last (the torch.cat result).
+at::Tensor fedkit_conv2d_dilated_bank(const at::Tensor& x,
+                                      const at::Tensor& w2d,
+                                      std::vector<long> dils,
+                                      std::vector<long> pads, long stride,
+                                      long R, long ktrue) {
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              w2d.scalar_type() == at::kBFloat16, "bank is bf16-only");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "bank input must be channels_last");
+  TORCH_CHECK(w2d.is_contiguous() && w2d.size(0) == 64,
+              "bank weight must be [64, n*R*R*C] contiguous");
+  int n = (int)dils.size();
+  TORCH_CHECK(n >= 1 && n <= 8 && (int)pads.size() == n, "1..8 taps");
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  TORCH_CHECK(C % 8 == 0, "bank needs C % 8 == 0");
+  int kg_per = (int)(R * R * C);
+  TORCH_CHECK(kg_per % BK == 0, "bank needs R*R*C % 64 == 0 per tap");
+  TORCH_CHECK(w2d.size(1) == (long)n * kg_per, "bank weight Kg mismatch");
+  int shared_pad = 0;
+  for (long p : pads) shared_pad = std::max(shared_pad, (int)p);
+  TapDesc td = {};
+  td.n = n;
+  td.kg_per = kg_per;
+  int P = -1;
+  for (int t = 0; t < n; ++t) {
+    td.dil[t] = (int)dils[t];
+    td.off[t] = shared_pad - (int)pads[t];
+    int Reff = ((int)R - 1) * (int)dils[t] + 1;
+    int Pt = (H + 2 * (int)pads[t] - Reff) / (int)stride + 1;
+    TORCH_CHECK(P < 0 || Pt == P, "bank taps disagree on output size");
+    P = Pt;
+  }
+  at::Tensor xp = shared_pad > 0
+      ? pad_nhwc(x, shared_pad, shared_pad, shared_pad, shared_pad) : x;
+  int Hp = xp.size(2), Wp = xp.size(3);
+  long long M = (long long)N * P * P;
+  auto y = at::empty({N, ktrue, P, P},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  bool bm64 = ((M + 127) / 128) < 256;
+  int Kg = n * kg_per;
+  auto stream = fedkit_stream();
+  dim3 grid((unsigned)((M + (bm64 ? 63 : 127)) / (bm64 ? 64 : 128)), 1);
+  auto LB = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
+                       (const bf16*)xp.data_ptr(),
+                       (const bf16*)w2d.data_ptr(), (bf16*)y.data_ptr(),
+                       N, Hp, Wp, C, 64, (int)R, (int)R, P, P, Kg, 1,
+                       (int)ktrue, (float*)nullptr, td);
+  };
+  if (stride == 1)
+    bm64 ? LB(conv_fwd_kernel<64, 1, 0, 3, true>)
+         : LB(conv_fwd_kernel<128, 1, 0, 3, true>);
+  else
+    bm64 ? LB(conv_fwd_kernel<64, 2, 0, 3, true>)
+         : LB(conv_fwd_kernel<128, 2, 0, 3, true>);
+  return y;
+}
 
 at::Tensor fedkit_conv_small_fwd(const at::Tensor& x, const at::Tensor& w,
                                  long stride, long padding);  // conv_small.hip

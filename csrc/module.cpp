@@ -71,6 +71,12 @@ at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
                                     long stride, long padding, long R, long S,
                                     long dil);
 
+at::Tensor fedkit_conv2d_dilated_bank(const at::Tensor& x,
+                                      const at::Tensor& w2d,
+                                      std::vector<long> dils,
+                                      std::vector<long> pads, long stride,
+                                      long R, long ktrue);
+
 at::Tensor fedkit_vae_elbo_fwd(const at::Tensor& recon, const at::Tensor& x,
                                const at::Tensor& mu, const at::Tensor& logvar);
 std::vector<at::Tensor> fedkit_vae_elbo_bwd(const at::Tensor& recon,
@@ -160,6 +166,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_weight", &fedkit_conv2d_bwd_weight, "conv bwd-weight",
         py::arg("gy"), py::arg("x"), py::arg("stride"), py::arg("padding"),
         py::arg("R"), py::arg("S"), py::arg("dil") = 1);
+  m.def("conv2d_dilated_bank", &fedkit_conv2d_dilated_bank,
+        "fused multi-dilation conv bank: n taps, one input, one launch "
+        "(block-diagonal combined weight)");
   m.def("vae_elbo_fwd", &fedkit_vae_elbo_fwd,
         "MSE(sum) + analytic KLD in one reduction pass");
   m.def("vae_elbo_bwd", &fedkit_vae_elbo_bwd, "VAE ELBO backward");

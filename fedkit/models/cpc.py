@@ -33,9 +33,22 @@ class EncoderCNN(nn.Module):
         self.conv4 = FedConvGeneric(latent_dim // 2, latent_dim, 4, stride=2, padding=1)       # 4 -> 2
 
     def forward(self, x):
-        bank = torch.cat(
-            (elu(self.conv1_1(x)), elu(self.conv1_2(x)), elu(self.conv1_4(x)),
-             elu(self.conv1_8(x)), elu(self.conv1_16(x))), dim=1)
+        from ..ops import native_enabled
+        from ..ops.conv import dilated_bank
+        mods = (self.conv1_1, self.conv1_2, self.conv1_4, self.conv1_8,
+                self.conv1_16)
+        if native_enabled(x):
+            # fused 5-tap bank: ONE MFMA launch for all dilations
+            # (block-diagonal combined weight); elu(conv_t(x)+b_t) cat ==
+            # elu(bank + b) since elu is elementwise and channels disjoint
+            if x.dtype != torch.bfloat16 and torch.is_autocast_enabled():
+                x = x.to(torch.bfloat16)
+            if x.dtype == torch.bfloat16:
+                bank = elu(dilated_bank(x, mods))
+            else:
+                bank = torch.cat([elu(m(x)) for m in mods], dim=1)
+        else:
+            bank = torch.cat([elu(m(x)) for m in mods], dim=1)
         h = elu(self.conv2(bank))
         h = elu(self.conv3(h))
         h = elu(self.conv4(h))

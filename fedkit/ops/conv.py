@@ -308,6 +308,89 @@ class FedConvTranspose2d(nn.ConvTranspose2d):
         return yb
 
 
+class _DilatedBankFn(torch.autograd.Function):
+    """Fused multi-dilation conv bank forward (one launch for all taps via
+    a block-diagonal combined weight, csrc conv2d_dilated_bank); backward
+    decomposes per tap onto the existing bwd-data / bwd-weight kernels.
+    Forward fusion is the hot win: CPC trains with LBFGS line searches,
+    whose probes are forward-only (SURVEY §2a dilated-bank row)."""
+
+    @staticmethod
+    def forward(ctx, x, w2d, dils, pads, stride, R, ktrue, *ws):
+        y = _ext().conv2d_dilated_bank(x, w2d, list(dils), list(pads),
+                                       stride, R, ktrue)
+        ctx.save_for_backward(x, *ws)
+        ctx.meta = (dils, pads, stride, R)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x = ctx.saved_tensors[0]
+        ws = ctx.saved_tensors[1:]
+        dils, pads, stride, R = ctx.meta
+        ext = _ext()
+        H, W = x.shape[2], x.shape[3]
+        gy = gy.contiguous(memory_format=torch.channels_last)
+        gx = None
+        gws = []
+        kt = ws[0].shape[0]
+        for t, wt in enumerate(ws):
+            gy_t = gy[:, t * kt:(t + 1) * kt].contiguous(
+                memory_format=torch.channels_last)
+            if ctx.needs_input_grad[0]:
+                wtb = wt.to(torch.bfloat16).contiguous(
+                    memory_format=torch.channels_last)
+                d = ext.conv2d_bwd_data(gy_t, wtb, stride, pads[t], H, W,
+                                        dil=dils[t])
+                gx = d if gx is None else gx + d
+            if ctx.needs_input_grad[7 + t]:
+                gws.append(ext.conv2d_bwd_weight(
+                    gy_t, x, stride, pads[t], R, R, dil=dils[t]))
+            else:
+                gws.append(None)
+        return (gx, None, None, None, None, None, None, *gws)
+
+
+def dilated_bank(x, mods):
+    """elu-less fused bank: x through every conv in `mods` (shared input,
+    per-module dilation/padding), outputs channel-concatenated.  Equivalent
+    to torch.cat([m(x) for m in mods], dim=1)."""
+    dils = [m.dilation[0] for m in mods]
+    pads = [m.padding[0] for m in mods]
+    stride = mods[0].stride[0]
+    R = mods[0].kernel_size[0]
+    C8 = _pad_mult(mods[0].in_channels, 8)
+    kt = mods[0].out_channels
+    ktrue = kt * len(mods)
+    xb = _chanpad_nhwc(x, C8)
+
+    m0 = mods[0]
+    vers = tuple(m.weight._version for m in mods)
+    cache = getattr(m0, "_wbank", None)
+    frozen = not any(m.weight.requires_grad for m in mods)
+    if frozen and cache is not None and cache[0] == vers:
+        w2d = cache[1]
+    else:
+        kg = R * R * C8
+        with torch.set_grad_enabled(False):
+            w2d = x.new_zeros(64, kg * len(mods), dtype=torch.bfloat16)
+            for t, m in enumerate(mods):
+                wt = m.weight.detach().to(torch.bfloat16)
+                if wt.shape[1] != C8:
+                    wt = F.pad(wt, (0, 0, 0, 0, 0, C8 - wt.shape[1]))
+                w2d[t * kt:(t + 1) * kt, t * kg:(t + 1) * kg] = \
+                    wt.permute(0, 2, 3, 1).reshape(kt, kg)
+        if frozen:
+            m0._wbank = (vers, w2d)
+    y = _DilatedBankFn.apply(xb, w2d, dils, pads, stride, R, ktrue,
+                             *[m.weight for m in mods])
+    bias = torch.cat([m.bias for m in mods]) if mods[0].bias is not None \
+        else None
+    if bias is not None:
+        y = y + bias.to(y.dtype).view(1, -1, 1, 1)
+    return y
+
+
 class FedConv2d(nn.Conv2d):
     def forward(self, x):
         if _native(x) and self.bias is None and self.groups == 1 \
