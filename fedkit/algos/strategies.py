@@ -123,8 +123,12 @@ class FedProx(Strategy):
     name = "fedprox"
     uses_penalty = True
 
-    def __init__(self, rho0: float = 1.0):
+    def __init__(self, rho0: float = 1.0, warmup_rounds: int = 0):
         self.rho0 = rho0
+        # engine extension, default 0 = reference behavior (see
+        # ConsensusADMM.warmup_rounds: z starts at 0 per block, so the
+        # round-0 proximal term pulls toward an uninformed target)
+        self.warmup_rounds = warmup_rounds
         # per-(block, client) proximal weight table, lazily filled at rho0.
         # The reference allocates rho = torch.ones(L,3)*admm_rho0
         # (fedprox_multi.py:142, "per layer, per slave") but only ever READS
@@ -146,6 +150,8 @@ class FedProx(Strategy):
         return st
 
     def penalty(self, state, ck, xvec):
+        if state.get("round", 0) < self.warmup_rounds:
+            return None
         xdelta = xvec - state["z"]
         rho = self._rho(state.get("block_idx", 0), ck)
         return 0.5 * rho * (torch.norm(xdelta, 2) ** 2)
@@ -170,6 +176,7 @@ class FedProx(Strategy):
                       for ck, xv in x.items()}
         rows = comm.gather_scalar_rows(per_client)
         primal_residual = float(rows[:, 0].sum()) / N
+        state["round"] = round_idx + 1
         return {"dual": dual_residual, "primal": primal_residual, "rho": rho}
 
 
@@ -178,9 +185,20 @@ class ConsensusADMM(Strategy):
     name = "admm"
     uses_penalty = True
 
-    def __init__(self, rho0: float = 0.1, bb: Optional[BBConfig] = None):
+    def __init__(self, rho0: float = 0.1, bb: Optional[BBConfig] = None,
+                 warmup_rounds: int = 0):
         self.rho0 = rho0
         self.bb = bb or BBConfig()
+        # Engine extension (default 0 = reference behavior): skip the
+        # augmented-Lagrangian penalty for the first `warmup_rounds`
+        # communication rounds of each block.  Every block restarts with
+        # z = 0 (consensus_multi.py:151-154), so round 0's penalty pulls
+        # the block toward an UNINFORMED zero target; deferring it until z
+        # is a real consensus (round >= 1) removes that damage without
+        # changing the z-/y-updates (round-0 z = mean(x) regardless since
+        # y = 0).  Measured on the synthetic protocol: see
+        # profiles/acc_synthetic.md.
+        self.warmup_rounds = warmup_rounds
 
     def init_block(self, comm, N, device, x0=None, block_idx=0):
         st = super().init_block(comm, N, device)
@@ -199,6 +217,8 @@ class ConsensusADMM(Strategy):
 
     def penalty(self, state, ck, xvec):
         """y^T (x-z) + (rho/2)||x-z||^2  (consensus_multi.py:209-218)."""
+        if state.get("round", 0) < self.warmup_rounds:
+            return None
         xdelta = xvec - state["z"]
         return torch.dot(state["y"][ck], xdelta) \
             + 0.5 * state["rho"] * (torch.norm(xdelta, 2) ** 2)
@@ -270,6 +290,7 @@ class ConsensusADMM(Strategy):
             state["y"][ck].add_(ydelta)
         rows = comm.gather_scalar_rows(per_client)
         primal_residual = float(rows[:, 0].sum()) / N
+        state["round"] = round_idx + 1
         return {"dual": dual_residual, "primal": primal_residual, "rho": rho}
 
 

@@ -90,6 +90,10 @@ class FedConfig:
     jsonl_path: Optional[str] = None   # structured per-round metrics
     max_steps_per_epoch: int = 0       # 0 = full shard (tests/bench shrink it)
     max_eval_batches: int = 0          # 0 = full test set
+    penalty_warmup_rounds: int = 0     # FedProx/ADMM: skip the penalty for
+                                       # the first N rounds of each block
+                                       # (z starts at 0 per block; 0 =
+                                       # reference behavior)
     round_checkpoint: bool = False     # save s{k}.model + RNG sidecar at each
                                        # (nloop, ci) boundary; load_model=True
                                        # then RESUMES at the recorded round
@@ -109,11 +113,13 @@ class FedConfig:
         if self.strategy == "fedavg":
             return FedAvg()
         if self.strategy == "fedprox":
-            return FedProx(rho0=self.admm_rho0)
+            return FedProx(rho0=self.admm_rho0,
+                           warmup_rounds=self.penalty_warmup_rounds)
         if self.strategy == "admm":
             bb = BBConfig(self.bb_update, self.bb_period_T,
                           self.bb_alphacorrmin, self.bb_epsilon, self.bb_rhomax)
-            return ConsensusADMM(rho0=self.admm_rho0, bb=bb)
+            return ConsensusADMM(rho0=self.admm_rho0, bb=bb,
+                                 warmup_rounds=self.penalty_warmup_rounds)
         raise ValueError(f"unknown strategy {self.strategy!r}")
 
 

@@ -49,6 +49,10 @@ def main():
                     help="comma list of configs to run")
     ap.add_argument("--rho", type=float, default=0.1)
     ap.add_argument("--bb", type=int, default=0)
+    ap.add_argument("--warmup", type=int, default=0,
+                    help="penalty warm-up rounds per block (FedProx/ADMM)")
+    ap.add_argument("--jsonl-dir", type=str, default="",
+                    help="write per-round residual JSONLs here")
     args = ap.parse_args()
     if args.quick:
         args.nloop, args.nadmm, args.max_steps = 1, 1, 3
@@ -81,7 +85,10 @@ def main():
         results["fedavg_K10"] = final_accs(job)
 
     if want("fedprox_K10"):
-        cfg = common(10, args, strategy="fedprox", admm_rho0=args.rho)
+        cfg = common(10, args, strategy="fedprox", admm_rho0=args.rho,
+                     penalty_warmup_rounds=args.warmup)
+        if args.jsonl_dir:
+            cfg.jsonl_path = os.path.join(args.jsonl_dir, "fedprox.jsonl")
         job = FederatedJob(cfg)
         job.run()
         results["fedprox_K10"] = final_accs(job)
@@ -89,7 +96,10 @@ def main():
     if want("admm_K10"):
         # consensus ADMM K=10 (rho0, optionally BB-adaptive)
         cfg = common(10, args, strategy="admm", admm_rho0=args.rho,
-                     bb_update=bool(args.bb))
+                     bb_update=bool(args.bb),
+                     penalty_warmup_rounds=args.warmup)
+        if args.jsonl_dir:
+            cfg.jsonl_path = os.path.join(args.jsonl_dir, "admm.jsonl")
         job = FederatedJob(cfg)
         job.run()
         results["admm_K10"] = final_accs(job)
