@@ -61,6 +61,19 @@ struct TapDesc {
   int off[8];
 };
 
+// Virtual-pad descriptor (bwd-data fast path): instead of materializing
+// dilate_pad(gy) — one extra kernel plus a full write+read of a y-sized
+// buffer per conv, 7%+ of round-1 GPU time — the A stage reads gy
+// DIRECTLY, bounds-checking each 16-B chunk (8 channels of ONE pixel,
+// C % 64 == 0 on every bwd-data shape) and substituting the zero page
+// outside.  The checks are per-SLOT scalars (4 per stage), not per-lane.
+//   hs, ws: source (gy) spatial dims
+//   pl:     top/left dilate-pad amount ((R-1)*dil - pad)
+//   vstr:   zero-insertion stride (the forward conv's stride)
+struct VPadDesc {
+  int hs, ws, pl, vstr;
+};
+
 // LDS byte offset of element (row, k) of a [rows][64] bf16 tile with the
 // chunk-XOR swizzle (16-B chunk index ^ (row & 7)).
 __device__ __forceinline__ int lds_off(int row, int k) {
@@ -82,7 +95,8 @@ __device__ __forceinline__ int lds_off(int row, int k) {
 // occupancy 2 with zero LDS conflicts, so more resident waves may hide
 // more latency than the deeper pipeline) — picked per measurement via
 // FEDKIT_CONV_STAGES.
-template <int BM, int STRIDE, int MODE, int STAGES, bool MT = false>
+template <int BM, int STRIDE, int MODE, int STAGES, bool MT = false,
+          bool VPAD = false>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      const bf16* __restrict__ w,   // [Kout][R*S*C]
@@ -92,7 +106,8 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      int Ktrue /* <= Kout: K rows beyond are zero-padding
                                   (VAE/CPC channel counts), skipped on
                                   store so y needs no unpad pass */,
-                     float* __restrict__ aux, TapDesc td = {}) {
+                     float* __restrict__ aux, TapDesc td = {},
+                     VPadDesc vp = {}) {
   constexpr int AB = BM * BK * 2;          // A tile bytes
   constexpr int BB = BN * BK * 2;          // B tile bytes
   __shared__ char smem[STAGES * (AB + BB)];
@@ -127,6 +142,7 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
   // source element (row, k8 = (d%8) ^ (row&7)).
   long long a_rowbase[A_SLOTS];   // &xp[n][p*STRIDE][q*STRIDE][0] offset
   int a_k8[A_SLOTS];
+  int a_p[A_SLOTS], a_q[A_SLOTS];          // VPAD: virtual-coord row origin
 #pragma unroll
   for (int i = 0; i < A_SLOTS; ++i) {
     int d = i * 256 + tid;
@@ -136,7 +152,14 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
     int q = (int)(m % Q);
     int p = (int)((m / Q) % P);
     int n = (int)(m / ((long long)P * Q));
-    a_rowbase[i] = (((long long)n * Hp + p * STRIDE) * Wp + q * STRIDE) * C;
+    if (VPAD) {
+      a_rowbase[i] = (long long)n * vp.hs * vp.ws * C;   // n plane base
+      a_p[i] = p * STRIDE;
+      a_q[i] = q * STRIDE;
+    } else {
+      a_rowbase[i] = (((long long)n * Hp + p * STRIDE) * Wp + q * STRIDE) * C;
+      a_p[i] = a_q[i] = 0;
+    }
     a_k8[i] = (d & 7) ^ (row & 7);
   }
   long long b_rowbase[B_SLOTS];
@@ -163,12 +186,18 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
     // sequence per slot right in the hot loop.
     long long tap_off = 0;
     int c0 = 0;
+    int vh = 0, vw = 0;                    // VPAD: tap offset in virtual coords
     if (c64) {
       int rs = (kt * BK) / C;
       c0 = kt * BK - rs * C;
       int s = rs % S;
       int r = rs / S;
-      tap_off = ((long long)r * dil * Wp + s * dil) * C;
+      if (VPAD) {
+        vh = r * dil - vp.pl;
+        vw = s * dil - vp.pl;
+      } else {
+        tap_off = ((long long)r * dil * Wp + s * dil) * C;
+      }
     }
     // multi-tap: the whole 64-k tile lies in ONE tap (kg_per % 64 == 0),
     // so the tap lookup is SCALAR per stage
@@ -182,7 +211,22 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
 #pragma unroll
     for (int i = 0; i < A_SLOTS; ++i) {
       const bf16* src;
-      if (c64) {
+      if (VPAD) {
+        // bounds + stride-validity per SLOT (scalar); zero page outside.
+        // C % 64 == 0 on every bwd-data shape, so (r,s) is stage-scalar.
+        int hv = a_p[i] + vh, wv = a_q[i] + vw;
+        bool ok = hv >= 0 && wv >= 0;
+        int hsrc = hv, wsrc = wv;
+        if (vp.vstr == 2) {
+          ok = ok && ((hv | wv) & 1) == 0;
+          hsrc >>= 1;
+          wsrc >>= 1;
+        }
+        ok = ok && hsrc < vp.hs && wsrc < vp.ws;
+        src = ok ? xp + a_rowbase[i] + ((long long)hsrc * vp.ws + wsrc) * C
+                       + c0 + a_k8[i] * 8
+                 : (const bf16*)g_kzero;
+      } else if (c64) {
         // C % 64 == 0 implies Kg % 64 == 0: no tail possible
         src = xp + a_rowbase[i] + tap_off + c0 + a_k8[i] * 8;
       } else {
@@ -841,9 +885,12 @@ void check_conv_inputs(const at::Tensor& x, const at::Tensor& w) {
 // launch on the PRE-PADDED input; pad already folded into Hp/Wp.
 // Ktrue < Kout means the trailing weight rows are channel zero-padding
 // (VAE/CPC shapes): y is allocated and stored dense at Ktrue channels.
+// vpad != nullptr: xp is the RAW gy (bwd-data) and the kernel performs the
+// dilate+pad virtually in the A stage (requires C % 64 == 0, stride 1).
 at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
                      int stride, int P, int Q, int dil = 1, int Ktrue = -1,
-                     float* bnpart = nullptr) {
+                     float* bnpart = nullptr,
+                     const VPadDesc* vpad = nullptr) {
   int N = xp.size(0), C = xp.size(1), Hp = xp.size(2), Wp = xp.size(3);
   int Kout = w_krs_c.size(0), R = w_krs_c.size(2), S = w_krs_c.size(3);
   int Kg = R * S * C;
@@ -884,12 +931,28 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
                        (const bf16*)xp.data_ptr(),
                        (const bf16*)w_krs_c.data_ptr(), (bf16*)y.data_ptr(),
                        N, Hp, Wp, C, Kout, R, S, P, Q, Kg, dil, Ktrue,
-                       aux, TapDesc{});
+                       aux, TapDesc{}, vpad ? *vpad : VPadDesc{});
   };
   static const int stages2 = []() {
     const char* e = getenv("FEDKIT_CONV_STAGES");
     return e && atoi(e) == 2;
   }();
+  if (vpad) {
+    TORCH_CHECK(C % 64 == 0 && stride == 1 && !bnpart,
+                "vpad path needs C % 64 == 0, stride 1");
+    if (splits > 1) {
+      bm64 ? L(conv_fwd_kernel<64, 1, 2, 3, false, true>)
+           : L(conv_fwd_kernel<128, 1, 2, 3, false, true>);
+      long long Ly = M * Kout;
+      hipLaunchKernelGGL(colsum_to_bf16_kernel, dim3(grid_1d(Ly, 256)),
+                         dim3(256), 0, stream, part.data_ptr<float>(),
+                         splits, Ly, (bf16*)y.data_ptr());
+    } else {
+      bm64 ? L(conv_fwd_kernel<64, 1, 0, 3, false, true>)
+           : L(conv_fwd_kernel<128, 1, 0, 3, false, true>);
+    }
+    return y;
+  }
   if (splits > 1) {
     if (stride == 1)
       bm64 ? L(conv_fwd_kernel<64, 1, 2, 3>) : L(conv_fwd_kernel<128, 1, 2, 3>);
@@ -973,7 +1036,7 @@ at::Tensor fedkit_conv2d_dilated_bank(const at::Tensor& x,
                        (const bf16*)xp.data_ptr(),
                        (const bf16*)w2d.data_ptr(), (bf16*)y.data_ptr(),
                        N, Hp, Wp, C, 64, (int)R, (int)R, P, P, Kg, 1,
-                       (int)ktrue, (float*)nullptr, td);
+                       (int)ktrue, (float*)nullptr, td, VPadDesc{});
   };
   if (stride == 1)
     bm64 ? LB(conv_fwd_kernel<64, 1, 0, 3, true>)
@@ -1104,9 +1167,22 @@ at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
     return dilate_pad_nhwc(dxs, 0, pb, 0, pr, 2);
   }
   // dilate+pad gy: pl = (R-1)*dil - pad, pr = pl + a with
-  // a = (H + 2p - Reff) % stride, Reff = (R-1)*dil + 1
+  // a = (H + 2p - Reff) % stride, Reff = (R-1)*dil + 1.
+  // Fast path: the pad/dilation happens VIRTUALLY in the conv kernel's A
+  // stage (zero-page substitution per out-of-range chunk) — no gyp buffer,
+  // no dilate_pad launch, no extra y-sized HBM round trip.
   int Reff = (R - 1) * (int)dil + 1;
   int pl = (Reff - 1) - (int)padding;
+  static const bool no_vpad = []() {
+    const char* e = getenv("FEDKIT_NO_VPAD");
+    return e && atoi(e) == 1;
+  }();
+  int K64 = gy.size(1);
+  if (!no_vpad && K64 % 64 == 0 && pl >= 0) {
+    VPadDesc vp{(int)gy.size(2), (int)gy.size(3), pl, (int)stride};
+    return conv_core(gy, wrot, 1, (int)H, (int)W, (int)dil, (int)ctrue,
+                     nullptr, &vp);
+  }
   int a = (int)((H + 2 * padding - Reff) % stride);
   at::Tensor gyp = dilate_pad_nhwc(gy, pl, pl + a, pl, pl + a, (int)stride);
   return conv_core(gyp, wrot, 1, (int)H, (int)W, (int)dil, (int)ctrue);
