@@ -153,9 +153,13 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
     int p = (int)((m / Q) % P);
     int n = (int)(m / ((long long)P * Q));
     if (VPAD) {
-      a_rowbase[i] = (long long)n * vp.hs * vp.ws * C;   // n plane base
-      a_p[i] = p * STRIDE;
-      a_q[i] = q * STRIDE;
+      // vstr == 1 only: base points at gy pixel (p - pl, q - pl); the
+      // stage adds a SCALAR tap offset and the per-slot work is just
+      // int compares against stage-scalar bounds — no muls in the loop
+      a_rowbase[i] =
+          (((long long)n * vp.hs + (p - vp.pl)) * vp.ws + (q - vp.pl)) * C;
+      a_p[i] = p - vp.pl;
+      a_q[i] = q - vp.pl;
     } else {
       a_rowbase[i] = (((long long)n * Hp + p * STRIDE) * Wp + q * STRIDE) * C;
       a_p[i] = a_q[i] = 0;
@@ -186,15 +190,16 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
     // sequence per slot right in the hot loop.
     long long tap_off = 0;
     int c0 = 0;
-    int vh = 0, vw = 0;                    // VPAD: tap offset in virtual coords
+    int vh = 0, vw = 0;                    // VPAD: tap offset (dil applied)
     if (c64) {
       int rs = (kt * BK) / C;
       c0 = kt * BK - rs * C;
       int s = rs % S;
       int r = rs / S;
       if (VPAD) {
-        vh = r * dil - vp.pl;
-        vw = s * dil - vp.pl;
+        vh = r * dil;
+        vw = s * dil;
+        tap_off = ((long long)vh * vp.ws + vw) * C;
       } else {
         tap_off = ((long long)r * dil * Wp + s * dil) * C;
       }
@@ -212,19 +217,12 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
     for (int i = 0; i < A_SLOTS; ++i) {
       const bf16* src;
       if (VPAD) {
-        // bounds + stride-validity per SLOT (scalar); zero page outside.
-        // C % 64 == 0 on every bwd-data shape, so (r,s) is stage-scalar.
+        // per-slot: 4 int compares vs stage-scalar bounds, then select.
+        // C % 64 == 0 on every bwd-data shape, so (r,s) is stage-scalar
+        // and the address is precomputed-base + scalar tap_off.
         int hv = a_p[i] + vh, wv = a_q[i] + vw;
-        bool ok = hv >= 0 && wv >= 0;
-        int hsrc = hv, wsrc = wv;
-        if (vp.vstr == 2) {
-          ok = ok && ((hv | wv) & 1) == 0;
-          hsrc >>= 1;
-          wsrc >>= 1;
-        }
-        ok = ok && hsrc < vp.hs && wsrc < vp.ws;
-        src = ok ? xp + a_rowbase[i] + ((long long)hsrc * vp.ws + wsrc) * C
-                       + c0 + a_k8[i] * 8
+        bool ok = hv >= 0 && wv >= 0 && hv < vp.hs && wv < vp.ws;
+        src = ok ? xp + a_rowbase[i] + tap_off + c0 + a_k8[i] * 8
                  : (const bf16*)g_kzero;
       } else if (c64) {
         // C % 64 == 0 implies Kg % 64 == 0: no tail possible
@@ -461,6 +459,44 @@ void transpose_mk_kernel(const bf16* __restrict__ in,  // [M][K]
   }
 }
 
+// Packed-Q plane transpose (layer4-class dw, Q == 4): out[k][m'] where
+// m' walks an [Nh][4] plane and the source pixel is
+//   m_in = (m' >> 2) * rowstride + rowoff + (m' & 3) * cs
+// i.e. 4 filter-column-aligned pixels per source row, rows subsampled by
+// rowstride (h-parity planes for stride 2).  After this, an 8-m' chunk =
+// TWO consecutive plane rows = the two output rows a Q=4 m-chunk spans —
+// contiguous 16 B, so the dw GEMM's glds staging applies unchanged.
+__global__ __launch_bounds__(256)
+void transpose_pack_kernel(const bf16* __restrict__ in,  // [Mfull][K]
+                           bf16* __restrict__ out,       // [K][Mp]
+                           long long Mp, int K, long long rowstride,
+                           long long rowoff, int cs) {
+  __shared__ char smem[64 * 128];
+  const int tid = threadIdx.x;
+  const long long mt = (long long)blockIdx.x * 64;
+  const int kt = blockIdx.y * 64;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {            // load: row = m', 8 k per lane
+    int d = i * 256 + tid;
+    int row = d >> 3, c = d & 7;
+    long long mp = mt + row;
+    long long m_in = (mp >> 2) * rowstride + rowoff + (mp & 3) * cs;
+    *(bf16x8*)(smem + tr_off(row, c * 8)) =
+        *(const bf16x8*)(in + m_in * K + kt + c * 8);
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {            // store: row = k, 8 m' per lane
+    int d = i * 256 + tid;
+    int kr = d >> 3, c = d & 7;
+    bf16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      v[j] = *(const bf16*)(smem + tr_off(c * 8 + j, kr));
+    *(bf16x8*)(out + (long long)(kt + kr) * Mp + mt + c * 8) = v;
+  }
+}
+
 // Vectorized bwd-data weight rotation: wrot[c][R-1-r][S-1-s][k] =
 // w[k][r][s][c] — i.e. the 64x64 LDS transpose above with the OUTPUT row
 // index permuted ((r,s,c) -> (c, R-1-r, S-1-s)).  The scalar rot_weight
@@ -523,7 +559,17 @@ void rot_weight64_kernel(const bf16* __restrict__ w,   // [K][R*S*C]
 // output rows (q 0..3 of p and of p+1), staged as two 8-byte register
 // loads — LDS-DMA cannot compose two pieces into one 16-B/lane image, so
 // these shapes force the register-staging commit path.
-template <int BMK, int BGLDS, int STRIDE, int NT = 1, int QH = 0>
+// QP != 0: packed-Q planes (Q == 4 shapes).  xpT/xpT2 are unused; the B
+// operand reads from ps.p[...]: QP=1 (fwd stride 1) plane s, row r;
+// QP=2 (fwd stride 2) plane s*2 + (r&1), row r>>1 (h-parity split).  In
+// both cases plane layout is [C][N][Hp_pl][4] and the m-chunk addressing
+// is the plain STRIDE=1 form with Wp = 4 — the plane construction
+// absorbed the filter column and the stride.
+struct PlaneSet {
+  const bf16* p[6];
+};
+
+template <int BMK, int BGLDS, int STRIDE, int NT = 1, int QH = 0, int QP = 0>
 __global__ __launch_bounds__(256)
 void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
                     const bf16* __restrict__ xpT,  // [C][N][Hp][Wp(lane)]
@@ -532,7 +578,8 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
                     int K, int C, int N, int Hp, int Wp, int S,
                     long long M, int RSC, int mtiles_per_split,
                     int qshift /* log2 Q */, int qmask,
-                    int pshift /* log2 (P*Q) */, int pmask) {
+                    int pshift /* log2 (P*Q) */, int pmask,
+                    PlaneSet ps = {}) {
   constexpr int BNT = BN * NT;             // rsc tile rows
   constexpr int AB = BMK * BK * 2;         // A tile bytes
   constexpr int BB = BNT * BK * 2;         // B tile bytes
@@ -575,7 +622,13 @@ void dw_gemm_kernel(const bf16* __restrict__ dyT,  // [K][M]
     int rs = rsc / C;
     int s = rs % S;
     int r = rs / S;
-    if (STRIDE == 2) {
+    if (QP == 1) {
+      b_plane[i] = ps.p[s];
+      b_base[i] = ((long long)c * N * Hp + r) * Wp;
+    } else if (QP == 2) {
+      b_plane[i] = ps.p[s * 2 + (r & 1)];
+      b_base[i] = ((long long)c * N * Hp + (r >> 1)) * Wp;
+    } else if (STRIDE == 2) {
       b_plane[i] = (s & 1) ? xpT2 : xpT;
       b_base[i] = ((long long)c * N * Hp + r) * Wp + (s >> 1);
     } else {
@@ -1178,8 +1231,11 @@ at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
     return e && atoi(e) == 1;
   }();
   int K64 = gy.size(1);
-  if (!no_vpad && K64 % 64 == 0 && pl >= 0) {
-    VPadDesc vp{(int)gy.size(2), (int)gy.size(3), pl, (int)stride};
+  if (!no_vpad && K64 % 64 == 0 && pl >= 0 && stride == 1) {
+    // stride 1 only: the no-mul compare form needs an affine gy address;
+    // stride-2's parity/shift addressing was measured SLOWER than the
+    // materialized dilate (per-slot 64-bit muls delay the DMA issue)
+    VPadDesc vp{(int)gy.size(2), (int)gy.size(3), pl, 1};
     return conv_core(gy, wrot, 1, (int)H, (int)W, (int)dil, (int)ctrue,
                      nullptr, &vp);
   }
@@ -1275,7 +1331,7 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
                          stride == 2 ? (const bf16*)xpT2.data_ptr() : nullptr,
                          part.data_ptr<float>(),
                          K, C, N, Hp, Wlane, S, M, (int)RSC, mps,
-                         qshift, Q - 1, pshift, P - 1);
+                         qshift, Q - 1, pshift, P - 1, PlaneSet{});
     };
     if (qh) {
       // Q == 4 shapes take the two-piece register commit (see kernel doc)
@@ -1317,6 +1373,84 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
                          dim3(256), 0, stream, part.data_ptr<float>(), splits,
                          L, (bf16*)dw.data_ptr());
     }
+    return dw;
+  }
+
+  // ---- packed-Q plane path (layer4-class shapes, Q == 4): the 17/17-
+  // native frontier VERDICT r1 #5 names.  An 8-m chunk spans TWO output
+  // rows; per-(filter-column[, h-parity]) planes of width exactly Q make
+  // it 16 contiguous bytes, so the same glds-staged MFMA dw GEMM applies.
+  static const bool no_qp = []() {
+    const char* e = getenv("FEDKIT_NO_QP");
+    return e && atoi(e) == 1;
+  }();
+  bool qp_ok = !no_qp && dil == 1 && pow2 && Q == 4 && (P % 2) == 0 &&
+      K % 64 == 0 && RSC % 64 == 0 && C % 64 == 0 && M % 64 == 0 &&
+      (stride == 1 || (stride == 2 && Hp % 2 == 0)) && S <= 3;
+  if (qp_ok) {
+    auto stream = fedkit_stream();
+    auto dyT = at::empty({(long long)K, M}, gy.options());
+    hipLaunchKernelGGL(transpose_mk_kernel, dim3((unsigned)(M / 64), K / 64),
+                       dim3(256), 0, stream, (const bf16*)gy.data_ptr(),
+                       (bf16*)dyT.data_ptr(), M, K, 1LL, 0LL);
+    int Hpl = stride == 1 ? Hp : Hp / 2;
+    long long Mp = (long long)N * Hpl * 4;
+    int nplanes = stride == 1 ? S : 2 * S;
+    TORCH_CHECK(Mp % 64 == 0, "packed-Q plane Mp % 64");
+    auto planes = at::empty({nplanes, (long long)C, Mp}, xp.options());
+    PlaneSet ps = {};
+    for (int t = 0; t < nplanes; ++t) {
+      long long rowstride, rowoff;
+      int cs;
+      if (stride == 1) {                   // plane t = filter column s
+        rowstride = Wp;
+        rowoff = t;
+        cs = 1;
+      } else {                             // plane t = (s, rpar)
+        int s = t / 2, rpar = t % 2;
+        rowstride = 2LL * Wp;
+        rowoff = (long long)rpar * Wp + s;
+        cs = 2;
+      }
+      bf16* outp = (bf16*)planes.data_ptr() + (long long)t * C * Mp;
+      hipLaunchKernelGGL(transpose_pack_kernel,
+                         dim3((unsigned)(Mp / 64), C / 64), dim3(256), 0,
+                         stream, (const bf16*)xp.data_ptr(), outp, Mp, C,
+                         rowstride, rowoff, cs);
+      ps.p[t] = outp;
+    }
+    int BMK = K % 128 == 0 ? 128 : 64;
+    int rsc_tiles = (int)(RSC / 64);
+    long long mtiles = M / 64;
+    long long tiles_xy = (long long)rsc_tiles * (K / BMK);
+    int splits = 1;
+    while (splits < 64 && tiles_xy * splits < 512 &&
+           (long long)splits * 2 <= mtiles)
+      splits *= 2;
+    int mps = (int)((mtiles + splits - 1) / splits);
+    auto part = at::empty({splits, (long long)K, RSC},
+                          xp.options().dtype(at::kFloat));
+    auto dw = at::empty({K, C, R, S},
+                        xp.options().memory_format(at::MemoryFormat::ChannelsLast));
+    int qshift = 2, pshift = 2 + __builtin_ctz((unsigned)P);
+    dim3 grid((unsigned)rsc_tiles, K / BMK, splits);
+    auto LQ = [&](auto kern) {
+      hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream,
+                         (const bf16*)dyT.data_ptr(), (const bf16*)nullptr,
+                         (const bf16*)nullptr, part.data_ptr<float>(),
+                         K, C, N, Hpl, 4, S, M, (int)RSC, mps,
+                         qshift, 3, pshift, P - 1, ps);
+    };
+    if (stride == 1)
+      BMK == 128 ? LQ(dw_gemm_kernel<128, 1, 1, 1, 0, 1>)
+                 : LQ(dw_gemm_kernel<64, 1, 1, 1, 0, 1>);
+    else
+      BMK == 128 ? LQ(dw_gemm_kernel<128, 1, 1, 1, 0, 2>)
+                 : LQ(dw_gemm_kernel<64, 1, 1, 1, 0, 2>);
+    long long L = (long long)K * RSC;
+    hipLaunchKernelGGL(colsum_to_bf16_kernel, dim3(grid_1d(L, 256)),
+                       dim3(256), 0, stream, part.data_ptr<float>(), splits,
+                       L, (bf16*)dw.data_ptr());
     return dw;
   }
 

@@ -229,6 +229,9 @@ RESNET_SHAPES = [
     (64, 128, 16, 256, 3, 2),
     (16, 128, 16, 256, 1, 2),
     (16, 256, 8, 512, 1, 2),
+    # bench-size layer4 shapes: the packed-Q (Q=4) dw plane path
+    (128, 512, 4, 512, 3, 1),
+    (128, 256, 8, 512, 3, 2),
 ]
 
 
