@@ -1376,15 +1376,20 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
     return dw;
   }
 
-  // ---- packed-Q plane path (layer4-class shapes, Q == 4): the 17/17-
-  // native frontier VERDICT r1 #5 names.  An 8-m chunk spans TWO output
-  // rows; per-(filter-column[, h-parity]) planes of width exactly Q make
-  // it 16 contiguous bytes, so the same glds-staged MFMA dw GEMM applies.
-  static const bool no_qp = []() {
-    const char* e = getenv("FEDKIT_NO_QP");
+  // ---- packed-Q plane path (layer4-class shapes, Q == 4): per-
+  // (filter-column[, h-parity]) transposed planes of width exactly Q make
+  // an 8-m chunk 16 contiguous bytes, so the same glds-staged MFMA dw
+  // GEMM applies.  MEASURED SLOWER than the im2col + rocBLAS split-K path
+  // it would replace (layer4 s1 44.9 vs 40.3 us, s2-C256 45.3 vs 34.0,
+  // gpurun_out/ab_*.log): at M = 2048 the extra plane transposes plus an
+  // underfilled 288-wg GEMM grid lose to the library's tiling — same
+  // conclusion as round 1's two-piece register variant (67 us).  Kept
+  // opt-in (FEDKIT_QP=1) as the measured ledger entry.
+  static const bool use_qp = []() {
+    const char* e = getenv("FEDKIT_QP");
     return e && atoi(e) == 1;
   }();
-  bool qp_ok = !no_qp && dil == 1 && pow2 && Q == 4 && (P % 2) == 0 &&
+  bool qp_ok = use_qp && dil == 1 && pow2 && Q == 4 && (P % 2) == 0 &&
       K % 64 == 0 && RSC % 64 == 0 && C % 64 == 0 && M % 64 == 0 &&
       (stride == 1 || (stride == 2 && Hp % 2 == 0)) && S <= 3;
   if (qp_ok) {

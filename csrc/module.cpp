@@ -71,6 +71,13 @@ at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
                                     long stride, long padding, long R, long S,
                                     long dil);
 
+at::Tensor fedkit_linear_fwd(const at::Tensor& x, const at::Tensor& w,
+                             const c10::optional<at::Tensor>& bias);
+at::Tensor fedkit_linear_bwd_data(const at::Tensor& gy, const at::Tensor& w);
+std::vector<at::Tensor> fedkit_linear_bwd_weight(const at::Tensor& gy,
+                                                 const at::Tensor& x,
+                                                 bool want_bias);
+
 at::Tensor fedkit_conv2d_dilated_bank(const at::Tensor& x,
                                       const at::Tensor& w2d,
                                       std::vector<long> dils,
@@ -166,6 +173,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_weight", &fedkit_conv2d_bwd_weight, "conv bwd-weight",
         py::arg("gy"), py::arg("x"), py::arg("stride"), py::arg("padding"),
         py::arg("R"), py::arg("S"), py::arg("dil") = 1);
+  m.def("linear_fwd", &fedkit_linear_fwd,
+        "fused linear fwd (x [M,K], w [N,K], bias) -> [M,N]",
+        py::arg("x"), py::arg("w"), py::arg("bias") = c10::nullopt);
+  m.def("linear_bwd_data", &fedkit_linear_bwd_data, "linear bwd-data");
+  m.def("linear_bwd_weight", &fedkit_linear_bwd_weight,
+        "linear bwd-weight (+bias): fp32 grads",
+        py::arg("gy"), py::arg("x"), py::arg("want_bias") = true);
   m.def("conv2d_dilated_bank", &fedkit_conv2d_dilated_bank,
         "fused multi-dilation conv bank: n taps, one input, one launch "
         "(block-diagonal combined weight)");

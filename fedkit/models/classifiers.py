@@ -13,6 +13,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.conv import FedConvGeneric
+from ..ops.linear import FedLinear
 from ..ops.elu import elu
 from ..ops.pool import FedMaxPool2d
 
@@ -25,9 +26,9 @@ class Net(nn.Module):
         self.conv1 = FedConvGeneric(3, 6, 5)
         self.pool = FedMaxPool2d(2, 2)
         self.conv2 = FedConvGeneric(6, 16, 5)
-        self.fc1 = nn.Linear(16 * 5 * 5, 120)
-        self.fc2 = nn.Linear(120, 84)
-        self.fc3 = nn.Linear(84, 10)
+        self.fc1 = FedLinear(16 * 5 * 5, 120)
+        self.fc2 = FedLinear(120, 84)
+        self.fc3 = FedLinear(84, 10)
 
     def forward(self, x):
         x = self.pool(elu(self.conv1(x)))
@@ -65,8 +66,8 @@ class Net1(nn.Module):
         self.conv4 = FedConvGeneric(64, 64, 3)
         self.pool1 = FedMaxPool2d(2, 2)
         self.pool2 = FedMaxPool2d(2, 2)
-        self.fc1 = nn.Linear(64 * 5 * 5, 512)
-        self.fc2 = nn.Linear(512, 10)
+        self.fc1 = FedLinear(64 * 5 * 5, 512)
+        self.fc2 = FedLinear(512, 10)
 
     def forward(self, x):
         x = elu(self.conv1(x))            # 32x32 -> 30x30
@@ -101,11 +102,11 @@ class Net2(nn.Module):
         self.pool2 = FedMaxPool2d(2, 2)
         self.pool3 = FedMaxPool2d(2, 2)
         self.pool4 = FedMaxPool2d(2, 2)
-        self.fc1 = nn.Linear(512 * 2 * 2, 128)
-        self.fc2 = nn.Linear(128, 256)
-        self.fc3 = nn.Linear(256, 512)
-        self.fc4 = nn.Linear(512, 1024)
-        self.fc5 = nn.Linear(1024, 10)
+        self.fc1 = FedLinear(512 * 2 * 2, 128)
+        self.fc2 = FedLinear(128, 256)
+        self.fc3 = FedLinear(256, 512)
+        self.fc4 = FedLinear(512, 1024)
+        self.fc5 = FedLinear(1024, 10)
 
     def forward(self, x):
         x = self.pool1(elu(self.conv1(x)))  # 32 -> 16

@@ -18,6 +18,7 @@ partition) is identical to the reference's.
 import torch.nn as nn
 
 from ..ops.conv import FedConv2d
+from ..ops.linear import FedLinear
 from ..ops.pool import avg_pool2d
 from ..ops.norm import FedBatchNorm2d, bn_elu
 
@@ -88,7 +89,7 @@ class ResNet(nn.Module):
         self.layer2 = self._make_layer(block, 128, num_blocks[1], 2)
         self.layer3 = self._make_layer(block, 256, num_blocks[2], 2)
         self.layer4 = self._make_layer(block, 512, num_blocks[3], 2)
-        self.linear = nn.Linear(512 * block.expansion, num_classes)
+        self.linear = FedLinear(512 * block.expansion, num_classes)
 
     def _make_layer(self, block, planes, n, stride):
         layers = []

@@ -22,6 +22,7 @@ import torch
 import torch.nn as nn
 
 from ..ops.conv import FedConvGeneric, FedConvTranspose2d
+from ..ops.linear import FedLinear
 import torch.nn.functional as F
 
 from ..ops.elu import elu
@@ -37,10 +38,10 @@ class AutoEncoderCNN(nn.Module):
         self.conv2 = FedConvGeneric(12, 24, 4, stride=2, padding=1)   # 16 -> 8
         self.conv3 = FedConvGeneric(24, 48, 4, stride=2, padding=1)   # 8 -> 4
         self.conv4 = FedConvGeneric(48, 96, 4, stride=2, padding=1)   # 4 -> 2
-        self.fc1 = nn.Linear(384, 16)
-        self.fc21 = nn.Linear(16, self.latent_dim)
-        self.fc22 = nn.Linear(16, self.latent_dim)
-        self.fc3 = nn.Linear(self.latent_dim, 384)
+        self.fc1 = FedLinear(384, 16)
+        self.fc21 = FedLinear(16, self.latent_dim)
+        self.fc22 = FedLinear(16, self.latent_dim)
+        self.fc3 = FedLinear(self.latent_dim, 384)
         self.tconv1 = FedConvTranspose2d(96, 48, 4, stride=2, padding=1)
         self.tconv2 = FedConvTranspose2d(48, 24, 4, stride=2, padding=1)
         self.tconv3 = FedConvTranspose2d(24, 12, 4, stride=2, padding=1)
@@ -96,21 +97,21 @@ class AutoEncoderCNNCL(nn.Module):
         self.conv3 = FedConvGeneric(24, 48, 4, stride=2, padding=1)
         self.conv4 = FedConvGeneric(48, 96, 4, stride=2, padding=1)
         # cluster head q(k|x)
-        self.fc11 = nn.Linear(384, 128)
-        self.fc12 = nn.Linear(128, 64)
-        self.fc13 = nn.Linear(64, self.K)
+        self.fc11 = FedLinear(384, 128)
+        self.fc12 = FedLinear(128, 64)
+        self.fc13 = FedLinear(64, self.K)
         # cluster-conditioned encoder q(z|x,k)
-        self.fc21 = nn.Linear(384 + self.K, 128)
-        self.fc22 = nn.Linear(128, 128)
-        self.fc23 = nn.Linear(128, self.L)
-        self.fc24 = nn.Linear(128, self.L)
+        self.fc21 = FedLinear(384 + self.K, 128)
+        self.fc22 = FedLinear(128, 128)
+        self.fc23 = FedLinear(128, self.L)
+        self.fc24 = FedLinear(128, self.L)
         # prior head p(z|k)
-        self.fc14 = nn.Linear(self.K, 64)
-        self.fc15 = nn.Linear(64, 64)
-        self.fc16 = nn.Linear(64, self.L)
-        self.fc17 = nn.Linear(64, self.L)
+        self.fc14 = FedLinear(self.K, 64)
+        self.fc15 = FedLinear(64, 64)
+        self.fc16 = FedLinear(64, self.L)
+        self.fc17 = FedLinear(64, self.L)
         # decoder p(x|z)
-        self.fc25 = nn.Linear(self.L, 384)
+        self.fc25 = FedLinear(self.L, 384)
         self.tconv1 = FedConvTranspose2d(96, 48, 4, stride=2, padding=1)
         self.tconv2 = FedConvTranspose2d(48, 24, 4, stride=2, padding=1)
         self.tconv3 = FedConvTranspose2d(24, 12, 4, stride=2, padding=1)

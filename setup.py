@@ -20,6 +20,7 @@ SOURCES = [
     "csrc/flat_ops.hip",
     "csrc/loss.hip",
     "csrc/losses_extra.hip",
+    "csrc/linear.hip",
     "csrc/batchnorm.hip",
     "csrc/conv2d_mfma.hip",
     "csrc/conv_small.hip",

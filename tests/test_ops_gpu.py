@@ -934,3 +934,62 @@ def test_encoder_bank_path_uses_fusion():
     finally:
         ops._NATIVE_ENV = old
     assert frob_err(lat, lat2) < 5e-2, frob_err(lat, lat2)
+
+
+@pytest.mark.parametrize("M,K,N,bias", [
+    (128, 512, 10, True),      # ResNet18 head
+    (128, 400, 120, True),     # Net fc1 (K % 8 != 0 tail)
+    (128, 84, 10, True),       # Net fc3 (tiny, K % 8 != 0)
+    (128, 1600, 512, True),    # Net1 fc1 (largest)
+    (128, 1024, 10, False),
+])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_linear_kernels_vs_torch(M, K, N, bias, dtype):
+    """csrc/linear.hip fwd/bwd-data/bwd-weight(+bias) vs fp32 F.linear."""
+    ext = _ext()
+    torch.manual_seed(23)
+    x = torch.randn(M, K, device="cuda", dtype=dtype)
+    w = torch.randn(N, K, device="cuda", dtype=dtype) * 0.05
+    b = torch.randn(N, device="cuda") if bias else None
+    y = ext.linear_fwd(x, w, b)
+    want = F.linear(x.float(), w.float(), b)
+    tol = 1e-5 if dtype == torch.float32 else 1e-2
+    assert frob_err(y, want) < tol
+
+    gy = torch.randn(M, N, device="cuda", dtype=dtype)
+    gx = ext.linear_bwd_data(gy, w)
+    outs = ext.linear_bwd_weight(gy, x, bias)
+    xr = x.float().detach().requires_grad_(True)
+    wr = w.float().detach().requires_grad_(True)
+    br = b.detach().requires_grad_(True) if bias else None
+    F.linear(xr, wr, br).backward(gy.float())
+    assert frob_err(gx, xr.grad) < tol * 3
+    assert frob_err(outs[0], wr.grad) < tol * 3
+    if bias:
+        assert frob_err(outs[1], br.grad) < tol * 3
+
+
+def test_fedlinear_module_autograd():
+    """FedLinear module end to end under bf16 vs eager fp32."""
+    from fedkit.ops.linear import FedLinear
+    import fedkit.ops as ops
+    torch.manual_seed(29)
+    m = FedLinear(512, 10).cuda()
+    x = torch.randn(64, 512, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = m(x)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    gx, gw, gb = x.grad.clone(), m.weight.grad.clone(), m.bias.grad.clone()
+    x.grad = m.weight.grad = m.bias.grad = None
+    old = ops._NATIVE_ENV
+    ops._NATIVE_ENV = False
+    try:
+        y2 = m(x)
+        y2.backward(gy)
+    finally:
+        ops._NATIVE_ENV = old
+    assert frob_err(y, y2) < 2e-2
+    assert frob_err(gx, x.grad) < 3e-2
+    assert frob_err(gw, m.weight.grad) < 3e-2
+    assert frob_err(gb, m.bias.grad) < 3e-2

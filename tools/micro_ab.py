@@ -3,7 +3,7 @@
 
 The gates are read once per process (static), so run this twice:
     python tools/micro_ab.py                      # new paths
-    FEDKIT_NO_VPAD=1 FEDKIT_NO_QP=1 python tools/micro_ab.py   # old paths
+    FEDKIT_NO_VPAD=1 python tools/micro_ab.py   # old paths (FEDKIT_QP=1 enables packed-Q dw)
 Prints one line per shape: op, shape, mean us.
 """
 
