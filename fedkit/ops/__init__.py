@@ -29,6 +29,34 @@ _EXT = None
 _EXT_ERR = None
 
 
+class _SyncDebugExt:
+    """Kernel-fault localization (SURVEY §5 sanitizer gap, the MI355X
+    analog of CUDA_LAUNCH_BLOCKING scoped to fedkit's own kernels):
+    FEDKIT_SYNC_DEBUG=1 wraps every native call with a device synchronize
+    + HIP error check, so an async fault (page fault, bad kernarg, OOB
+    glds) surfaces AT the offending op with its name instead of crashing
+    many launches later inside unrelated torch code."""
+
+    def __init__(self, mod):
+        self._mod = mod
+
+    def __getattr__(self, name):
+        fn = getattr(self._mod, name)
+        if not callable(fn):
+            return fn
+
+        def wrapped(*a, **kw):
+            out = fn(*a, **kw)
+            try:
+                torch.cuda.synchronize()
+            except RuntimeError as e:
+                raise RuntimeError(
+                    f"fedkit._C.{name} faulted (FEDKIT_SYNC_DEBUG): {e}"
+                ) from e
+            return out
+        return wrapped
+
+
 def _try_load():
     global _EXT, _EXT_ERR
     if _EXT is not None or _EXT_ERR is not None:
@@ -36,6 +64,8 @@ def _try_load():
     try:
         from .. import _C  # in-tree built extension: fedkit/_C*.so
         _EXT = _C
+        if os.environ.get("FEDKIT_SYNC_DEBUG") == "1":
+            _EXT = _SyncDebugExt(_C)
     except ImportError as e:  # pragma: no cover - exercised only pre-build
         _EXT_ERR = e
 
@@ -77,10 +107,11 @@ from .conv import (FedConv2d, FedConvGeneric,  # noqa: E402
                    FedConvTranspose2d)
 from .pool import FedMaxPool2d, max_pool2d, avg_pool2d  # noqa: E402
 from .norm import FedBatchNorm2d         # noqa: E402
+from .linear import FedLinear            # noqa: E402
 
 __all__ = [
     "ext", "has_ext", "require_ext", "native_enabled",
     "elu", "flat", "losses", "FedConv2d", "FedConvGeneric",
     "FedMaxPool2d", "max_pool2d", "avg_pool2d",
-    "FedConvTranspose2d", "FedBatchNorm2d",
+    "FedConvTranspose2d", "FedBatchNorm2d", "FedLinear",
 ]
