@@ -95,8 +95,15 @@ __device__ __forceinline__ int lds_off(int row, int k) {
 // occupancy 2 with zero LDS conflicts, so more resident waves may hide
 // more latency than the deeper pipeline) — picked per measurement via
 // FEDKIT_CONV_STAGES.
+// VPAD: 0 = pre-materialized input; 1 = virtual pad, stride-1 source
+// (per-slot int compares, affine address); 2 = virtual pad over a
+// stride-2 zero-inserted source (bwd-data of stride-2 convs): the A
+// stage reads the RAW gy — besides killing the dilate_pad launch and
+// buffer, it skips the 3/4-zeros A traffic the materialized dilated
+// image pays.  Parity/bounds per slot via 2 precomputed parity bits and
+// 4 stage-scalar class offsets (no 64-bit muls in the loop).
 template <int BM, int STRIDE, int MODE, int STAGES, bool MT = false,
-          bool VPAD = false>
+          int VPAD = 0>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
                      const bf16* __restrict__ w,   // [Kout][R*S*C]
@@ -152,14 +159,24 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
     int q = (int)(m % Q);
     int p = (int)((m / Q) % P);
     int n = (int)(m / ((long long)P * Q));
-    if (VPAD) {
-      // vstr == 1 only: base points at gy pixel (p - pl, q - pl); the
-      // stage adds a SCALAR tap offset and the per-slot work is just
-      // int compares against stage-scalar bounds — no muls in the loop
+    if (VPAD == 1) {
+      // base points at gy pixel (p - pl, q - pl); the stage adds a
+      // SCALAR tap offset and the per-slot work is just int compares
+      // against stage-scalar bounds — no muls in the loop
       a_rowbase[i] =
           (((long long)n * vp.hs + (p - vp.pl)) * vp.ws + (q - vp.pl)) * C;
       a_p[i] = p - vp.pl;
       a_q[i] = q - vp.pl;
+    } else if (VPAD == 2) {
+      // virtual coords hv = (p - pl) + r*dil map to source row hv/2 when
+      // hv is even; base uses the FLOOR halves, the stage adds a class
+      // offset selected by this slot's parity bits (a_p/a_q carry the
+      // full virtual origin for parity + bounds)
+      int vpp = p - vp.pl, vqq = q - vp.pl;
+      a_rowbase[i] =
+          (((long long)n * vp.hs + (vpp >> 1)) * vp.ws + (vqq >> 1)) * C;
+      a_p[i] = vpp;
+      a_q[i] = vqq;
     } else {
       a_rowbase[i] = (((long long)n * Hp + p * STRIDE) * Wp + q * STRIDE) * C;
       a_p[i] = a_q[i] = 0;
@@ -191,15 +208,32 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
     long long tap_off = 0;
     int c0 = 0;
     int vh = 0, vw = 0;                    // VPAD: tap offset (dil applied)
+    // VPAD == 2 stage scalars: class tap offsets + bounds per parity bit
+    int t00 = 0, t01 = 0, t10 = 0, t11 = 0;
+    int rlo0 = 0, rhi0 = 0, rlo1 = 0, rhi1 = 0;
+    int clo0 = 0, chi0 = 0, clo1 = 0, chi1 = 0;
     if (c64) {
       int rs = (kt * BK) / C;
       c0 = kt * BK - rs * C;
       int s = rs % S;
       int r = rs / S;
-      if (VPAD) {
+      if (VPAD == 1) {
         vh = r * dil;
         vw = s * dil;
         tap_off = ((long long)vh * vp.ws + vw) * C;
+      } else if (VPAD == 2) {
+        vh = r * dil;
+        vw = s * dil;
+        int rt0 = (vh + 0) >> 1, rt1 = (vh + 1) >> 1;
+        int ct0 = (vw + 0) >> 1, ct1 = (vw + 1) >> 1;
+        t00 = (rt0 * vp.ws + ct0) * C + c0;
+        t01 = (rt0 * vp.ws + ct1) * C + c0;
+        t10 = (rt1 * vp.ws + ct0) * C + c0;
+        t11 = (rt1 * vp.ws + ct1) * C + c0;
+        rlo0 = -rt0; rhi0 = vp.hs - rt0;
+        rlo1 = -rt1; rhi1 = vp.hs - rt1;
+        clo0 = -ct0; chi0 = vp.ws - ct0;
+        clo1 = -ct1; chi1 = vp.ws - ct1;
       } else {
         tap_off = ((long long)r * dil * Wp + s * dil) * C;
       }
@@ -216,13 +250,23 @@ void conv_fwd_kernel(const bf16* __restrict__ xp,  // [N][Hp][Wp][C] padded
 #pragma unroll
     for (int i = 0; i < A_SLOTS; ++i) {
       const bf16* src;
-      if (VPAD) {
+      if (VPAD == 1) {
         // per-slot: 4 int compares vs stage-scalar bounds, then select.
         // C % 64 == 0 on every bwd-data shape, so (r,s) is stage-scalar
         // and the address is precomputed-base + scalar tap_off.
         int hv = a_p[i] + vh, wv = a_q[i] + vw;
         bool ok = hv >= 0 && wv >= 0 && hv < vp.hs && wv < vp.ws;
         src = ok ? xp + a_rowbase[i] + tap_off + c0 + a_k8[i] * 8
+                 : (const bf16*)g_kzero;
+      } else if (VPAD == 2) {
+        // parity + bounds per slot, class tap offset by 2 parity bits
+        int pa = a_p[i] & 1, qa = a_q[i] & 1;
+        int ah = a_p[i] >> 1, aq = a_q[i] >> 1;
+        bool ok = (((a_p[i] ^ vh) | (a_q[i] ^ vw)) & 1) == 0 &&
+                  ah >= (pa ? rlo1 : rlo0) && ah < (pa ? rhi1 : rhi0) &&
+                  aq >= (qa ? clo1 : clo0) && aq < (qa ? chi1 : chi0);
+        int off = pa ? (qa ? t11 : t10) : (qa ? t01 : t00);
+        src = ok ? xp + a_rowbase[i] + off + a_k8[i] * 8
                  : (const bf16*)g_kzero;
       } else if (c64) {
         // C % 64 == 0 implies Kg % 64 == 0: no tail possible
@@ -992,17 +1036,26 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
   }();
   if (vpad) {
     TORCH_CHECK(C % 64 == 0 && stride == 1 && !bnpart,
-                "vpad path needs C % 64 == 0, stride 1");
+                "vpad path needs C % 64 == 0, stride-1 conv");
+    bool v2 = vpad->vstr == 2;
     if (splits > 1) {
-      bm64 ? L(conv_fwd_kernel<64, 1, 2, 3, false, true>)
-           : L(conv_fwd_kernel<128, 1, 2, 3, false, true>);
+      if (v2)
+        bm64 ? L(conv_fwd_kernel<64, 1, 2, 3, false, 2>)
+             : L(conv_fwd_kernel<128, 1, 2, 3, false, 2>);
+      else
+        bm64 ? L(conv_fwd_kernel<64, 1, 2, 3, false, 1>)
+             : L(conv_fwd_kernel<128, 1, 2, 3, false, 1>);
       long long Ly = M * Kout;
       hipLaunchKernelGGL(colsum_to_bf16_kernel, dim3(grid_1d(Ly, 256)),
                          dim3(256), 0, stream, part.data_ptr<float>(),
                          splits, Ly, (bf16*)y.data_ptr());
     } else {
-      bm64 ? L(conv_fwd_kernel<64, 1, 0, 3, false, true>)
-           : L(conv_fwd_kernel<128, 1, 0, 3, false, true>);
+      if (v2)
+        bm64 ? L(conv_fwd_kernel<64, 1, 0, 3, false, 2>)
+             : L(conv_fwd_kernel<128, 1, 0, 3, false, 2>);
+      else
+        bm64 ? L(conv_fwd_kernel<64, 1, 0, 3, false, 1>)
+             : L(conv_fwd_kernel<128, 1, 0, 3, false, 1>);
     }
     return y;
   }
@@ -1231,11 +1284,13 @@ at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
     return e && atoi(e) == 1;
   }();
   int K64 = gy.size(1);
-  if (!no_vpad && K64 % 64 == 0 && pl >= 0 && stride == 1) {
-    // stride 1 only: the no-mul compare form needs an affine gy address;
-    // stride-2's parity/shift addressing was measured SLOWER than the
-    // materialized dilate (per-slot 64-bit muls delay the DMA issue)
-    VPadDesc vp{(int)gy.size(2), (int)gy.size(3), pl, 1};
+  if (!no_vpad && K64 % 64 == 0 && pl >= 0 && stride <= 2) {
+    // stride 1: affine gy address + per-slot int compares (first cut
+    // with per-slot 64-bit muls measured SLOWER than the materialized
+    // dilate; this form measured layer1 43.0 -> 29.6 us).
+    // stride 2: parity-class form — also skips the 3/4-zeros A traffic
+    // the materialized zero-inserted image pays.
+    VPadDesc vp{(int)gy.size(2), (int)gy.size(3), pl, (int)stride};
     return conv_core(gy, wrot, 1, (int)H, (int)W, (int)dil, (int)ctrue,
                      nullptr, &vp);
   }

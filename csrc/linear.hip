@@ -78,8 +78,11 @@ void linear_bwd_data_kernel(const T* __restrict__ gy,  // [M][N]
   for (int j = 0; j < kw; ++j) from_f32(acc[j], gx[m * K + k0 + j]);
 }
 
-// gw[n][k8] = sum_m gy[m][n] * x[m][k8..]; k-chunk 0 threads also emit
-// gb[n] = sum_m gy[m][n].  Thread per (n, k-chunk).
+// gw[n][k8] = sum_m gy[m][n] * x[m][k8..]; k-chunk 0 waves also emit
+// gb[n] = sum_m gy[m][n].  One WAVE per (n, k-chunk), lanes split the M
+// (reduction) axis, xor-shuffle combine — grids of N*kc8 THREADS left the
+// chip idle (the [128,512]x[512,10] head made a 640-thread launch that
+// measured 126 us; waves over M give 64x the parallelism).
 template <typename T>
 __global__ __launch_bounds__(256)
 void linear_bwd_weight_kernel(const T* __restrict__ gy,  // [M][N]
@@ -88,21 +91,31 @@ void linear_bwd_weight_kernel(const T* __restrict__ gy,  // [M][N]
                               float* __restrict__ gb,    // [N] fp32 or null
                               long long M, int N, int K) {
   const int kc8 = (K + 7) / 8;
-  const long long e = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (e >= (long long)N * kc8) return;
-  const int n = (int)(e / kc8);
-  const int k0 = (int)(e % kc8) * 8;
+  const long long wid =
+      ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int lane = threadIdx.x & 63;
+  if (wid >= (long long)N * kc8) return;
+  const int n = (int)(wid / kc8);
+  const int k0 = (int)(wid % kc8) * 8;
   const int kw = min(8, K - k0);
   float acc[8] = {};
   float bacc = 0.f;
-  for (long long m = 0; m < M; ++m) {
+  for (long long m = lane; m < M; m += 64) {
     float g = to_f32(gy[m * N + n]);
     const T* xr = x + m * K + k0;
     for (int j = 0; j < kw; ++j) acc[j] += g * to_f32(xr[j]);
-    if (k0 == 0) bacc += g;
+    bacc += g;
   }
-  for (int j = 0; j < kw; ++j) gw[(long long)n * K + k0 + j] = acc[j];
-  if (gb && k0 == 0) gb[n] = bacc;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    for (int off = 32; off > 0; off >>= 1)
+      acc[j] += __shfl_xor(acc[j], off, 64);
+  for (int off = 32; off > 0; off >>= 1)
+    bacc += __shfl_xor(bacc, off, 64);
+  if (lane == 0) {
+    for (int j = 0; j < kw; ++j) gw[(long long)n * K + k0 + j] = acc[j];
+    if (gb && k0 == 0) gb[n] = bacc;
+  }
 }
 
 }  // namespace
@@ -171,7 +184,7 @@ std::vector<at::Tensor> fedkit_linear_bwd_weight(const at::Tensor& gy,
   auto stream = fedkit_stream();
   DISPATCH_F32_BF16(gc, "linear_bwd_weight", {
     hipLaunchKernelGGL((linear_bwd_weight_kernel<scalar_t>),
-                       dim3(grid_1d((long long)N * kc8, 256, 1u << 20)),
+                       dim3(grid_1d((long long)N * kc8 * 64, 256, 1u << 20)),
                        dim3(256), 0, stream,
                        (const scalar_t*)gc.data_ptr(),
                        (const scalar_t*)xc.data_ptr(), gw.data_ptr<float>(),

@@ -46,6 +46,22 @@ def main():
              * 0.05).contiguous(memory_format=torch.channels_last)
         us = timeit(lambda: ext.conv2d_bwd_data(gy, w, 1, 1, H, H))
         print(f"{tag} bwd_data_s1 C{C} H{H} K{K}: {us:8.1f} us")
+    # bwd-data stride-2 shapes (vpad v2: parity-class virtual dilate)
+    for C, H, K in [(64, 32, 128), (128, 16, 256), (256, 8, 512)]:
+        P = H // 2
+        gy = torch.randn(BATCH, K, P, P, device="cuda",
+                         dtype=torch.bfloat16).contiguous(
+            memory_format=torch.channels_last)
+        w = (torch.randn(K, C, 3, 3, device="cuda", dtype=torch.bfloat16)
+             * 0.05).contiguous(memory_format=torch.channels_last)
+        us = timeit(lambda: ext.conv2d_bwd_data(gy, w, 2, 1, H, H))
+        print(f"{tag} bwd_data_s2 C{C} H{H} K{K}: {us:8.1f} us")
+    # linear bwd-weight (the 126 us head case + the big Net1 fc1)
+    for M, K, N in [(128, 512, 10), (128, 1600, 512)]:
+        gy = torch.randn(M, N, device="cuda", dtype=torch.bfloat16)
+        x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+        us = timeit(lambda: ext.linear_bwd_weight(gy, x, True))
+        print(f"{tag} linear_dw M{M} K{K} N{N}: {us:8.1f} us")
     # layer4-class dw shapes (packed-Q)
     for C, H, K, stride in [(512, 4, 512, 1), (256, 8, 512, 2)]:
         P = H // stride
