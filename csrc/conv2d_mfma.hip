@@ -541,6 +541,54 @@ void transpose_pack_kernel(const bf16* __restrict__ in,  // [Mfull][K]
   }
 }
 
+// Batched transpose: the 2-3 operand transposes of ONE dw call (dyT +
+// xpT[+ xpT2]) issued as a SINGLE launch — each costs ~6 us of which most
+// is launch/ramp latency at these sizes (~32 transpose launches/step,
+// ~190 us, round-2 profile).  1-D grid linearizes all jobs' 64x64 tiles.
+struct TransJob {
+  const bf16* in;
+  bf16* out;
+  long long M;
+  int K;
+  long long mscale, moff;
+  long long tile0;          // first linear tile of this job
+};
+
+__global__ __launch_bounds__(256)
+void transpose_mk_batch_kernel(TransJob j0, TransJob j1, TransJob j2,
+                               int njobs, long long total_tiles) {
+  __shared__ char smem[64 * 128];
+  const int tid = threadIdx.x;
+  long long t = blockIdx.x;
+  if (t >= total_tiles) return;
+  TransJob j = j0;
+  if (njobs > 2 && t >= j2.tile0) j = j2;
+  else if (njobs > 1 && t >= j1.tile0) j = j1;
+  long long rel = t - j.tile0;
+  const long long mtiles = j.M / 64;
+  const long long mt = (rel % mtiles) * 64;
+  const int kt = (int)(rel / mtiles) * 64;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int d = i * 256 + tid;
+    int row = d >> 3, c = d & 7;
+    *(bf16x8*)(smem + tr_off(row, c * 8)) =
+        *(const bf16x8*)(j.in + ((mt + row) * j.mscale + j.moff) * j.K +
+                         kt + c * 8);
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int d = i * 256 + tid;
+    int kr = d >> 3, c = d & 7;
+    bf16x8 v;
+#pragma unroll
+    for (int jj = 0; jj < 8; ++jj)
+      v[jj] = *(const bf16*)(smem + tr_off(c * 8 + jj, kr));
+    *(bf16x8*)(j.out + (long long)(kt + kr) * j.M + mt + c * 8) = v;
+  }
+}
+
 // Vectorized bwd-data weight rotation: wrot[c][R-1-r][S-1-s][k] =
 // w[k][r][s][c] — i.e. the 64x64 LDS transpose above with the OUTPUT row
 // index permuted ((r,s,c) -> (c, R-1-r, S-1-s)).  The scalar rot_weight
@@ -1331,31 +1379,35 @@ at::Tensor fedkit_conv2d_bwd_weight_prepadded(const at::Tensor& gy,
       (stride == 2 && common && Wp % 2 == 0 && (NHW / 2) % 64 == 0)) {
     auto stream = fedkit_stream();
     auto dyT = at::empty({(long long)K, M}, gy.options());
-    hipLaunchKernelGGL(transpose_mk_kernel, dim3((unsigned)(M / 64), K / 64),
-                       dim3(256), 0, stream, (const bf16*)gy.data_ptr(),
-                       (bf16*)dyT.data_ptr(), M, K, 1LL, 0LL);
     at::Tensor xpT, xpT2;
     int Wlane = Wp;
+    // all operand transposes of this dw call in ONE launch
+    TransJob jd = {(const bf16*)gy.data_ptr(), (bf16*)dyT.data_ptr(),
+                   M, K, 1LL, 0LL, 0LL};
+    TransJob jx = {}, jx2 = {};
+    int njobs;
+    long long tiles = (M / 64) * (K / 64);
     if (stride == 1) {
       xpT = at::empty({(long long)C, NHW}, xp.options());
-      hipLaunchKernelGGL(transpose_mk_kernel,
-                         dim3((unsigned)(NHW / 64), C / 64), dim3(256), 0,
-                         stream, (const bf16*)xp.data_ptr(),
-                         (bf16*)xpT.data_ptr(), NHW, C, 1LL, 0LL);
+      jx = {(const bf16*)xp.data_ptr(), (bf16*)xpT.data_ptr(),
+            NHW, C, 1LL, 0LL, tiles};
+      tiles += (NHW / 64) * (C / 64);
+      njobs = 2;
     } else {
       Wlane = Wp / 2;
       long long NHW2 = NHW / 2;
       xpT = at::empty({(long long)C, NHW2}, xp.options());
       xpT2 = at::empty({(long long)C, NHW2}, xp.options());
-      hipLaunchKernelGGL(transpose_mk_kernel,
-                         dim3((unsigned)(NHW2 / 64), C / 64), dim3(256), 0,
-                         stream, (const bf16*)xp.data_ptr(),
-                         (bf16*)xpT.data_ptr(), NHW2, C, 2LL, 0LL);
-      hipLaunchKernelGGL(transpose_mk_kernel,
-                         dim3((unsigned)(NHW2 / 64), C / 64), dim3(256), 0,
-                         stream, (const bf16*)xp.data_ptr(),
-                         (bf16*)xpT2.data_ptr(), NHW2, C, 2LL, 1LL);
+      jx = {(const bf16*)xp.data_ptr(), (bf16*)xpT.data_ptr(),
+            NHW2, C, 2LL, 0LL, tiles};
+      tiles += (NHW2 / 64) * (C / 64);
+      jx2 = {(const bf16*)xp.data_ptr(), (bf16*)xpT2.data_ptr(),
+             NHW2, C, 2LL, 1LL, tiles};
+      tiles += (NHW2 / 64) * (C / 64);
+      njobs = 3;
     }
+    hipLaunchKernelGGL(transpose_mk_batch_kernel, dim3((unsigned)tiles),
+                       dim3(256), 0, stream, jd, jx, jx2, njobs, tiles);
     int qshift = __builtin_ctz((unsigned)Q);
     int pshift = qshift + __builtin_ctz((unsigned)P);
     int BMK = K % 128 == 0 ? 128 : 64;     // kout tile (128 ~1.7x faster)

@@ -49,8 +49,15 @@ class _LinearFn(torch.autograd.Function):
 
 
 class FedLinear(nn.Linear):
+    # Above this weight size a linear is a plain MFMA-shaped GEMM and the
+    # library is the right tool (mandate: hand-written kernels for fused
+    # hot ops, rocBLAS for plain GEMMs).  Net1 fc1 [512,1600] measured
+    # 259 us on the wave-reduction bwd-weight kernel vs ~10 us library;
+    # every head/VAE/Net-small fc sits far below the bound.
+    _NATIVE_MAX_WEIGHT_NUMEL = 256 * 1024
+
     def forward(self, x):
-        if _native(x):
+        if _native(x) and self.weight.numel() <= self._NATIVE_MAX_WEIGHT_NUMEL:
             if x.dtype != torch.bfloat16 and torch.is_autocast_enabled():
                 x = x.to(torch.bfloat16)
             if x.dtype == torch.bfloat16 and x.dim() == 2:
