@@ -80,8 +80,12 @@ class _VaeElboFn(torch.autograd.Function):
 
 def vae_loss(recon_x, x, mu, logvar):
     """MSE(sum) reconstruction + analytic KL (federated_vae.py:97-108)."""
-    if _native(recon_x) and recon_x.dtype == x.dtype \
-            and mu.dtype == logvar.dtype == recon_x.dtype:
+    if _native(recon_x):
+        if not (recon_x.dtype == x.dtype == mu.dtype == logvar.dtype):
+            # mixed autocast dtypes: upcast so the fused kernel matches
+            # the torch path's fp32 promotion exactly
+            recon_x, x, mu, logvar = (t.float()
+                                      for t in (recon_x, x, mu, logvar))
         return _VaeElboFn.apply(recon_x.contiguous(), x.contiguous(),
                                 mu.contiguous(), logvar.contiguous())
     mse = F.mse_loss(recon_x, x, reduction="sum")
