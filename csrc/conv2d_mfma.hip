@@ -1332,12 +1332,16 @@ at::Tensor fedkit_conv2d_bwd_data(const at::Tensor& gy, const at::Tensor& w,
     return e && atoi(e) == 1;
   }();
   int K64 = gy.size(1);
-  if (!no_vpad && K64 % 64 == 0 && pl >= 0 && stride <= 2) {
-    // stride 1: affine gy address + per-slot int compares (first cut
-    // with per-slot 64-bit muls measured SLOWER than the materialized
-    // dilate; this form measured layer1 43.0 -> 29.6 us).
-    // stride 2: parity-class form — also skips the 3/4-zeros A traffic
-    // the materialized zero-inserted image pays.
+  // stride 1: affine gy address + per-slot int compares (first cut with
+  // per-slot 64-bit muls measured SLOWER than the materialized dilate;
+  // this form measured layer1 dx 42.4 -> 29.1 us).
+  // stride 2: parity-class form — skips the 3/4-zeros A traffic of the
+  // zero-inserted image; wins at large spatial (60.1 -> 49.8 us at H=32)
+  // but LOSES at H=8 (52.4 -> 61.4, latency-bound small grid), so gated
+  // on the dx spatial size (gpurun_out/ab3_*.log).
+  bool vpad_ok = !no_vpad && K64 % 64 == 0 && pl >= 0 &&
+      (stride == 1 || (stride == 2 && H >= 16));
+  if (vpad_ok) {
     VPadDesc vp{(int)gy.size(2), (int)gy.size(3), pl, (int)stride};
     return conv_core(gy, wrot, 1, (int)H, (int)W, (int)dil, (int)ctrue,
                      nullptr, &vp);
