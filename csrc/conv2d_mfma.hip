@@ -1049,7 +1049,13 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
                      xp.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto stream = fedkit_stream();
   // pick BM so the grid fills the 256 CUs when possible
-  bool bm64 = ((M + 127) / 128) * (Kout / BN) < 256;
+  // (FEDKIT_CONV_BM64=1 forces the 64-row tile everywhere: tuning knob —
+  // half the MFMA density per wg but 2x the workgroups and 3 wg/CU)
+  static const int force_bm64 = []() {
+    const char* e = getenv("FEDKIT_CONV_BM64");
+    return e ? atoi(e) : 0;
+  }();
+  bool bm64 = force_bm64 || ((M + 127) / 128) * (Kout / BN) < 256;
   int BM = bm64 ? 64 : 128;
   dim3 grid((unsigned)((M + BM - 1) / BM), Kout / BN);
   TORCH_CHECK(stride == 1 || stride == 2, "conv kernel supports stride 1/2");

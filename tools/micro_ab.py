@@ -77,3 +77,30 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def fwd_sweep():
+    """Forward conv timings per flagship shape (run under env combos:
+    FEDKIT_CONV_STAGES=2, FEDKIT_CONV_BM64=1)."""
+    tag = []
+    if os.environ.get("FEDKIT_CONV_BM64") == "1":
+        tag.append("bm64")
+    if os.environ.get("FEDKIT_CONV_STAGES") == "2":
+        tag.append("st2")
+    tag = "+".join(tag) or "default"
+    for C, H, K, stride in [(64, 32, 64, 1), (128, 16, 128, 1),
+                            (256, 8, 256, 1), (512, 4, 512, 1),
+                            (64, 32, 128, 2), (128, 16, 256, 2),
+                            (256, 8, 512, 2)]:
+        x = torch.randn(BATCH, C, H, H, device="cuda",
+                        dtype=torch.bfloat16).contiguous(
+            memory_format=torch.channels_last)
+        w = (torch.randn(K, C, 3, 3, device="cuda", dtype=torch.bfloat16)
+             * 0.05).contiguous(memory_format=torch.channels_last)
+        xp = ext.conv2d_pad_input(x, 1)
+        us = timeit(lambda: ext.conv2d_fwd_prepadded(xp, w, stride))
+        print(f"FWD[{tag}] C{C} H{H} K{K} s{stride}: {us:8.1f} us")
+
+
+if os.environ.get("FEDKIT_FWD_SWEEP") == "1":
+    fwd_sweep()
