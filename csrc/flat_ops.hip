@@ -386,3 +386,48 @@ void fedkit_cast_bf16_to_f32(std::vector<at::Tensor> srcs,
   hipLaunchKernelGGL(cast_bf16_to_f32_kernel, dim3(grid_1d(total, 256)),
                      dim3(256), 0, fedkit_stream(), d, total);
 }
+
+// ------------------------------------------------- fused Welford update
+// LBFGS batch mode, new-minibatch statistics (SURVEY §2a "Online grad
+// mean/variance (Welford) ... fused into the flat-grad kernel"; reference
+// lbfgsnew.py:600-615).  One kernel replaces 2 clones + 2 axpys +
+// 1 addcmul + 1 sum (~6 launches + 2 allocations):
+//   d_old = g - avg;  avg += d_old/n;  d_new = g - avg;
+//   avg_sq += d_new * d_old;  out = sum(avg_sq)
+namespace {
+__global__ void welford_update_kernel(const float* __restrict__ g,
+                                      float* __restrict__ avg,
+                                      float* __restrict__ avg_sq,
+                                      float inv_n,
+                                      float* __restrict__ out,
+                                      long long n) {
+  float acc = 0.f;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < n; i += (long long)gridDim.x * blockDim.x) {
+    float gi = g[i];
+    float d_old = gi - avg[i];
+    float a_new = avg[i] + d_old * inv_n;
+    avg[i] = a_new;
+    float sq = avg_sq[i] + (gi - a_new) * d_old;
+    avg_sq[i] = sq;
+    acc += sq;
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    acc += __shfl_xor(acc, off, 64);
+  if ((threadIdx.x & 63) == 0) atomicAdd(out, acc);
+}
+}  // namespace
+
+at::Tensor fedkit_welford_update(const at::Tensor& g, at::Tensor avg,
+                                 at::Tensor avg_sq, double inv_n) {
+  TORCH_CHECK(g.scalar_type() == at::kFloat && avg.scalar_type() == at::kFloat
+              && avg_sq.scalar_type() == at::kFloat,
+              "welford_update is fp32");
+  long long n = g.numel();
+  auto out = at::zeros({}, g.options());
+  hipLaunchKernelGGL(welford_update_kernel, dim3(grid_1d(n, 256)), dim3(256),
+                     0, fedkit_stream(), g.contiguous().data_ptr<float>(),
+                     avg.data_ptr<float>(), avg_sq.data_ptr<float>(),
+                     (float)inv_n, out.data_ptr<float>(), n);
+  return out;
+}

@@ -71,6 +71,8 @@ at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
                                     long stride, long padding, long R, long S,
                                     long dil);
 
+at::Tensor fedkit_welford_update(const at::Tensor& g, at::Tensor avg,
+                                 at::Tensor avg_sq, double inv_n);
 at::Tensor fedkit_linear_fwd(const at::Tensor& x, const at::Tensor& w,
                              const c10::optional<at::Tensor>& bias);
 at::Tensor fedkit_linear_bwd_data(const at::Tensor& gy, const at::Tensor& w);
@@ -173,6 +175,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_weight", &fedkit_conv2d_bwd_weight, "conv bwd-weight",
         py::arg("gy"), py::arg("x"), py::arg("stride"), py::arg("padding"),
         py::arg("R"), py::arg("S"), py::arg("dil") = 1);
+  m.def("welford_update", &fedkit_welford_update,
+        "fused LBFGS Welford grad-stats update; returns sum(avg_sq)");
   m.def("linear_fwd", &fedkit_linear_fwd,
         "fused linear fwd (x [M,K], w [N,K], bias) -> [M,N]",
         py::arg("x"), py::arg("w"), py::arg("bias") = c10::nullopt);

@@ -447,12 +447,22 @@ class LBFGSNew(Optimizer):
                 # first iteration of a new step() call == new minibatch
                 batch_changed = batch_mode and (n_iter == 1 and state["n_iter"] > 1)
                 if batch_changed:
-                    # online inter-batch Welford estimate drives alphabar
-                    g_del_old = flat_grad.clone().add_(running_avg, alpha=-1.0)
-                    running_avg.add_(g_del_old, alpha=1.0 / state["n_iter"])
-                    g_del_new = flat_grad.clone().add_(running_avg, alpha=-1.0)
-                    running_avg_sq.addcmul_(g_del_new, g_del_old, value=1)
-                    alphabar = 1.0 / (1.0 + float(running_avg_sq.sum())
+                    # online inter-batch Welford estimate drives alphabar;
+                    # fused path: ONE kernel updates avg/avg_sq in place
+                    # and returns sum(avg_sq) (csrc welford_update) vs the
+                    # ~6-launch clone/axpy/addcmul/sum chain
+                    if fused:
+                        from ..ops import require_ext
+                        sq_sum = float(require_ext().welford_update(
+                            flat_grad, running_avg, running_avg_sq,
+                            1.0 / state["n_iter"]))
+                    else:
+                        g_del_old = flat_grad.clone().add_(running_avg, alpha=-1.0)
+                        running_avg.add_(g_del_old, alpha=1.0 / state["n_iter"])
+                        g_del_new = flat_grad.clone().add_(running_avg, alpha=-1.0)
+                        running_avg_sq.addcmul_(g_del_new, g_del_old, value=1)
+                        sq_sum = float(running_avg_sq.sum())
+                    alphabar = 1.0 / (1.0 + sq_sum
                                       / ((state["n_iter"] - 1) * grad_nrm))
 
                 if ys > 1e-10 * sn2 and not batch_changed:

@@ -993,3 +993,22 @@ def test_fedlinear_module_autograd():
     assert frob_err(gx, x.grad) < 3e-2
     assert frob_err(gw, m.weight.grad) < 3e-2
     assert frob_err(gb, m.bias.grad) < 3e-2
+
+
+def test_welford_update_fused_vs_torch():
+    """Fused Welford grad-stats kernel vs the clone/axpy/addcmul chain."""
+    ext = _ext()
+    torch.manual_seed(31)
+    n = 100000
+    g = torch.randn(n, device="cuda")
+    avg = torch.randn(n, device="cuda")
+    sq = torch.rand(n, device="cuda")
+    avg_r, sq_r = avg.clone(), sq.clone()
+    it = 5
+    out = ext.welford_update(g, avg, sq, 1.0 / it)
+    d_old = g - avg_r
+    avg_r += d_old / it
+    sq_r += (g - avg_r) * d_old
+    assert frob_err(avg, avg_r) < 1e-6
+    assert frob_err(sq, sq_r) < 1e-6
+    assert abs(float(out) - float(sq_r.sum())) / abs(float(sq_r.sum())) < 1e-4
