@@ -40,10 +40,12 @@ def main():
     ap.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     ap.add_argument("--native", type=int, default=1,
                     help="0 = eager torch ops for A/B comparison")
-    ap.add_argument("--graph", type=int, default=0,
-                    help="1 = hipGraph-capture the local step (measured: no "
-                         "gain at this size — the GPU, not the host enqueue "
-                         "path, is saturated; kept for larger configs)")
+    ap.add_argument("--graph", type=int, default=1,
+                    help="hipGraph-capture the local step (default ON since "
+                         "round 2: after the kernel fusions the step is "
+                         "launch-bound enough that one replay beats ~380 "
+                         "enqueues — measured 3.83 -> 3.61 ms same box; "
+                         "falls back to eager if capture fails)")
     args = ap.parse_args()
 
     if not args.native:

@@ -1055,7 +1055,10 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
     const char* e = getenv("FEDKIT_CONV_BM64");
     return e ? atoi(e) : 0;
   }();
-  bool bm64 = force_bm64 || ((M + 127) / 128) * (Kout / BN) < 256;
+  // threshold 512 (2 wg/CU at 256 CUs): measured round 2 — BM64 wins
+  // 20.5 -> 15.2 us on the C128-s2 class and ties where the BM128 grid
+  // already fills (gpurun_out/fwd_sweep.log)
+  bool bm64 = force_bm64 || ((M + 127) / 128) * (Kout / BN) < 512;
   int BM = bm64 ? 64 : 128;
   dim3 grid((unsigned)((M + BM - 1) / BM), Kout / BN);
   TORCH_CHECK(stride == 1 || stride == 2, "conv kernel supports stride 1/2");
@@ -1084,10 +1087,16 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
                        N, Hp, Wp, C, Kout, R, S, P, Q, Kg, dil, Ktrue,
                        aux, TapDesc{}, vpad ? *vpad : VPadDesc{});
   };
-  static const int stages2 = []() {
+  // STAGES=2 (single lookahead, 48 KB LDS -> 3 wg/CU) measured faster or
+  // tied on EVERY MODE-0 shape in the round-2 sweep (layer1 25.2 -> 22.6
+  // us; fwd_sweep.log) — the extra resident waves hide more latency than
+  // the deeper pipeline, consistent with the 43% wave-wait PMC.  Default
+  // 2; FEDKIT_CONV_STAGES=3 restores the old pipeline.
+  static const int stages3 = []() {
     const char* e = getenv("FEDKIT_CONV_STAGES");
-    return e && atoi(e) == 2;
+    return e && atoi(e) == 3;
   }();
+  const bool stages2 = !stages3;
   if (vpad) {
     TORCH_CHECK(C % 64 == 0 && stride == 1 && !bnpart,
                 "vpad path needs C % 64 == 0, stride-1 conv");
