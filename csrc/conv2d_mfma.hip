@@ -1123,7 +1123,12 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
     return y;
   }
   if (splits > 1) {
-    if (stride == 1)
+    if (stages2) {
+      if (stride == 1)
+        bm64 ? L(conv_fwd_kernel<64, 1, 2, 2>) : L(conv_fwd_kernel<128, 1, 2, 2>);
+      else
+        bm64 ? L(conv_fwd_kernel<64, 2, 2, 2>) : L(conv_fwd_kernel<128, 2, 2, 2>);
+    } else if (stride == 1)
       bm64 ? L(conv_fwd_kernel<64, 1, 2, 3>) : L(conv_fwd_kernel<128, 1, 2, 3>);
     else
       bm64 ? L(conv_fwd_kernel<64, 2, 2, 3>) : L(conv_fwd_kernel<128, 2, 2, 3>);
