@@ -40,12 +40,12 @@ def main():
     ap.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     ap.add_argument("--native", type=int, default=1,
                     help="0 = eager torch ops for A/B comparison")
-    ap.add_argument("--graph", type=int, default=1,
-                    help="hipGraph-capture the local step (default ON since "
-                         "round 2: after the kernel fusions the step is "
-                         "launch-bound enough that one replay beats ~380 "
-                         "enqueues — measured 3.83 -> 3.61 ms same box; "
-                         "falls back to eager if capture fails)")
+    ap.add_argument("--graph", type=int, default=0,
+                    help="1 = hipGraph-capture the local step.  Measured "
+                         "BOTH ways across boxes (r2: -6% on one box, +1.5% "
+                         "on another with the run order reversed) — the "
+                         "apparent gain tracks thermal run order, not the "
+                         "graph; default off, kept for A/B")
     args = ap.parse_args()
 
     if not args.native:
