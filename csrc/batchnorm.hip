@@ -215,6 +215,26 @@ __global__ void bn_conv_colsum_finalize_kernel(
 // epilogue (the reference's `elu(bn(conv) [+ shortcut])` patterns,
 // simple_models.py:150-153) — one pass instead of bn/add/elu separate
 // kernels (each an extra HBM read+write of the activation).
+// pad geometry for the apply-into-pad fusion (pad == 0 -> identity).
+// BN's normalized output almost always feeds a padded conv; writing the
+// PADDED image directly from the apply kernel removes the separate pad
+// launch and a full read+write pass of y (round 2).
+struct PadGeom {
+  int H, W, Hp, Wp, pad;
+};
+
+// vector index over the unpadded domain -> index into the padded image
+__device__ __forceinline__ long long pad_vec_idx(const PadGeom& pg,
+                                                 long long i, int Cv) {
+  long long m = i / Cv;
+  int cv = (int)(i - m * Cv);
+  int w = (int)(m % pg.W);
+  long long t = m / pg.W;
+  int h = (int)(t % pg.H);
+  long long n = t / pg.H;
+  return (((n * pg.Hp) + h + pg.pad) * pg.Wp + (w + pg.pad)) * Cv + cv;
+}
+
 template <typename T, int VEC, bool ELU, bool RES>
 __global__ void bn_apply_kernel(const T* __restrict__ x,
                                 const T* __restrict__ res, T* __restrict__ y,
@@ -222,17 +242,40 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
                                 const float* __restrict__ invstd,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
-                                long long nvec, int Cv /* C / VEC */) {
+                                long long nvec /* PADDED count if pad>0 */,
+                                int Cv /* C / VEC */, PadGeom pg) {
   using V = VecT<T, VEC>;
   const V* xv = reinterpret_cast<const V*>(x);
   const V* rv = reinterpret_cast<const V*>(res);
   V* yv = reinterpret_cast<V*>(y);
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
        i < nvec; i += (long long)gridDim.x * blockDim.x) {
-    int c0 = (int)(i % Cv) * VEC;
-    V a = xv[i], r;
+    long long ix = i;
+    bool border = false;
+    if (pg.pad) {
+      // i walks the PADDED image; borders get zeros, interior reads x
+      long long mp = i / Cv;
+      int cv = (int)(i - mp * Cv);
+      int wp = (int)(mp % pg.Wp);
+      long long t = mp / pg.Wp;
+      int hp = (int)(t % pg.Hp);
+      long long n = t / pg.Hp;
+      int h = hp - pg.pad, w = wp - pg.pad;
+      border = h < 0 || w < 0 || h >= pg.H || w >= pg.W;
+      if (!border)
+        ix = (((n * pg.H) + h) * pg.W + w) * Cv + cv;
+    }
+    V r;
+    if (border) {
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) from_f32(0.f, r.v[j]);
+      yv[i] = r;
+      continue;
+    }
+    int c0 = (int)(ix % Cv) * VEC;
+    V a = xv[ix];
     V rr;
-    if (RES) rr = rv[i];
+    if (RES) rr = rv[ix];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       int c = c0 + j;
@@ -256,7 +299,8 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
                                        const T* __restrict__ yout, long long M,
                                        int Cv, const float* __restrict__ mean,
                                        const float* __restrict__ invstd,
-                                       float* __restrict__ part /* [nb][2][C] */) {
+                                       float* __restrict__ part /* [nb][2][C] */,
+                                       PadGeom pg) {
   using V = VecT<T, VEC>;
   const int C = Cv * VEC;
   const V* xv = reinterpret_cast<const V*>(x);
@@ -274,14 +318,19 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
     m[j] = mean[c0 + j];
     is[j] = invstd[c0 + j];
   }
+  // gy / yout may live in a PADDED image (apply-into-pad forward)
+  auto gidx = [&](long long i) {
+    return pg.pad ? pad_vec_idx(pg, i, Cv) : i;
+  };
   float sdy[VEC] = {}, sdyx[VEC] = {};
   long long i = i0;
   for (; i + stride < total; i += 2 * stride) {
-    V a0 = xv[i], g0 = gv[i], a1 = xv[i + stride], g1 = gv[i + stride];
+    long long ig0 = gidx(i), ig1 = gidx(i + stride);
+    V a0 = xv[i], g0 = gv[ig0], a1 = xv[i + stride], g1 = gv[ig1];
     V y0, y1;
     if (ELU) {
-      y0 = yv[i];
-      y1 = yv[i + stride];
+      y0 = yv[ig0];
+      y1 = yv[ig1];
     }
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
@@ -296,9 +345,10 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
     }
   }
   for (; i < total; i += stride) {
-    V a = xv[i], g = gv[i];
+    long long ig = gidx(i);
+    V a = xv[i], g = gv[ig];
     V yy;
-    if (ELU) yy = yv[i];
+    if (ELU) yy = yv[ig];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       float gg = to_f32(g.v[j]);
@@ -340,7 +390,8 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
                                     const float* __restrict__ invstd,
                                     const float* __restrict__ gamma,
                                     const float* __restrict__ ws,
-                                    long long M, long long nvec, int Cv) {
+                                    long long M, long long nvec, int Cv,
+                                    PadGeom pg) {
   using V = VecT<T, VEC>;
   const V* xv = reinterpret_cast<const V*>(x);
   const V* gv = reinterpret_cast<const V*>(gy);
@@ -351,10 +402,11 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
   float inv_count = 1.f / (float)M;
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
        i < nvec; i += (long long)gridDim.x * blockDim.x) {
+    long long ig = pg.pad ? pad_vec_idx(pg, i, Cv) : i;
     int c0 = (int)(i % Cv) * VEC;
-    V a = xv[i], g = gv[i], r, go;
+    V a = xv[i], g = gv[ig], r, go;
     V yy;
-    if (ELU) yy = yv[i];
+    if (ELU) yy = yv[ig];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       int c = c0 + j;
@@ -388,14 +440,21 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
                                       double momentum, double eps,
                                       c10::optional<at::Tensor> residual,
                                       bool elu,
-                                      c10::optional<at::Tensor> conv_part) {
+                                      c10::optional<at::Tensor> conv_part,
+                                      long pad_out) {
   check_nhwc(x);
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   long long M = (long long)N * H * W;
   auto fopts = x.options().dtype(at::kFloat);
   auto save_mean = at::empty({C}, fopts);
   auto save_invstd = at::empty({C}, fopts);
-  auto y = at::empty_like(x);
+  // pad_out > 0: the apply writes the PADDED image the next conv
+  // consumes (border zeros included) — no separate pad launch/pass
+  const int pad = (int)pad_out;
+  at::Tensor y = pad == 0
+      ? at::empty_like(x)
+      : at::empty({N, C, H + 2 * pad, W + 2 * pad},
+                  x.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto stream = fedkit_stream();
   auto gamma_f = gamma.contiguous();
   auto beta_f = beta.contiguous();
@@ -459,7 +518,9 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
   DISPATCH_F32_BF16(x, "bn_apply", {
     constexpr int VEC = 16 / sizeof(scalar_t);
     TORCH_CHECK(C % VEC == 0, "bn needs C % ", VEC, " == 0");
-    long long nvec = M * C / VEC;
+    PadGeom pg = {H, W, H + 2 * pad, W + 2 * pad, pad};
+    long long nvec = (pad == 0 ? M
+                      : (long long)N * pg.Hp * pg.Wp) * C / VEC;
     auto launch = [&](auto ekind, auto rkind) {
       hipLaunchKernelGGL(
           (bn_apply_kernel<scalar_t, VEC, decltype(ekind)::value,
@@ -468,7 +529,7 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
           (const scalar_t*)x.data_ptr(), (const scalar_t*)res_ptr,
           (scalar_t*)y.data_ptr(), save_mean.data_ptr<float>(),
           save_invstd.data_ptr<float>(), gamma_f.data_ptr<float>(),
-          beta_f.data_ptr<float>(), nvec, C / VEC);
+          beta_f.data_ptr<float>(), nvec, C / VEC, pg);
     };
     using T0 = std::integral_constant<bool, false>;
     using T1 = std::integral_constant<bool, true>;
@@ -488,11 +549,15 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& save_mean,
                                       const at::Tensor& save_invstd,
                                       c10::optional<at::Tensor> elu_y,
-                                      bool want_g) {
+                                      bool want_g, long pad_in) {
   check_nhwc(x);
   check_nhwc(gy);
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   long long M = (long long)N * H * W;
+  const int pad = (int)pad_in;
+  PadGeom pg = {H, W, H + 2 * pad, W + 2 * pad, pad};
+  TORCH_CHECK(gy.size(2) == pg.Hp && gy.size(3) == pg.Wp,
+              "bn_bwd: gy geometry mismatch for pad_in=", pad);
   auto fopts = x.options().dtype(at::kFloat);
   auto ws = at::empty({2, C}, fopts);
   auto gx = at::empty_like(x);
@@ -502,7 +567,7 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
   const void* y_ptr = elu ? elu_y->data_ptr() : nullptr;
   TORCH_CHECK(!want_g || elu, "want_g requires the fused-ELU backward");
   at::Tensor gout;
-  if (want_g) gout = at::empty_like(gy);
+  if (want_g) gout = at::empty_like(x);   // residual grad is UNPADDED
   using T0 = std::integral_constant<bool, false>;
   using T1 = std::integral_constant<bool, true>;
   DISPATCH_F32_BF16(x, "bn_bwd_partials", {
@@ -520,7 +585,8 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                          (const scalar_t*)gy.data_ptr(),
                          (const scalar_t*)y_ptr, M, C / VEC,
                          save_mean.data_ptr<float>(),
-                         save_invstd.data_ptr<float>(), part.data_ptr<float>());
+                         save_invstd.data_ptr<float>(), part.data_ptr<float>(),
+                         pg);
     };
     if (elu) launch1(T1{}); else launch1(T0{});
     hipLaunchKernelGGL(bn_colsum_kernel, dim3((2 * C + 3) / 4), dim3(256),
@@ -542,7 +608,7 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                          save_mean.data_ptr<float>(),
                          save_invstd.data_ptr<float>(),
                          gamma_f.data_ptr<float>(), ws.data_ptr<float>(), M,
-                         nvec, C / VEC);
+                         nvec, C / VEC, pg);
     };
     if (elu && want_g)  launch2(T1{}, T1{});
     else if (elu)       launch2(T1{}, T0{});

@@ -48,7 +48,10 @@ class BasicBlock(nn.Module):
             )
 
     def forward(self, x):
-        out = bn_elu(self.bn1, self.conv1(x))
+        # bn1's output feeds ONLY conv2 (pad 1): the apply kernel writes
+        # the padded image directly (bn_elu pad_out fusion) and conv2
+        # skips its pad pass
+        out = bn_elu(self.bn1, self.conv1(x), pad_out=1)
         return bn_elu(self.bn2, self.conv2(out), residual=self.shortcut(x))
 
 

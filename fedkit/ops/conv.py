@@ -39,10 +39,14 @@ def _ext():
 
 class _ConvFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, stride, padding):
+    def forward(ctx, x, w, stride, padding, prepadded=False):
         # x: [N,C,H,W] logical, carried NHWC-contiguous; w: [K,C,R,S].
         # The padded input is computed ONCE and saved, so bwd-weight reuses
         # it instead of re-padding (and x itself need not be kept).
+        # prepadded=True (round 2): x arrived ALREADY padded from the
+        # producing bn_elu(pad_out=...) apply kernel — skip the pad pass;
+        # backward then returns the padded-domain input gradient (the
+        # bn backward reads it at padded coordinates).
         # FEDKIT_CONV_BNSTATS=1: the epilogue also emits the BatchNorm
         # stage-1 partials and the downstream FedBatchNorm2d skips its own
         # reduction pass.  MEASURED NET LOSS at CIFAR sizes (step 3.9 ->
@@ -57,6 +61,10 @@ class _ConvFn(torch.autograd.Function):
             y = ext.conv2d_fwd(x, w, stride, padding)
             part = y.new_empty(0)
             ctx.save_for_backward(x, w)
+        elif prepadded:
+            y = ext.conv2d_fwd_prepadded(x, w, stride)
+            part = y.new_empty(0)
+            ctx.save_for_backward(x, w)
         else:
             xp = ext.conv2d_pad_input(x, padding)
             if want_stats:
@@ -66,6 +74,7 @@ class _ConvFn(torch.autograd.Function):
                 part = y.new_empty(0)
             ctx.save_for_backward(xp, w)
         ctx.small_c = small_c
+        ctx.prepadded = prepadded
         ctx.stride = stride
         ctx.padding = padding
         ctx.hw = (x.shape[2], x.shape[3])
@@ -78,8 +87,14 @@ class _ConvFn(torch.autograd.Function):
         gy = gy.contiguous(memory_format=torch.channels_last)
         gx = gw = None
         if ctx.needs_input_grad[0]:
-            gx = _ext().conv2d_bwd_data(gy, w, ctx.stride, ctx.padding,
-                                        ctx.hw[0], ctx.hw[1])
+            if ctx.prepadded:
+                # grad wrt the PADDED input: same conv adjoint with pad=0
+                # over the padded geometry
+                gx = _ext().conv2d_bwd_data(gy, w, ctx.stride, 0,
+                                            ctx.hw[0], ctx.hw[1])
+            else:
+                gx = _ext().conv2d_bwd_data(gy, w, ctx.stride, ctx.padding,
+                                            ctx.hw[0], ctx.hw[1])
         if ctx.needs_input_grad[1]:
             if ctx.small_c:
                 gw = _ext().conv2d_bwd_weight(gy, xsaved, ctx.stride,
@@ -87,7 +102,7 @@ class _ConvFn(torch.autograd.Function):
             else:
                 gw = _ext().conv2d_bwd_weight_prepadded(
                     gy, xsaved, ctx.stride, w.shape[2], w.shape[3])
-        return gx, gw, None, None
+        return gx, gw, None, None, None
 
 
 def _pad_mult(n, m):
@@ -393,6 +408,8 @@ def dilated_bank(x, mods):
 
 class FedConv2d(nn.Conv2d):
     def forward(self, x):
+        # producer-padded input (bn_elu(pad_out=...) fusion)?
+        prepad = getattr(x, "_fedkit_prepad", 0)
         if _native(x) and self.bias is None and self.groups == 1 \
                 and self.dilation == (1, 1) and self.kernel_size[0] in (1, 3):
             if x.dtype != torch.bfloat16 and torch.is_autocast_enabled():
@@ -410,10 +427,17 @@ class FedConv2d(nn.Conv2d):
                         self, "_w16", self.weight,
                         lambda: self.weight.to(torch.bfloat16).contiguous(
                             memory_format=torch.channels_last))
-                y, part = _ConvFn.apply(x, w, self.stride[0], self.padding[0])
+                y, part = _ConvFn.apply(x, w, self.stride[0], self.padding[0],
+                                        prepad == self.padding[0] > 0)
                 if self.training and part.numel():
                     y._fedkit_bn_stats = part
                 return y
+        if prepad and prepad == self.padding[0]:
+            # fallback consuming an already-padded image: pad 0
+            return F.conv2d(x, self.weight.to(x.dtype),
+                            self.bias.to(x.dtype) if self.bias is not None
+                            else None, self.stride, 0, self.dilation,
+                            self.groups)
         return F.conv2d(x, self.weight.to(x.dtype),
                         self.bias.to(x.dtype) if self.bias is not None else None,
                         self.stride, self.padding, self.dilation, self.groups)

@@ -56,17 +56,26 @@ class _BnActFn(torch.autograd.Function):
     (simple_models.py:150-153) with a single apply kernel.  Backward:
     g = elu'(y) * gy (from the saved post-activation output), which is both
     the residual grad and the BN-output grad.
+
+    pad_out > 0 (round 2): the apply writes directly into the PADDED image
+    the next conv consumes (borders zeroed in-kernel) — removes the
+    separate pad launch plus a full read+write pass of y.  The output is
+    padded; backward receives the padded-domain gradient and the bn_bwd
+    kernels read gy / elu-y at padded coordinates.
     """
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
-                training, momentum, eps, residual, conv_part=None):
+                training, momentum, eps, residual, conv_part=None,
+                pad_out=0):
         y, save_mean, save_invstd = _ext().bn_fwd(
             x, weight, bias, running_mean, running_var,
             bool(training), float(momentum), float(eps),
-            residual=residual, elu=True, conv_part=conv_part)
+            residual=residual, elu=True, conv_part=conv_part,
+            pad_out=int(pad_out))
         ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
         ctx.has_res = residual is not None
+        ctx.pad_out = int(pad_out)
         return y
 
     @staticmethod
@@ -77,10 +86,11 @@ class _BnActFn(torch.autograd.Function):
         # output y); the residual grad g is streamed out of the apply kernel
         # only when the residual branch needs it
         out = _ext().bn_bwd(gy, x, weight, save_mean, save_invstd,
-                            elu_y=y, want_g=ctx.has_res)
+                            elu_y=y, want_g=ctx.has_res,
+                            pad_in=ctx.pad_out)
         gres = out[3] if ctx.has_res else None
         return (out[0], out[1], out[2], None, None, None, None, None, gres,
-                None)
+                None, None)
 
 
 class FedBatchNorm2d(nn.BatchNorm2d):
@@ -138,8 +148,12 @@ class FedBatchNorm2d(nn.BatchNorm2d):
 
 
 def bn_elu(bn: FedBatchNorm2d, x: torch.Tensor,
-           residual: torch.Tensor = None) -> torch.Tensor:
-    """elu(bn(x) [+ residual]) — fused on GPU, composed on CPU."""
+           residual: torch.Tensor = None, pad_out: int = 0) -> torch.Tensor:
+    """elu(bn(x) [+ residual]) — fused on GPU, composed on CPU.
+
+    pad_out > 0: the GPU path emits the PADDED image (marked with
+    `_fedkit_prepad`) so the consuming conv skips its pad pass; the CPU
+    path ignores it (the conv pads itself)."""
     if _native(x):
         bn._prep(x)
         part = getattr(x, "_fedkit_bn_stats", None) if bn.training else None
@@ -148,10 +162,13 @@ def bn_elu(bn: FedBatchNorm2d, x: torch.Tensor,
             residual = residual.contiguous(memory_format=torch.channels_last)
             if residual.dtype != x.dtype:
                 residual = residual.to(x.dtype)
-        return _BnActFn.apply(x, bn.weight, bn.bias,
-                              bn.running_mean, bn.running_var,
-                              bn.training, bn.momentum, bn.eps, residual,
-                              part)
+        y = _BnActFn.apply(x, bn.weight, bn.bias,
+                           bn.running_mean, bn.running_var,
+                           bn.training, bn.momentum, bn.eps, residual,
+                           part, pad_out)
+        if pad_out:
+            y._fedkit_prepad = pad_out
+        return y
     y = nn.BatchNorm2d.forward(bn, x)
     if residual is not None:
         y = y + residual

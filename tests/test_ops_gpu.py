@@ -1012,3 +1012,65 @@ def test_welford_update_fused_vs_torch():
     assert frob_err(avg, avg_r) < 1e-6
     assert frob_err(sq, sq_r) < 1e-6
     assert abs(float(out) - float(sq_r.sum())) / abs(float(sq_r.sum())) < 1e-4
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_bn_elu_pad_out_fusion(dtype):
+    """bn_elu(pad_out=1) emits the padded image (borders zero) and its
+    backward matches the composed pad(elu(bn(x))) autograd exactly."""
+    from fedkit.ops.norm import FedBatchNorm2d, bn_elu
+    torch.manual_seed(37)
+    C = 64
+    bn = FedBatchNorm2d(C).cuda().train()
+    x = torch.randn(8, C, 16, 16, device="cuda", dtype=dtype,
+                    requires_grad=True)
+    xp = bn_elu(bn, x.contiguous(memory_format=torch.channels_last),
+                pad_out=1)
+    assert xp.shape == (8, C, 18, 18)
+    assert getattr(xp, "_fedkit_prepad", 0) == 1
+    gy = torch.randn_like(xp)
+    xp.backward(gy)
+    gx = x.grad.clone()
+    gw, gb = bn.weight.grad.clone(), bn.bias.grad.clone()
+    x.grad = bn.weight.grad = bn.bias.grad = None
+
+    bn2 = FedBatchNorm2d(C).cuda().train()
+    with torch.no_grad():
+        bn2.weight.copy_(bn.weight)
+        bn2.bias.copy_(bn.bias)
+    yref = F.pad(F.elu(F.batch_norm(
+        x.float(), bn2.running_mean, bn2.running_var, bn2.weight, bn2.bias,
+        True, 0.1, 1e-5)), (1, 1, 1, 1))
+    yref.backward(gy.float())
+    tol = 1e-3 if dtype == torch.float32 else 5e-2
+    assert rel_err(xp, yref) < tol
+    assert rel_err(gx, x.grad) < tol
+    assert rel_err(gw, bn2.weight.grad) < tol
+    assert rel_err(gb, bn2.bias.grad) < tol
+
+
+def test_resnet_block_pad_fusion_end_to_end():
+    """A BasicBlock through the apply-into-pad path vs eager fp32."""
+    import fedkit.ops as ops
+    from fedkit.models.resnet import BasicBlock
+    torch.manual_seed(41)
+    blk = BasicBlock(64, 64).cuda().to(memory_format=torch.channels_last)
+    x = torch.randn(16, 64, 16, 16, device="cuda").contiguous(
+        memory_format=torch.channels_last)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y = blk(x)
+    loss = y.float().square().mean()
+    loss.backward()
+    nat = {n: p.grad.float().clone() for n, p in blk.named_parameters()}
+    blk.zero_grad(set_to_none=True)
+    old = ops._NATIVE_ENV
+    ops._NATIVE_ENV = False
+    try:
+        y2 = blk(x)
+        loss2 = y2.square().mean()
+        loss2.backward()
+    finally:
+        ops._NATIVE_ENV = old
+    assert frob_err(y, y2) < 5e-2
+    for n, p in blk.named_parameters():
+        assert frob_err(nat[n], p.grad.float()) < 8e-2, n
