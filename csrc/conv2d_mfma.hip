@@ -1055,10 +1055,13 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
     const char* e = getenv("FEDKIT_CONV_BM64");
     return e ? atoi(e) : 0;
   }();
-  // threshold 512 (2 wg/CU at 256 CUs): measured round 2 — BM64 wins
+  // threshold 512 (2 wg/CU at 256 CUs) for the ordinary path: BM64 wins
   // 20.5 -> 15.2 us on the C128-s2 class and ties where the BM128 grid
-  // already fills (gpurun_out/fwd_sweep.log)
-  bool bm64 = force_bm64 || ((M + 127) / 128) * (Kout / BN) < 512;
+  // already fills (fwd_sweep.log).  vpad (bwd-data) keeps 256: the wider
+  // threshold REGRESSED layer3-class dx 32.4 -> 49.8 us (stats_v8.txt) —
+  // the per-slot compare overhead wants the denser BM128 tile.
+  bool bm64 = force_bm64 ||
+      ((M + 127) / 128) * (Kout / BN) < (vpad ? 256 : 512);
   int BM = bm64 ? 64 : 128;
   dim3 grid((unsigned)((M + BM - 1) / BM), Kout / BN);
   TORCH_CHECK(stride == 1 || stride == 2, "conv kernel supports stride 1/2");
@@ -1112,6 +1115,13 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
       hipLaunchKernelGGL(colsum_to_bf16_kernel, dim3(grid_1d(Ly, 256)),
                          dim3(256), 0, stream, part.data_ptr<float>(),
                          splits, Ly, (bf16*)y.data_ptr());
+    } else if (stages2) {
+      if (v2)
+        bm64 ? L(conv_fwd_kernel<64, 1, 0, 2, false, 2>)
+             : L(conv_fwd_kernel<128, 1, 0, 2, false, 2>);
+      else
+        bm64 ? L(conv_fwd_kernel<64, 1, 0, 2, false, 1>)
+             : L(conv_fwd_kernel<128, 1, 0, 2, false, 1>);
     } else {
       if (v2)
         bm64 ? L(conv_fwd_kernel<64, 1, 0, 3, false, 2>)
