@@ -243,7 +243,8 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
                                 long long nvec /* PADDED count if pad>0 */,
-                                int Cv /* C / VEC */, PadGeom pg) {
+                                int Cv /* C / VEC */, PadGeom pg,
+                                PadGeom rpg /* residual geometry */) {
   using V = VecT<T, VEC>;
   const V* xv = reinterpret_cast<const V*>(x);
   const V* rv = reinterpret_cast<const V*>(res);
@@ -275,7 +276,7 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
     int c0 = (int)(ix % Cv) * VEC;
     V a = xv[ix];
     V rr;
-    if (RES) rr = rv[ix];
+    if (RES) rr = rv[rpg.pad ? pad_vec_idx(rpg, ix, Cv) : ix];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       int c = c0 + j;
@@ -391,7 +392,8 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
                                     const float* __restrict__ gamma,
                                     const float* __restrict__ ws,
                                     long long M, long long nvec, int Cv,
-                                    PadGeom pg) {
+                                    PadGeom pg,
+                                    PadGeom rpg /* gout geometry */) {
   using V = VecT<T, VEC>;
   const V* xv = reinterpret_cast<const V*>(x);
   const V* gv = reinterpret_cast<const V*>(gy);
@@ -420,7 +422,7 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
       from_f32(val, r.v[j]);
     }
     ov[i] = r;
-    if (WANTG) gov[i] = go;
+    if (WANTG) gov[rpg.pad ? pad_vec_idx(rpg, i, Cv) : i] = go;
   }
 }
 
@@ -441,7 +443,7 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
                                       c10::optional<at::Tensor> residual,
                                       bool elu,
                                       c10::optional<at::Tensor> conv_part,
-                                      long pad_out) {
+                                      long pad_out, long res_pad) {
   check_nhwc(x);
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   long long M = (long long)N * H * W;
@@ -508,17 +510,21 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
   }
   const bool has_res = residual.has_value();
   const void* res_ptr = nullptr;
+  const int rpad = (int)res_pad;
   if (has_res) {
     check_nhwc(*residual);
-    TORCH_CHECK(residual->scalar_type() == x.scalar_type() &&
-                residual->sizes() == x.sizes(),
-                "bn residual must match input shape/dtype");
+    TORCH_CHECK(residual->scalar_type() == x.scalar_type(),
+                "bn residual must match input dtype");
+    TORCH_CHECK(residual->size(2) == H + 2 * rpad &&
+                residual->size(3) == W + 2 * rpad,
+                "bn residual geometry mismatch for res_pad=", rpad);
     res_ptr = residual->data_ptr();
   }
   DISPATCH_F32_BF16(x, "bn_apply", {
     constexpr int VEC = 16 / sizeof(scalar_t);
     TORCH_CHECK(C % VEC == 0, "bn needs C % ", VEC, " == 0");
     PadGeom pg = {H, W, H + 2 * pad, W + 2 * pad, pad};
+    PadGeom rpg = {H, W, H + 2 * rpad, W + 2 * rpad, rpad};
     long long nvec = (pad == 0 ? M
                       : (long long)N * pg.Hp * pg.Wp) * C / VEC;
     auto launch = [&](auto ekind, auto rkind) {
@@ -529,7 +535,7 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
           (const scalar_t*)x.data_ptr(), (const scalar_t*)res_ptr,
           (scalar_t*)y.data_ptr(), save_mean.data_ptr<float>(),
           save_invstd.data_ptr<float>(), gamma_f.data_ptr<float>(),
-          beta_f.data_ptr<float>(), nvec, C / VEC, pg);
+          beta_f.data_ptr<float>(), nvec, C / VEC, pg, rpg);
     };
     using T0 = std::integral_constant<bool, false>;
     using T1 = std::integral_constant<bool, true>;
@@ -549,7 +555,8 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& save_mean,
                                       const at::Tensor& save_invstd,
                                       c10::optional<at::Tensor> elu_y,
-                                      bool want_g, long pad_in) {
+                                      bool want_g, long pad_in,
+                                      long res_pad) {
   check_nhwc(x);
   check_nhwc(gy);
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
@@ -566,8 +573,18 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
   const bool elu = elu_y.has_value();
   const void* y_ptr = elu ? elu_y->data_ptr() : nullptr;
   TORCH_CHECK(!want_g || elu, "want_g requires the fused-ELU backward");
+  const int rpad = (int)res_pad;
+  PadGeom rpg = {H, W, H + 2 * rpad, W + 2 * rpad, rpad};
   at::Tensor gout;
-  if (want_g) gout = at::empty_like(x);   // residual grad is UNPADDED
+  if (want_g) {
+    // residual grad matches the residual's geometry; PADDED residuals get
+    // zero borders (at::zeros) and the kernel writes the interior
+    gout = rpad == 0
+        ? at::empty_like(x)
+        : at::zeros({(long long)N, (long long)C, (long long)rpg.Hp,
+                     (long long)rpg.Wp},
+                    x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  }
   using T0 = std::integral_constant<bool, false>;
   using T1 = std::integral_constant<bool, true>;
   DISPATCH_F32_BF16(x, "bn_bwd_partials", {
@@ -608,7 +625,7 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                          save_mean.data_ptr<float>(),
                          save_invstd.data_ptr<float>(),
                          gamma_f.data_ptr<float>(), ws.data_ptr<float>(), M,
-                         nvec, C / VEC, pg);
+                         nvec, C / VEC, pg, rpg);
     };
     if (elu && want_g)  launch2(T1{}, T1{});
     else if (elu)       launch2(T1{}, T0{});

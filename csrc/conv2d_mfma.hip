@@ -1035,7 +1035,8 @@ void check_conv_inputs(const at::Tensor& x, const at::Tensor& w) {
 at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
                      int stride, int P, int Q, int dil = 1, int Ktrue = -1,
                      float* bnpart = nullptr,
-                     const VPadDesc* vpad = nullptr) {
+                     const VPadDesc* vpad = nullptr,
+                     int bm64_below = 256 /* FWD callers pass 512 */) {
   int N = xp.size(0), C = xp.size(1), Hp = xp.size(2), Wp = xp.size(3);
   int Kout = w_krs_c.size(0), R = w_krs_c.size(2), S = w_krs_c.size(3);
   int Kg = R * S * C;
@@ -1055,13 +1056,12 @@ at::Tensor conv_core(const at::Tensor& xp, const at::Tensor& w_krs_c,
     const char* e = getenv("FEDKIT_CONV_BM64");
     return e ? atoi(e) : 0;
   }();
-  // threshold 512 (2 wg/CU at 256 CUs) for the ordinary path: BM64 wins
-  // 20.5 -> 15.2 us on the C128-s2 class and ties where the BM128 grid
-  // already fills (fwd_sweep.log).  vpad (bwd-data) keeps 256: the wider
-  // threshold REGRESSED layer3-class dx 32.4 -> 49.8 us (stats_v8.txt) —
-  // the per-slot compare overhead wants the denser BM128 tile.
+  // per-call-site threshold: FWD shapes measured BM64 wins below 512 wgs
+  // (C128-s2 fwd 20.5 -> 15.2 us, fwd_sweep.log) but BOTH bwd-data forms
+  // regressed under it (vpad layer3 dx 32.4 -> 49.8, materialized s2-C256
+  // 52.4 -> 64.0) — backward callers keep the 256 default.
   bool bm64 = force_bm64 ||
-      ((M + 127) / 128) * (Kout / BN) < (vpad ? 256 : 512);
+      ((M + 127) / 128) * (Kout / BN) < (vpad ? 256 : bm64_below);
   int BM = bm64 ? 64 : 128;
   dim3 grid((unsigned)((M + BM - 1) / BM), Kout / BN);
   TORCH_CHECK(stride == 1 || stride == 2, "conv kernel supports stride 1/2");
@@ -1263,7 +1263,8 @@ at::Tensor fedkit_conv2d_fwd_prepadded(const at::Tensor& xp,
   int Reff = (R - 1) * (int)dil + 1, Seff = (S - 1) * (int)dil + 1;
   int P = (Hp - Reff) / (int)stride + 1;
   int Q = (Wp - Seff) / (int)stride + 1;
-  return conv_core(xp, w, (int)stride, P, Q, (int)dil, (int)ktrue);
+  return conv_core(xp, w, (int)stride, P, Q, (int)dil, (int)ktrue,
+                   nullptr, nullptr, 512);
 }
 
 // prepadded forward that ALSO emits the BatchNorm stage-1 partials
@@ -1284,7 +1285,7 @@ std::vector<at::Tensor> fedkit_conv2d_fwd_prepadded_bnstats(
   auto part = at::empty({Kout / BN, mtiles, 2, 64},
                         xp.options().dtype(at::kFloat));
   auto y = conv_core(xp, w, (int)stride, P, Q, 1, -1,
-                     part.data_ptr<float>());
+                     part.data_ptr<float>(), nullptr, 512);
   return {y, part};
 }
 
@@ -1304,7 +1305,8 @@ at::Tensor fedkit_conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   int Q = (W + 2 * (int)padding - Reff) / (int)stride + 1;
   at::Tensor xp = padding > 0 ? pad_nhwc(x, padding, padding, padding, padding)
                               : x;
-  return conv_core(xp, w, (int)stride, P, Q, (int)dil, (int)ktrue);
+  return conv_core(xp, w, (int)stride, P, Q, (int)dil, (int)ktrue,
+                   nullptr, nullptr, 512);
 }
 
 // dx = conv_dil(dilate_str(gy), rot180(w)); gy must carry the same

@@ -43,13 +43,13 @@ std::vector<at::Tensor> fedkit_bn_fwd(const at::Tensor& x,
                                       c10::optional<at::Tensor> residual,
                                       bool elu,
                                       c10::optional<at::Tensor> conv_part,
-                                      long pad_out);
+                                      long pad_out, long res_pad);
 std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& gamma,
                                       const at::Tensor& save_mean,
                                       const at::Tensor& save_invstd,
                                       c10::optional<at::Tensor> elu_y,
-                                      bool want_g, long pad_in);
+                                      bool want_g, long pad_in, long res_pad);
 
 at::Tensor fedkit_conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                              long stride, long padding, long dil, long ktrue);
@@ -143,13 +143,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("running_mean"), py::arg("running_var"), py::arg("training"),
         py::arg("momentum"), py::arg("eps"),
         py::arg("residual") = c10::nullopt, py::arg("elu") = false,
-        py::arg("conv_part") = c10::nullopt, py::arg("pad_out") = 0);
+        py::arg("conv_part") = c10::nullopt, py::arg("pad_out") = 0,
+        py::arg("res_pad") = 0);
   m.def("bn_bwd", &fedkit_bn_bwd,
         "NHWC BatchNorm bwd (optionally fused with ELU backward from the "
         "saved output): returns (gx, gw, gb[, g])",
         py::arg("gy"), py::arg("x"), py::arg("gamma"), py::arg("save_mean"),
         py::arg("save_invstd"), py::arg("elu_y") = c10::nullopt,
-        py::arg("want_g") = false, py::arg("pad_in") = 0);
+        py::arg("want_g") = false, py::arg("pad_in") = 0,
+        py::arg("res_pad") = 0);
   m.def("conv2d_fwd", &fedkit_conv2d_fwd,
         "NHWC implicit-GEMM conv fwd on MFMA (square filter, stride 1/2, "
         "optional dilation; ktrue marks channel-padded Kout)",

@@ -50,9 +50,13 @@ class BasicBlock(nn.Module):
     def forward(self, x):
         # bn1's output feeds ONLY conv2 (pad 1): the apply kernel writes
         # the padded image directly (bn_elu pad_out fusion) and conv2
-        # skips its pad pass
+        # skips its pad pass.  bn2's output is padded too when the NEXT
+        # block is identity-shortcut (pad_out_next, set by ResNet): its
+        # conv1 consumes the marker and its bn2 reads the padded residual
+        # at interior coordinates.
         out = bn_elu(self.bn1, self.conv1(x), pad_out=1)
-        return bn_elu(self.bn2, self.conv2(out), residual=self.shortcut(x))
+        return bn_elu(self.bn2, self.conv2(out), residual=self.shortcut(x),
+                      pad_out=getattr(self, "pad_out_next", 0))
 
 
 class Bottleneck(nn.Module):
@@ -96,8 +100,15 @@ class ResNet(nn.Module):
 
     def _make_layer(self, block, planes, n, stride):
         layers = []
-        for s in [stride] + [1] * (n - 1):
-            layers.append(block(self.in_planes, planes, s))
+        for i, s in enumerate([stride] + [1] * (n - 1)):
+            blk = block(self.in_planes, planes, s)
+            # bn2 -> next block's conv1 pad fusion: legal when the
+            # successor is the NEXT block of the SAME stage (identity
+            # shortcut: conv1 takes the marker, the residual is read at
+            # interior coords); stage-crossing successors have a 1x1
+            # shortcut that needs the unpadded image
+            blk.pad_out_next = 1 if (i + 1 < n and block is BasicBlock) else 0
+            layers.append(blk)
             self.in_planes = planes * block.expansion
         return nn.Sequential(*layers)
 
@@ -119,7 +130,8 @@ class ResNet(nn.Module):
 
     def forward(self, x):
         self._precast(x)
-        out = bn_elu(self.bn1, self.conv1(x))
+        # the stem's output feeds layer1.0 (identity shortcut): fused pad
+        out = bn_elu(self.bn1, self.conv1(x), pad_out=1)
         out = self.layer1(out)
         out = self.layer2(out)
         out = self.layer3(out)

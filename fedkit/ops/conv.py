@@ -410,6 +410,11 @@ class FedConv2d(nn.Conv2d):
     def forward(self, x):
         # producer-padded input (bn_elu(pad_out=...) fusion)?
         prepad = getattr(x, "_fedkit_prepad", 0)
+        if prepad and prepad != self.padding[0]:
+            # marker mismatch (defensive; the resnet wiring never hits
+            # this): slice back to the interior and take the normal path
+            x = x[:, :, prepad:-prepad, prepad:-prepad]
+            prepad = 0
         if _native(x) and self.bias is None and self.groups == 1 \
                 and self.dilation == (1, 1) and self.kernel_size[0] in (1, 3):
             if x.dtype != torch.bfloat16 and torch.is_autocast_enabled():

@@ -67,15 +67,16 @@ class _BnActFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
                 training, momentum, eps, residual, conv_part=None,
-                pad_out=0):
+                pad_out=0, res_pad=0):
         y, save_mean, save_invstd = _ext().bn_fwd(
             x, weight, bias, running_mean, running_var,
             bool(training), float(momentum), float(eps),
             residual=residual, elu=True, conv_part=conv_part,
-            pad_out=int(pad_out))
+            pad_out=int(pad_out), res_pad=int(res_pad))
         ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
         ctx.has_res = residual is not None
         ctx.pad_out = int(pad_out)
+        ctx.res_pad = int(res_pad)
         return y
 
     @staticmethod
@@ -87,10 +88,10 @@ class _BnActFn(torch.autograd.Function):
         # only when the residual branch needs it
         out = _ext().bn_bwd(gy, x, weight, save_mean, save_invstd,
                             elu_y=y, want_g=ctx.has_res,
-                            pad_in=ctx.pad_out)
+                            pad_in=ctx.pad_out, res_pad=ctx.res_pad)
         gres = out[3] if ctx.has_res else None
         return (out[0], out[1], out[2], None, None, None, None, None, gres,
-                None, None)
+                None, None, None)
 
 
 class FedBatchNorm2d(nn.BatchNorm2d):
@@ -158,14 +159,19 @@ def bn_elu(bn: FedBatchNorm2d, x: torch.Tensor,
         bn._prep(x)
         part = getattr(x, "_fedkit_bn_stats", None) if bn.training else None
         x = x.contiguous(memory_format=torch.channels_last)
+        res_pad = 0
         if residual is not None:
+            # a producer-padded residual (marker) is read at interior
+            # coordinates by the apply kernel; its gradient comes back
+            # padded with zero borders
+            res_pad = getattr(residual, "_fedkit_prepad", 0)
             residual = residual.contiguous(memory_format=torch.channels_last)
             if residual.dtype != x.dtype:
                 residual = residual.to(x.dtype)
         y = _BnActFn.apply(x, bn.weight, bn.bias,
                            bn.running_mean, bn.running_var,
                            bn.training, bn.momentum, bn.eps, residual,
-                           part, pad_out)
+                           part, pad_out, res_pad)
         if pad_out:
             y._fedkit_prepad = pad_out
         return y
