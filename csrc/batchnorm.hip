@@ -578,12 +578,17 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
   at::Tensor gout;
   if (want_g) {
     // residual grad matches the residual's geometry; PADDED residuals get
-    // zero borders (at::zeros) and the kernel writes the interior
-    gout = rpad == 0
-        ? at::empty_like(x)
-        : at::zeros({(long long)N, (long long)C, (long long)rpg.Hp,
-                     (long long)rpg.Wp},
-                    x.options().memory_format(at::MemoryFormat::ChannelsLast));
+    // zero borders and the kernel writes the interior.
+    // (empty + zero_: at::zeros DROPS the memory_format in options — the
+    // same quirk bwd_data's wrot works around)
+    if (rpad == 0) {
+      gout = at::empty_like(x);
+    } else {
+      gout = at::empty({(long long)N, (long long)C, (long long)rpg.Hp,
+                        (long long)rpg.Wp},
+                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
+      gout.zero_();
+    }
   }
   using T0 = std::integral_constant<bool, false>;
   using T1 = std::integral_constant<bool, true>;

@@ -1074,3 +1074,43 @@ def test_resnet_block_pad_fusion_end_to_end():
     assert frob_err(y, y2) < 5e-2
     for n, p in blk.named_parameters():
         assert frob_err(nat[n], p.grad.float()) < 8e-2, n
+
+
+def test_bn_elu_padded_residual():
+    """bn_elu with a producer-PADDED residual (phase-B fusion): values and
+    the residual's padded-domain gradient vs the composed form."""
+    from fedkit.ops.norm import FedBatchNorm2d, bn_elu
+    torch.manual_seed(43)
+    C = 64
+    bn_a = FedBatchNorm2d(C).cuda().train()
+    bn_b = FedBatchNorm2d(C).cuda().train()
+    x0 = torch.randn(8, C, 16, 16, device="cuda", requires_grad=True)
+    xmid = torch.randn(8, C, 16, 16, device="cuda")
+    # producer: padded output (the "block input")
+    blk_in = bn_elu(bn_a, x0.contiguous(memory_format=torch.channels_last),
+                    pad_out=1)
+    # consumer bn2: unpadded conv output + the padded residual
+    y = bn_elu(bn_b, xmid.contiguous(memory_format=torch.channels_last),
+               residual=blk_in)
+    assert y.shape == (8, C, 16, 16)
+    loss = y.float().square().sum()
+    loss.backward()
+    gx0 = x0.grad.clone()
+    x0.grad = None
+    for m in (bn_a, bn_b):
+        m.zero_grad(set_to_none=True)
+
+    bn_a2 = FedBatchNorm2d(C).cuda().train()
+    bn_b2 = FedBatchNorm2d(C).cuda().train()
+    with torch.no_grad():
+        for dst, src in ((bn_a2, bn_a), (bn_b2, bn_b)):
+            dst.weight.copy_(src.weight)
+            dst.bias.copy_(src.bias)
+    ref_in = F.elu(F.batch_norm(x0, bn_a2.running_mean, bn_a2.running_var,
+                                bn_a2.weight, bn_a2.bias, True, 0.1, 1e-5))
+    yref = F.elu(F.batch_norm(xmid, bn_b2.running_mean, bn_b2.running_var,
+                              bn_b2.weight, bn_b2.bias, True, 0.1, 1e-5)
+                 + ref_in)
+    yref.float().square().sum().backward()
+    assert rel_err(y, yref) < 1e-3
+    assert rel_err(gx0, x0.grad) < 1e-3
