@@ -29,7 +29,8 @@ def common(K, args, **kw):
     return FedConfig(
         K=K, model="Net", default_batch=128,
         Nloop=args.nloop, Nepoch=1, Nadmm=args.nadmm,
-        check_results=False, save_model=False, load_model=False,
+        check_results=bool(getattr(args, "trajectory", False)),
+        save_model=False, load_model=False,
         init_model=True, biased_input=True, be_verbose=False,
         use_cuda=True, max_steps_per_epoch=args.max_steps, **kw)
 
@@ -53,6 +54,8 @@ def main():
                     help="penalty warm-up rounds per block (FedProx/ADMM)")
     ap.add_argument("--jsonl-dir", type=str, default="",
                     help="write per-round residual JSONLs here")
+    ap.add_argument("--trajectory", action="store_true",
+                    help="evaluate every round (acc curves in the JSONLs)")
     args = ap.parse_args()
     if args.quick:
         args.nloop, args.nadmm, args.max_steps = 1, 1, 3
@@ -74,12 +77,16 @@ def main():
 
     if want("standalone_K10"):
         cfg = common(10, args, strategy="none")
+        if args.jsonl_dir:
+            cfg.jsonl_path = os.path.join(args.jsonl_dir, "standalone10.jsonl")
         job = FederatedJob(cfg)
         job.run()
         results["standalone_K10"] = final_accs(job)
 
     if want("fedavg_K10"):
         cfg = common(10, args, strategy="fedavg")
+        if args.jsonl_dir:
+            cfg.jsonl_path = os.path.join(args.jsonl_dir, "fedavg.jsonl")
         job = FederatedJob(cfg)
         job.run()
         results["fedavg_K10"] = final_accs(job)
