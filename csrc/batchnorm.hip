@@ -319,19 +319,19 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
     m[j] = mean[c0 + j];
     is[j] = invstd[c0 + j];
   }
-  // gy / yout may live in a PADDED image (apply-into-pad forward)
-  auto gidx = [&](long long i) {
+  // the SAVED y may live in a PADDED image (apply-into-pad forward);
+  // gy and every gradient stay in the unpadded domain
+  auto yidx = [&](long long i) {
     return pg.pad ? pad_vec_idx(pg, i, Cv) : i;
   };
   float sdy[VEC] = {}, sdyx[VEC] = {};
   long long i = i0;
   for (; i + stride < total; i += 2 * stride) {
-    long long ig0 = gidx(i), ig1 = gidx(i + stride);
-    V a0 = xv[i], g0 = gv[ig0], a1 = xv[i + stride], g1 = gv[ig1];
+    V a0 = xv[i], g0 = gv[i], a1 = xv[i + stride], g1 = gv[i + stride];
     V y0, y1;
     if (ELU) {
-      y0 = yv[ig0];
-      y1 = yv[ig1];
+      y0 = yv[yidx(i)];
+      y1 = yv[yidx(i + stride)];
     }
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
@@ -346,10 +346,9 @@ __global__ void bn_bwd_partials_kernel(const T* __restrict__ x,
     }
   }
   for (; i < total; i += stride) {
-    long long ig = gidx(i);
-    V a = xv[i], g = gv[ig];
+    V a = xv[i], g = gv[i];
     V yy;
-    if (ELU) yy = yv[ig];
+    if (ELU) yy = yv[yidx(i)];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       float gg = to_f32(g.v[j]);
@@ -392,8 +391,7 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
                                     const float* __restrict__ gamma,
                                     const float* __restrict__ ws,
                                     long long M, long long nvec, int Cv,
-                                    PadGeom pg,
-                                    PadGeom rpg /* gout geometry */) {
+                                    PadGeom pg /* saved-y geometry */) {
   using V = VecT<T, VEC>;
   const V* xv = reinterpret_cast<const V*>(x);
   const V* gv = reinterpret_cast<const V*>(gy);
@@ -404,11 +402,10 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
   float inv_count = 1.f / (float)M;
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
        i < nvec; i += (long long)gridDim.x * blockDim.x) {
-    long long ig = pg.pad ? pad_vec_idx(pg, i, Cv) : i;
     int c0 = (int)(i % Cv) * VEC;
-    V a = xv[i], g = gv[ig], r, go;
+    V a = xv[i], g = gv[i], r, go;
     V yy;
-    if (ELU) yy = yv[ig];
+    if (ELU) yy = yv[pg.pad ? pad_vec_idx(pg, i, Cv) : i];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       int c = c0 + j;
@@ -422,7 +419,7 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
       from_f32(val, r.v[j]);
     }
     ov[i] = r;
-    if (WANTG) gov[rpg.pad ? pad_vec_idx(rpg, i, Cv) : i] = go;
+    if (WANTG) gov[i] = go;
   }
 }
 
@@ -555,16 +552,19 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& save_mean,
                                       const at::Tensor& save_invstd,
                                       c10::optional<at::Tensor> elu_y,
-                                      bool want_g, long pad_in,
-                                      long res_pad) {
+                                      bool want_g, long pad_in) {
   check_nhwc(x);
   check_nhwc(gy);
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   long long M = (long long)N * H * W;
+  // pad_in: the SAVED y (elu_y) lives in a padded image (apply-into-pad
+  // forward); gy and every gradient stay in the UNPADDED domain
   const int pad = (int)pad_in;
   PadGeom pg = {H, W, H + 2 * pad, W + 2 * pad, pad};
-  TORCH_CHECK(gy.size(2) == pg.Hp && gy.size(3) == pg.Wp,
-              "bn_bwd: gy geometry mismatch for pad_in=", pad);
+  TORCH_CHECK(gy.size(2) == H && gy.size(3) == W,
+              "bn_bwd: gy must be unpadded");
+  TORCH_CHECK(!pad || (elu_y.has_value() && elu_y->size(2) == pg.Hp),
+              "bn_bwd: elu_y geometry mismatch for pad_in=", pad);
   auto fopts = x.options().dtype(at::kFloat);
   auto ws = at::empty({2, C}, fopts);
   auto gx = at::empty_like(x);
@@ -573,23 +573,8 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
   const bool elu = elu_y.has_value();
   const void* y_ptr = elu ? elu_y->data_ptr() : nullptr;
   TORCH_CHECK(!want_g || elu, "want_g requires the fused-ELU backward");
-  const int rpad = (int)res_pad;
-  PadGeom rpg = {H, W, H + 2 * rpad, W + 2 * rpad, rpad};
   at::Tensor gout;
-  if (want_g) {
-    // residual grad matches the residual's geometry; PADDED residuals get
-    // zero borders and the kernel writes the interior.
-    // (empty + zero_: at::zeros DROPS the memory_format in options — the
-    // same quirk bwd_data's wrot works around)
-    if (rpad == 0) {
-      gout = at::empty_like(x);
-    } else {
-      gout = at::empty({(long long)N, (long long)C, (long long)rpg.Hp,
-                        (long long)rpg.Wp},
-                       x.options().memory_format(at::MemoryFormat::ChannelsLast));
-      gout.zero_();
-    }
-  }
+  if (want_g) gout = at::empty_like(x);   // residual grad is UNPADDED
   using T0 = std::integral_constant<bool, false>;
   using T1 = std::integral_constant<bool, true>;
   DISPATCH_F32_BF16(x, "bn_bwd_partials", {
@@ -630,7 +615,7 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                          save_mean.data_ptr<float>(),
                          save_invstd.data_ptr<float>(),
                          gamma_f.data_ptr<float>(), ws.data_ptr<float>(), M,
-                         nvec, C / VEC, pg, rpg);
+                         nvec, C / VEC, pg);
     };
     if (elu && want_g)  launch2(T1{}, T1{});
     else if (elu)       launch2(T1{}, T0{});

@@ -49,7 +49,7 @@ std::vector<at::Tensor> fedkit_bn_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& save_mean,
                                       const at::Tensor& save_invstd,
                                       c10::optional<at::Tensor> elu_y,
-                                      bool want_g, long pad_in, long res_pad);
+                                      bool want_g, long pad_in);
 
 at::Tensor fedkit_conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                              long stride, long padding, long dil, long ktrue);
@@ -150,8 +150,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "saved output): returns (gx, gw, gb[, g])",
         py::arg("gy"), py::arg("x"), py::arg("gamma"), py::arg("save_mean"),
         py::arg("save_invstd"), py::arg("elu_y") = c10::nullopt,
-        py::arg("want_g") = false, py::arg("pad_in") = 0,
-        py::arg("res_pad") = 0);
+        py::arg("want_g") = false, py::arg("pad_in") = 0);
   m.def("conv2d_fwd", &fedkit_conv2d_fwd,
         "NHWC implicit-GEMM conv fwd on MFMA (square filter, stride 1/2, "
         "optional dilation; ktrue marks channel-padded Kout)",

@@ -39,14 +39,15 @@ def _ext():
 
 class _ConvFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, stride, padding, prepadded=False):
+    def forward(ctx, x, w, stride, padding, pbuf=None):
         # x: [N,C,H,W] logical, carried NHWC-contiguous; w: [K,C,R,S].
         # The padded input is computed ONCE and saved, so bwd-weight reuses
         # it instead of re-padding (and x itself need not be kept).
-        # prepadded=True (round 2): x arrived ALREADY padded from the
-        # producing bn_elu(pad_out=...) apply kernel — skip the pad pass;
-        # backward then returns the padded-domain input gradient (the
-        # bn backward reads it at padded coordinates).
+        # pbuf (round 2): the producing bn_elu(pad_out=...) already wrote
+        # the PADDED image; x is its interior view (the autograd edge).
+        # The forward and bwd-weight consume pbuf directly (no pad pass);
+        # bwd-data computes the input gradient at the ORIGINAL unpadded
+        # geometry, so no border-gradient work is wasted.
         # FEDKIT_CONV_BNSTATS=1: the epilogue also emits the BatchNorm
         # stage-1 partials and the downstream FedBatchNorm2d skips its own
         # reduction pass.  MEASURED NET LOSS at CIFAR sizes (step 3.9 ->
@@ -61,10 +62,10 @@ class _ConvFn(torch.autograd.Function):
             y = ext.conv2d_fwd(x, w, stride, padding)
             part = y.new_empty(0)
             ctx.save_for_backward(x, w)
-        elif prepadded:
-            y = ext.conv2d_fwd_prepadded(x, w, stride)
+        elif pbuf is not None:
+            y = ext.conv2d_fwd_prepadded(pbuf, w, stride)
             part = y.new_empty(0)
-            ctx.save_for_backward(x, w)
+            ctx.save_for_backward(pbuf, w)
         else:
             xp = ext.conv2d_pad_input(x, padding)
             if want_stats:
@@ -74,7 +75,6 @@ class _ConvFn(torch.autograd.Function):
                 part = y.new_empty(0)
             ctx.save_for_backward(xp, w)
         ctx.small_c = small_c
-        ctx.prepadded = prepadded
         ctx.stride = stride
         ctx.padding = padding
         ctx.hw = (x.shape[2], x.shape[3])
@@ -87,14 +87,8 @@ class _ConvFn(torch.autograd.Function):
         gy = gy.contiguous(memory_format=torch.channels_last)
         gx = gw = None
         if ctx.needs_input_grad[0]:
-            if ctx.prepadded:
-                # grad wrt the PADDED input: same conv adjoint with pad=0
-                # over the padded geometry
-                gx = _ext().conv2d_bwd_data(gy, w, ctx.stride, 0,
-                                            ctx.hw[0], ctx.hw[1])
-            else:
-                gx = _ext().conv2d_bwd_data(gy, w, ctx.stride, ctx.padding,
-                                            ctx.hw[0], ctx.hw[1])
+            gx = _ext().conv2d_bwd_data(gy, w, ctx.stride, ctx.padding,
+                                        ctx.hw[0], ctx.hw[1])
         if ctx.needs_input_grad[1]:
             if ctx.small_c:
                 gw = _ext().conv2d_bwd_weight(gy, xsaved, ctx.stride,
@@ -408,13 +402,12 @@ def dilated_bank(x, mods):
 
 class FedConv2d(nn.Conv2d):
     def forward(self, x):
-        # producer-padded input (bn_elu(pad_out=...) fusion)?
-        prepad = getattr(x, "_fedkit_prepad", 0)
-        if prepad and prepad != self.padding[0]:
-            # marker mismatch (defensive; the resnet wiring never hits
-            # this): slice back to the interior and take the normal path
-            x = x[:, :, prepad:-prepad, prepad:-prepad]
-            prepad = 0
+        # producer-padded input? (bn_elu(pad_out=...) fusion: x is the
+        # interior view, mk[0] the padded buffer.)  A marker mismatch is
+        # harmless — x alone is fully valid, the marker is just skipped.
+        mk = getattr(x, "_fedkit_padded", None)
+        pbuf = mk[0] if (mk is not None and mk[1] == self.padding[0]
+                         and self.padding[0] > 0) else None
         if _native(x) and self.bias is None and self.groups == 1 \
                 and self.dilation == (1, 1) and self.kernel_size[0] in (1, 3):
             if x.dtype != torch.bfloat16 and torch.is_autocast_enabled():
@@ -422,8 +415,11 @@ class FedConv2d(nn.Conv2d):
             if x.dtype == torch.bfloat16:
                 # the MI355X fast path: bf16 NHWC through the MFMA kernels;
                 # the fp32->bf16 weight cast is autograd-tracked, so bwd-weight
-                # gradients land on the fp32 master copy
-                x = x.contiguous(memory_format=torch.channels_last)
+                # gradients land on the fp32 master copy.  With a producer-
+                # padded buffer, x is only the autograd edge — do NOT
+                # materialize the (non-contiguous) interior view.
+                if pbuf is None:
+                    x = x.contiguous(memory_format=torch.channels_last)
                 once = self.__dict__.pop("_w16_once", None)
                 if once is not None and once[0] == self.weight._version:
                     w = once[1]
@@ -432,17 +428,11 @@ class FedConv2d(nn.Conv2d):
                         self, "_w16", self.weight,
                         lambda: self.weight.to(torch.bfloat16).contiguous(
                             memory_format=torch.channels_last))
-                y, part = _ConvFn.apply(x, w, self.stride[0], self.padding[0],
-                                        prepad == self.padding[0] > 0)
+                y, part = _ConvFn.apply(x, w, self.stride[0],
+                                        self.padding[0], pbuf)
                 if self.training and part.numel():
                     y._fedkit_bn_stats = part
                 return y
-        if prepad and prepad == self.padding[0]:
-            # fallback consuming an already-padded image: pad 0
-            return F.conv2d(x, self.weight.to(x.dtype),
-                            self.bias.to(x.dtype) if self.bias is not None
-                            else None, self.stride, 0, self.dilation,
-                            self.groups)
         return F.conv2d(x, self.weight.to(x.dtype),
                         self.bias.to(x.dtype) if self.bias is not None else None,
                         self.stride, self.padding, self.dilation, self.groups)

@@ -160,20 +160,26 @@ def bn_elu(bn: FedBatchNorm2d, x: torch.Tensor,
         part = getattr(x, "_fedkit_bn_stats", None) if bn.training else None
         x = x.contiguous(memory_format=torch.channels_last)
         res_pad = 0
+        res_buf = None
         if residual is not None:
-            # a producer-padded residual (marker) is read at interior
-            # coordinates by the apply kernel; its gradient comes back
-            # padded with zero borders
-            res_pad = getattr(residual, "_fedkit_prepad", 0)
-            residual = residual.contiguous(memory_format=torch.channels_last)
-            if residual.dtype != x.dtype:
-                residual = residual.to(x.dtype)
-        y = _BnActFn.apply(x, bn.weight, bn.bias,
-                           bn.running_mean, bn.running_var,
-                           bn.training, bn.momentum, bn.eps, residual,
-                           part, pad_out, res_pad)
+            # a producer-padded residual ships its padded buffer for the
+            # in-kernel interior read; the view keeps the autograd edge
+            mk = getattr(residual, "_fedkit_padded", None)
+            if mk is not None:
+                res_buf, res_pad = mk
+                if res_buf.dtype != x.dtype:
+                    res_buf = res_buf.to(x.dtype)
+            else:
+                residual = residual.contiguous(
+                    memory_format=torch.channels_last)
+                if residual.dtype != x.dtype:
+                    residual = residual.to(x.dtype)
+        y, ypad = _BnActFn.apply(x, bn.weight, bn.bias,
+                                 bn.running_mean, bn.running_var,
+                                 bn.training, bn.momentum, bn.eps, residual,
+                                 part, pad_out, res_pad, res_buf)
         if pad_out:
-            y._fedkit_prepad = pad_out
+            y._fedkit_padded = (ypad.detach(), pad_out)
         return y
     y = nn.BatchNorm2d.forward(bn, x)
     if residual is not None:

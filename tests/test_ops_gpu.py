@@ -1016,20 +1016,24 @@ def test_welford_update_fused_vs_torch():
 
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
 def test_bn_elu_pad_out_fusion(dtype):
-    """bn_elu(pad_out=1) emits the padded image (borders zero) and its
-    backward matches the composed pad(elu(bn(x))) autograd exactly."""
+    """bn_elu(pad_out=1): the padded side-buffer equals pad(elu(bn(x)))
+    with zero borders, the autograd output is the interior view, and the
+    backward (gy in the unpadded domain, saved-y read at padded coords)
+    matches the composed autograd."""
     from fedkit.ops.norm import FedBatchNorm2d, bn_elu
     torch.manual_seed(37)
     C = 64
     bn = FedBatchNorm2d(C).cuda().train()
     x = torch.randn(8, C, 16, 16, device="cuda", dtype=dtype,
                     requires_grad=True)
-    xp = bn_elu(bn, x.contiguous(memory_format=torch.channels_last),
-                pad_out=1)
-    assert xp.shape == (8, C, 18, 18)
-    assert getattr(xp, "_fedkit_prepad", 0) == 1
-    gy = torch.randn_like(xp)
-    xp.backward(gy)
+    y = bn_elu(bn, x.contiguous(memory_format=torch.channels_last),
+               pad_out=1)
+    assert y.shape == (8, C, 16, 16)
+    buf, pad = y._fedkit_padded
+    assert pad == 1 and buf.shape == (8, C, 18, 18)
+    assert float(buf[:, :, 0, :].abs().max()) == 0.0   # borders zero
+    gy = torch.randn_like(x)
+    y.backward(gy)
     gx = x.grad.clone()
     gw, gb = bn.weight.grad.clone(), bn.bias.grad.clone()
     x.grad = bn.weight.grad = bn.bias.grad = None
@@ -1038,12 +1042,13 @@ def test_bn_elu_pad_out_fusion(dtype):
     with torch.no_grad():
         bn2.weight.copy_(bn.weight)
         bn2.bias.copy_(bn.bias)
-    yref = F.pad(F.elu(F.batch_norm(
+    yref = F.elu(F.batch_norm(
         x.float(), bn2.running_mean, bn2.running_var, bn2.weight, bn2.bias,
-        True, 0.1, 1e-5)), (1, 1, 1, 1))
+        True, 0.1, 1e-5))
     yref.backward(gy.float())
     tol = 1e-3 if dtype == torch.float32 else 5e-2
-    assert rel_err(xp, yref) < tol
+    assert rel_err(y, yref) < tol
+    assert rel_err(buf[:, :, 1:-1, 1:-1], yref) < tol
     assert rel_err(gx, x.grad) < tol
     assert rel_err(gw, bn2.weight.grad) < tol
     assert rel_err(gb, bn2.bias.grad) < tol
@@ -1086,10 +1091,12 @@ def test_bn_elu_padded_residual():
     bn_b = FedBatchNorm2d(C).cuda().train()
     x0 = torch.randn(8, C, 16, 16, device="cuda", requires_grad=True)
     xmid = torch.randn(8, C, 16, 16, device="cuda")
-    # producer: padded output (the "block input")
+    # producer: padded side-buffer (the "block input")
     blk_in = bn_elu(bn_a, x0.contiguous(memory_format=torch.channels_last),
                     pad_out=1)
-    # consumer bn2: unpadded conv output + the padded residual
+    assert blk_in._fedkit_padded[1] == 1
+    # consumer bn2: unpadded conv output + the marked residual (its padded
+    # buffer is read at interior coordinates in-kernel)
     y = bn_elu(bn_b, xmid.contiguous(memory_format=torch.channels_last),
                residual=blk_in)
     assert y.shape == (8, C, 16, 16)
