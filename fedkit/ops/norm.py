@@ -57,41 +57,55 @@ class _BnActFn(torch.autograd.Function):
     g = elu'(y) * gy (from the saved post-activation output), which is both
     the residual grad and the BN-output grad.
 
-    pad_out > 0 (round 2): the apply writes directly into the PADDED image
-    the next conv consumes (borders zeroed in-kernel) — removes the
-    separate pad launch plus a full read+write pass of y.  The output is
-    padded; backward receives the padded-domain gradient and the bn_bwd
-    kernels read gy / elu-y at padded coordinates.
+    pad_out > 0 (round 2): the apply ALSO writes the PADDED image the
+    next conv consumes (borders zeroed in-kernel) — removes the separate
+    pad launch plus a full read+write pass of y.  The AUTOGRAD output is
+    the interior VIEW of that buffer (unpadded shape), so every gradient
+    stays in the unpadded domain — the consuming conv reads the padded
+    buffer out-of-band (the `_fedkit_padded` marker) but computes its
+    input gradient at the ORIGINAL geometry (no wasted border-gradient
+    work); only the bn_bwd kernels' reads of the saved y use the pad
+    geometry.  A producer-padded RESIDUAL ships its buffer via res_buf
+    (read at interior coords in-kernel) while the interior view carries
+    the autograd edge.
     """
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
                 training, momentum, eps, residual, conv_part=None,
-                pad_out=0, res_pad=0):
+                pad_out=0, res_pad=0, res_buf=None):
         y, save_mean, save_invstd = _ext().bn_fwd(
             x, weight, bias, running_mean, running_var,
             bool(training), float(momentum), float(eps),
-            residual=residual, elu=True, conv_part=conv_part,
+            residual=res_buf if res_buf is not None else residual,
+            elu=True, conv_part=conv_part,
             pad_out=int(pad_out), res_pad=int(res_pad))
         ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
         ctx.has_res = residual is not None
         ctx.pad_out = int(pad_out)
-        ctx.res_pad = int(res_pad)
-        return y
+        if pad_out:
+            p = int(pad_out)
+            ypad = y
+            yview = y[:, :, p:-p, p:-p]
+        else:
+            ypad = y.new_empty(0)
+            yview = y
+        ctx.mark_non_differentiable(ypad)
+        return yview, ypad
 
     @staticmethod
-    def backward(ctx, gy):
+    def backward(ctx, gy, _gypad):
         x, weight, save_mean, save_invstd, y = ctx.saved_tensors
         gy = gy.contiguous(memory_format=torch.channels_last)
         # elu' is fused into both bn_bwd kernels (recomputed from the saved
-        # output y); the residual grad g is streamed out of the apply kernel
-        # only when the residual branch needs it
+        # output y, read at padded coords when pad_out was used); the
+        # residual grad g streams out of the apply kernel on demand
         out = _ext().bn_bwd(gy, x, weight, save_mean, save_invstd,
                             elu_y=y, want_g=ctx.has_res,
-                            pad_in=ctx.pad_out, res_pad=ctx.res_pad)
+                            pad_in=ctx.pad_out)
         gres = out[3] if ctx.has_res else None
         return (out[0], out[1], out[2], None, None, None, None, None, gres,
-                None, None, None)
+                None, None, None, None)
 
 
 class FedBatchNorm2d(nn.BatchNorm2d):
