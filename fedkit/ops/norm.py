@@ -166,9 +166,10 @@ def bn_elu(bn: FedBatchNorm2d, x: torch.Tensor,
            residual: torch.Tensor = None, pad_out: int = 0) -> torch.Tensor:
     """elu(bn(x) [+ residual]) — fused on GPU, composed on CPU.
 
-    pad_out > 0: the GPU path emits the PADDED image (marked with
-    `_fedkit_prepad`) so the consuming conv skips its pad pass; the CPU
-    path ignores it (the conv pads itself)."""
+    pad_out > 0: the GPU path returns the interior view of a padded
+    side-buffer (shipped via the `_fedkit_padded` marker) so the consuming
+    conv skips its pad pass; the CPU path ignores it (the conv pads
+    itself)."""
     if _native(x):
         bn._prep(x)
         part = getattr(x, "_fedkit_bn_stats", None) if bn.training else None
