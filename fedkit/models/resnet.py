@@ -15,12 +15,18 @@ The module graph (and therefore state_dict / parameter order / block
 partition) is identical to the reference's.
 """
 
+import os
+
 import torch.nn as nn
 
 from ..ops.conv import FedConv2d
 from ..ops.linear import FedLinear
 from ..ops.pool import avg_pool2d
 from ..ops.norm import FedBatchNorm2d, bn_elu
+
+
+# A/B gate for the bn-apply-into-pad fusion (default on)
+_PAD_FUSE = 0 if os.environ.get("FEDKIT_NO_PADFUSE") == "1" else 1
 
 
 def _conv3x3(cin, cout, stride=1):
@@ -54,9 +60,9 @@ class BasicBlock(nn.Module):
         # block is identity-shortcut (pad_out_next, set by ResNet): its
         # conv1 consumes the marker and its bn2 reads the padded residual
         # at interior coordinates.
-        out = bn_elu(self.bn1, self.conv1(x), pad_out=1)
+        out = bn_elu(self.bn1, self.conv1(x), pad_out=_PAD_FUSE)
         return bn_elu(self.bn2, self.conv2(out), residual=self.shortcut(x),
-                      pad_out=getattr(self, "pad_out_next", 0))
+                      pad_out=_PAD_FUSE and getattr(self, "pad_out_next", 0))
 
 
 class Bottleneck(nn.Module):
@@ -131,7 +137,7 @@ class ResNet(nn.Module):
     def forward(self, x):
         self._precast(x)
         # the stem's output feeds layer1.0 (identity shortcut): fused pad
-        out = bn_elu(self.bn1, self.conv1(x), pad_out=1)
+        out = bn_elu(self.bn1, self.conv1(x), pad_out=_PAD_FUSE)
         out = self.layer1(out)
         out = self.layer2(out)
         out = self.layer3(out)
