@@ -13,7 +13,7 @@ def _run(opt, closure, steps):
     losses = []
     for _ in range(steps):
         loss = opt.step(closure)
-        losses.append(float(loss))
+        losses.append(float(loss.detach() if hasattr(loss, 'detach') else loss))
     return losses
 
 
@@ -119,7 +119,7 @@ def test_zero_grad_early_exit():
         return loss
 
     loss = opt.step(closure)     # grad is 0 at the optimum: returns at once
-    assert float(loss) == 0.0
+    assert float(loss.detach()) == 0.0
     assert float(x) == 0.0
 
 
