@@ -120,7 +120,7 @@ def test_zero_grad_early_exit():
 
     loss = opt.step(closure)     # grad is 0 at the optimum: returns at once
     assert float(loss.detach()) == 0.0
-    assert float(x) == 0.0
+    assert float(x.detach()) == 0.0
 
 
 def test_multi_tensor_params():
