@@ -1121,3 +1121,47 @@ def test_bn_elu_padded_residual():
     yref.float().square().sum().backward()
     assert rel_err(y, yref) < 1e-3
     assert rel_err(gx0, x0.grad) < 1e-3
+
+
+def test_round_checkpoint_resume_gpu(tmp_path):
+    """Kill-and-resume on GPU (CUDA RNG state in the sidecar): the resumed
+    FedAvg run reproduces the uninterrupted run's final weights."""
+    from fedkit.parallel import FedConfig, FederatedJob
+
+    def cfg_for(sub, **kw):
+        d = tmp_path / sub
+        d.mkdir(exist_ok=True)
+        return FedConfig(K=2, default_batch=32, Nloop=2, Nepoch=1, Nadmm=1,
+                         use_cuda=True, check_results=False, dtype="bf16",
+                         max_steps_per_epoch=2, save_model=False,
+                         strategy="fedavg", model="Net",
+                         round_checkpoint=True, ckpt_prefix=str(d / "s"),
+                         **kw)
+
+    job_a = FederatedJob(cfg_for("a"))
+    job_a.run()
+    sd_a = {ck: {k: v.clone() for k, v in job_a.nets[ck].state_dict().items()}
+            for ck in (0, 1)}
+
+    class Killed(Exception):
+        pass
+
+    nl = {"v": 0, "seen": -1}
+
+    def hook(job, ci):
+        if ci <= nl["seen"]:
+            nl["v"] += 1
+        nl["seen"] = ci
+        if (nl["v"], ci) == (1, 2):
+            raise Killed
+
+    job_b = FederatedJob(cfg_for("b"), block_hook=hook)
+    with pytest.raises(Killed):
+        job_b.run()
+    job_c = FederatedJob(cfg_for("b", load_model=True))
+    job_c.run()
+    for ck in (0, 1):
+        sd_c = job_c.nets[ck].state_dict()
+        for k in sd_a[ck]:
+            assert torch.allclose(sd_a[ck][k].float(), sd_c[k].float(),
+                                  atol=1e-6), (ck, k)
