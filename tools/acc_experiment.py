@@ -27,9 +27,10 @@ from fedkit.parallel.runtime import run_standalone
 
 def common(K, args, **kw):
     return FedConfig(
-        K=K, model="Net", default_batch=128,
+        K=K, model=args.model, default_batch=128,
         Nloop=args.nloop, Nepoch=1, Nadmm=args.nadmm,
         check_results=bool(getattr(args, "trajectory", False)),
+        dtype=getattr(args, "dtype", "fp32"),
         save_model=False, load_model=False,
         init_model=True, biased_input=True, be_verbose=False,
         use_cuda=True, max_steps_per_epoch=args.max_steps, **kw)
@@ -56,6 +57,9 @@ def main():
                     help="write per-round residual JSONLs here")
     ap.add_argument("--trajectory", action="store_true",
                     help="evaluate every round (acc curves in the JSONLs)")
+    ap.add_argument("--model", type=str, default="Net")
+    ap.add_argument("--K", type=int, default=10)
+    ap.add_argument("--dtype", type=str, default="fp32")
     args = ap.parse_args()
     if args.quick:
         args.nloop, args.nadmm, args.max_steps = 1, 1, 3
@@ -76,7 +80,7 @@ def main():
         results["standalone_K1"] = final_accs(job)
 
     if want("standalone_K10"):
-        cfg = common(10, args, strategy="none")
+        cfg = common(args.K, args, strategy="none")
         if args.jsonl_dir:
             cfg.jsonl_path = os.path.join(args.jsonl_dir, "standalone10.jsonl")
         job = FederatedJob(cfg)
@@ -84,7 +88,7 @@ def main():
         results["standalone_K10"] = final_accs(job)
 
     if want("fedavg_K10"):
-        cfg = common(10, args, strategy="fedavg")
+        cfg = common(args.K, args, strategy="fedavg")
         if args.jsonl_dir:
             cfg.jsonl_path = os.path.join(args.jsonl_dir, "fedavg.jsonl")
         job = FederatedJob(cfg)
@@ -92,7 +96,7 @@ def main():
         results["fedavg_K10"] = final_accs(job)
 
     if want("fedprox_K10"):
-        cfg = common(10, args, strategy="fedprox", admm_rho0=args.rho,
+        cfg = common(args.K, args, strategy="fedprox", admm_rho0=args.rho,
                      penalty_warmup_rounds=args.warmup)
         if args.jsonl_dir:
             cfg.jsonl_path = os.path.join(args.jsonl_dir, "fedprox.jsonl")
@@ -102,7 +106,7 @@ def main():
 
     if want("admm_K10"):
         # consensus ADMM K=10 (rho0, optionally BB-adaptive)
-        cfg = common(10, args, strategy="admm", admm_rho0=args.rho,
+        cfg = common(args.K, args, strategy="admm", admm_rho0=args.rho,
                      bb_update=bool(args.bb),
                      penalty_warmup_rounds=args.warmup)
         if args.jsonl_dir:
@@ -115,8 +119,9 @@ def main():
                   "per_client": [round(v, 1) for v in a]}
            for name, a in results.items()}
     out["wall_s"] = round(time.time() - t0, 1)
-    out["protocol"] = (f"Net CNN, synthetic class-structured CIFAR-shaped "
-                       f"data, Nloop={args.nloop} Nadmm={args.nadmm} "
+    out["protocol"] = (f"{args.model}, K={args.K}, dtype={args.dtype}, "
+                       f"synthetic class-structured CIFAR-shaped data, "
+                       f"Nloop={args.nloop} Nadmm={args.nadmm} "
                        f"Nepoch=1, Adam lr=1e-3, batch 128")
     print(json.dumps(out))
     keys = ["standalone_K1", "fedavg_K10", "fedprox_K10", "admm_K10",
