@@ -199,3 +199,33 @@ def test_round_checkpoint_kill_and_resume(tmp_path):
     assert done, "resumed run never reached the final block"
     assert abs(done[-1]["dual"] - tail_a[-1]["dual"]) < 1e-9
     assert abs(done[-1]["primal"] - tail_a[-1]["primal"]) < 1e-9
+
+
+def test_acc_experiment_tool_quick(tmp_path, monkeypatch):
+    """Guard the accuracy-protocol tool against bitrot (CPU quick mode)."""
+    import subprocess
+    import sys
+    r = subprocess.run(
+        [sys.executable, "tools/acc_experiment.py", "--quick",
+         "--only", "fedavg_K10", "--model", "Net", "--K", "2"],
+        capture_output=True, text=True, timeout=300,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 0, r.stderr[-500:]
+    rec = json.loads([l for l in r.stdout.splitlines()
+                      if l.startswith("{")][-1])
+    assert "fedavg_K10" in rec and rec["fedavg_K10"]["mean_acc"] > 0
+
+
+def test_sync_debug_wrapper():
+    """FEDKIT_SYNC_DEBUG wraps native calls with a post-call check."""
+    from fedkit.ops import _SyncDebugExt
+
+    class FakeMod:
+        const = 7
+
+        def op(self, a):
+            return a + 1
+
+    w = _SyncDebugExt(FakeMod())
+    assert w.const == 7
+    assert w.op(1) == 2   # torch.cuda.synchronize is a no-op without GPU
