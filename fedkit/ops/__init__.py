@@ -47,12 +47,13 @@ class _SyncDebugExt:
 
         def wrapped(*a, **kw):
             out = fn(*a, **kw)
-            try:
-                torch.cuda.synchronize()
-            except RuntimeError as e:
-                raise RuntimeError(
-                    f"fedkit._C.{name} faulted (FEDKIT_SYNC_DEBUG): {e}"
-                ) from e
+            if torch.cuda.is_available():
+                try:
+                    torch.cuda.synchronize()
+                except RuntimeError as e:
+                    raise RuntimeError(
+                        f"fedkit._C.{name} faulted (FEDKIT_SYNC_DEBUG): {e}"
+                    ) from e
             return out
         return wrapped
 
