@@ -229,3 +229,49 @@ def test_sync_debug_wrapper():
     w = _SyncDebugExt(FakeMod())
     assert w.const == 7
     assert w.op(1) == 2   # torch.cuda.synchronize is a no-op without GPU
+
+
+def test_round_checkpoint_resume_vae_rng(tmp_path):
+    """Resume reproducibility for an RNG-consuming model (VAE
+    reparametrization draws from the global torch RNG, which the sidecar
+    saves/restores)."""
+    from fedkit.ops.losses import vae_loss
+
+    def loss_fn(net, images, _labels):
+        out, mu, logvar = net(images)
+        return vae_loss(out, images, mu, logvar)
+
+    def cfg_for(sub, **kw):
+        d = tmp_path / sub
+        d.mkdir(exist_ok=True)
+        return FedConfig(K=1, default_batch=16, Nloop=2, Nepoch=1, Nadmm=1,
+                         use_cuda=False, check_results=False,
+                         max_steps_per_epoch=1, save_model=False,
+                         strategy="fedavg", model="AutoEncoderCNN",
+                         per_layer=True, round_checkpoint=True,
+                         ckpt_prefix=str(d / "s"), **kw)
+
+    job_a = FederatedJob(cfg_for("a"), loss_fn=loss_fn)
+    job_a.run()
+    sd_a = {k: v.clone() for k, v in job_a.nets[0].state_dict().items()}
+
+    class Killed(Exception):
+        pass
+
+    nl = {"v": 0, "seen": -1}
+
+    def hook(job, ci):
+        if ci <= nl["seen"]:
+            nl["v"] += 1
+        nl["seen"] = ci
+        if (nl["v"], ci) == (1, 3):
+            raise Killed
+
+    job_b = FederatedJob(cfg_for("b"), loss_fn=loss_fn, block_hook=hook)
+    with pytest.raises(Killed):
+        job_b.run()
+    job_c = FederatedJob(cfg_for("b", load_model=True), loss_fn=loss_fn)
+    job_c.run()
+    sd_c = job_c.nets[0].state_dict()
+    for k in sd_a:
+        assert torch.allclose(sd_a[k], sd_c[k], atol=1e-7), k
