@@ -81,12 +81,17 @@ def main():
     net = ResNet18().to(device)
     if use_cuda:
         net = net.to(memory_format=torch.channels_last)
-    # foreach (default) Adam: rocprof showed fused=True trading the 5
-    # foreach sweeps for 1 fused kernel PLUS one scalar step-counter add
-    # per parameter tensor (62 launches, ~185 us/step) — net loss.
-    # capturable only when hipGraph capture is requested.
-    opt = torch.optim.Adam(net.parameters(), lr=1e-3,
-                           capturable=bool(args.graph and use_cuda))
+    # fedkit FusedAdam: the whole update in ONE HIP kernel (torch's
+    # foreach Adam spends ~110 us/step over ~10 multi-tensor sweeps;
+    # torch's fused=True traded them for 1 kernel PLUS one step-counter
+    # add per tensor — 62 launches, net loss).  Falls back to stock under
+    # graph capture (the host-side step counter is not capture-safe).
+    if use_cuda and args.native and not args.graph:
+        from fedkit.optim import FusedAdam
+        opt = FusedAdam(net.parameters(), lr=1e-3)
+    else:
+        opt = torch.optim.Adam(net.parameters(), lr=1e-3,
+                               capturable=bool(args.graph and use_cuda))
     blocks = net.train_order_block_ids()
     params = list(net.parameters())
 

@@ -72,6 +72,12 @@ at::Tensor fedkit_conv2d_bwd_weight(const at::Tensor& gy, const at::Tensor& x,
                                     long stride, long padding, long R, long S,
                                     long dil);
 
+void fedkit_adam_step(std::vector<at::Tensor> params,
+                      std::vector<at::Tensor> grads,
+                      std::vector<at::Tensor> exp_avg,
+                      std::vector<at::Tensor> exp_avg_sq, double lr,
+                      double beta1, double beta2, double eps, long step,
+                      double weight_decay);
 at::Tensor fedkit_welford_update(const at::Tensor& g, at::Tensor avg,
                                  at::Tensor avg_sq, double inv_n);
 at::Tensor fedkit_linear_fwd(const at::Tensor& x, const at::Tensor& w,
@@ -177,6 +183,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_weight", &fedkit_conv2d_bwd_weight, "conv bwd-weight",
         py::arg("gy"), py::arg("x"), py::arg("stride"), py::arg("padding"),
         py::arg("R"), py::arg("S"), py::arg("dil") = 1);
+  m.def("adam_step", &fedkit_adam_step,
+        "fused Adam: one kernel updates every parameter (fp32 master)");
   m.def("welford_update", &fedkit_welford_update,
         "fused LBFGS Welford grad-stats update; returns sum(avg_sq)");
   m.def("linear_fwd", &fedkit_linear_fwd,

@@ -1,3 +1,4 @@
 from .lbfgsnew import LBFGSNew
+from .fusedadam import FusedAdam
 
-__all__ = ["LBFGSNew"]
+__all__ = ["LBFGSNew", "FusedAdam"]

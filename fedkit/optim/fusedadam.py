@@ -1,0 +1,62 @@
+"""FusedAdam — torch.optim.Adam semantics, ONE HIP kernel per step.
+
+The reference trains with stock Adam (federated_multi.py:156-159); the
+foreach implementation spends ~110 us/step on ResNet18 across ~10
+multi-tensor sweeps re-reading the moment buffers.  csrc/adam.hip does
+the whole update (m/v moments, bias correction, optional L2) in a single
+pass at the traffic floor.  State dict layout matches torch.optim.Adam
+(`step`/`exp_avg`/`exp_avg_sq` per param), so the reference's checkpoint
+contract (optimizer_state_dict) is unchanged.  Falls back to the stock
+update on CPU / non-fp32 / sparse.
+"""
+
+import torch
+
+
+class FusedAdam(torch.optim.Adam):
+    def _fused_applicable(self, group, params):
+        from ..ops import native_enabled, has_ext
+        if group["amsgrad"] or group.get("maximize"):
+            return False
+        return (has_ext() and params
+                and all(p.is_cuda and p.dtype == torch.float32
+                        and p.grad is not None and not p.grad.is_sparse
+                        for p in params))
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        from ..ops import require_ext
+        # all-or-nothing: any inapplicable group -> entirely stock path
+        # (CPU tests, exotic options); applicability is uniform in practice
+        if not all(self._fused_applicable(
+                g, [p for p in g["params"] if p.grad is not None])
+                for g in self.param_groups):
+            super().step()
+            return loss
+        for group in self.param_groups:
+            params = [p for p in group["params"] if p.grad is not None]
+            ps, gs, ms, vs = [], [], [], []
+            step_t = None
+            for p in params:
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = torch.tensor(0.0)
+                    state["exp_avg"] = torch.zeros_like(p)
+                    state["exp_avg_sq"] = torch.zeros_like(p)
+                state["step"] += 1
+                step_t = state["step"]
+                ps.append(p)
+                gs.append(p.grad)
+                ms.append(state["exp_avg"])
+                vs.append(state["exp_avg_sq"])
+            b1, b2 = group["betas"]
+            require_ext().adam_step(
+                ps, gs, ms, vs, group["lr"], b1, b2, group["eps"],
+                int(step_t.item()), group["weight_decay"])
+            for p in params:
+                torch.autograd.graph.increment_version(p)
+        return loss

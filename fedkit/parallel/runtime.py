@@ -40,7 +40,7 @@ import torch.optim as optim
 from ..algos import BBConfig, ConsensusADMM, FedAvg, FedProx, NoConsensus, Strategy
 from ..data import make_client_datasets
 from ..models import MODEL_FACTORIES
-from ..optim import LBFGSNew
+from ..optim import FusedAdam, LBFGSNew
 from ..ops import losses as loss_ops
 from ..utils import (flat_trainable, freeze_all_layers, get_trainable_values,
                      init_weights, number_of_blocks, put_trainable_values,
@@ -194,7 +194,9 @@ class FederatedJob:
         if self.cfg.optimizer == "lbfgs":
             return LBFGSNew(params, history_size=10, max_iter=4,
                             line_search_fn=True, batch_mode=True)
-        return optim.Adam(params, lr=self.cfg.lr)
+        # FusedAdam == torch Adam semantics/state-dict; one HIP kernel per
+        # step on GPU, stock update elsewhere
+        return FusedAdam(params, lr=self.cfg.lr)
 
     def _autocast(self):
         if self.bf16:

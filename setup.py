@@ -21,6 +21,7 @@ SOURCES = [
     "csrc/loss.hip",
     "csrc/losses_extra.hip",
     "csrc/linear.hip",
+    "csrc/adam.hip",
     "csrc/batchnorm.hip",
     "csrc/conv2d_mfma.hip",
     "csrc/conv_small.hip",
