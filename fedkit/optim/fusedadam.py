@@ -18,8 +18,9 @@ class FusedAdam(torch.optim.Adam):
         from ..ops import native_enabled, has_ext
         if group["amsgrad"] or group.get("maximize"):
             return False
+        # native_enabled also honors the FEDKIT_NATIVE=0 eager kill switch
         return (has_ext() and params
-                and all(p.is_cuda and p.dtype == torch.float32
+                and all(native_enabled(p) and p.dtype == torch.float32
                         and p.grad is not None and not p.grad.is_sparse
                         for p in params))
 
