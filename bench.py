@@ -86,7 +86,8 @@ def main():
     # torch's fused=True traded them for 1 kernel PLUS one step-counter
     # add per tensor — 62 launches, net loss).  Falls back to stock under
     # graph capture (the host-side step counter is not capture-safe).
-    if use_cuda and args.native and not args.graph:
+    if use_cuda and args.native and not args.graph \
+            and os.environ.get("FEDKIT_FUSED_ADAM", "1") != "0":
         from fedkit.optim import FusedAdam
         opt = FusedAdam(net.parameters(), lr=1e-3)
     else:
