@@ -31,7 +31,7 @@ def summarize(path):
     con.close()
     agg = {}
     for name, cname, val in rows:
-        agg.setdefault(name.split("(")[0][:70], {})[cname] = val
+        agg.setdefault(name.replace("(anonymous namespace)::", "")[:80], {})[cname] = val
     print(f"# {path}")
     key = "SQ_WAVE_CYCLES"
     for name, cs in sorted(agg.items(), key=lambda kv: -kv[1].get(key, 0))[:16]:
