@@ -33,25 +33,29 @@ void linear_fwd_kernel(const T* __restrict__ x,   // [M][K]
                        const float* __restrict__ b,  // [N] or null
                        T* __restrict__ y,         // [M][N]
                        long long M, int N, int K) {
-  const long long wid = ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int lane = threadIdx.x & 63;
-  if (wid >= M * N) return;
-  const int n = (int)(wid % N);
-  const long long m = wid / N;
-  const T* xr = x + m * K;
-  const T* wr = w + (long long)n * K;
-  float acc = 0.f;
-  for (int k = lane * 8; k + 8 <= K; k += 64 * 8)
-    acc += dot8(xr + k, wr + k);
-  // K tail (K % 8 != 0): lane 0 picks it up serially
-  if (lane == 0)
-    for (int k = (K / 8) * 8; k < K; ++k)
-      acc += to_f32(xr[k]) * to_f32(wr[k]);
-  for (int off = 32; off > 0; off >>= 1)
-    acc += __shfl_xor(acc, off, 64);
-  if (lane == 0) {
-    if (b) acc += b[n];
-    from_f32(acc, y[m * N + n]);
+  const long long wstride =
+      ((long long)gridDim.x * blockDim.x) >> 6;   // waves in the grid
+  for (long long wid = ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+       wid < M * N; wid += wstride) {
+    const int n = (int)(wid % N);
+    const long long m = wid / N;
+    const T* xr = x + m * K;
+    const T* wr = w + (long long)n * K;
+    float acc = 0.f;
+    for (int k = lane * 8; k + 8 <= K; k += 64 * 8)
+      acc += dot8(xr + k, wr + k);
+    // K tail (K % 8 != 0): lane 0 picks it up serially
+    if (lane == 0)
+      for (int k = (K / 8) * 8; k < K; ++k)
+        acc += to_f32(xr[k]) * to_f32(wr[k]);
+    for (int off = 32; off > 0; off >>= 1)
+      acc += __shfl_xor(acc, off, 64);
+    if (lane == 0) {
+      float out = acc;
+      if (b) out += b[n];
+      from_f32(out, y[m * N + n]);
+    }
   }
 }
 
@@ -63,19 +67,20 @@ void linear_bwd_data_kernel(const T* __restrict__ gy,  // [M][N]
                             T* __restrict__ gx,        // [M][K]
                             long long M, int N, int K) {
   const int kc8 = (K + 7) / 8;
-  const long long e = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (e >= M * kc8) return;
-  const long long m = e / kc8;
-  const int k0 = (int)(e % kc8) * 8;
-  const int kw = min(8, K - k0);
-  float acc[8] = {};
-  const T* gr = gy + m * N;
-  for (int n = 0; n < N; ++n) {
-    float g = to_f32(gr[n]);
-    const T* wr = w + (long long)n * K + k0;
-    for (int j = 0; j < kw; ++j) acc[j] += g * to_f32(wr[j]);
+  for (long long e = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       e < M * kc8; e += (long long)gridDim.x * blockDim.x) {
+    const long long m = e / kc8;
+    const int k0 = (int)(e % kc8) * 8;
+    const int kw = min(8, K - k0);
+    float acc[8] = {};
+    const T* gr = gy + m * N;
+    for (int n = 0; n < N; ++n) {
+      float g = to_f32(gr[n]);
+      const T* wr = w + (long long)n * K + k0;
+      for (int j = 0; j < kw; ++j) acc[j] += g * to_f32(wr[j]);
+    }
+    for (int j = 0; j < kw; ++j) from_f32(acc[j], gx[m * K + k0 + j]);
   }
-  for (int j = 0; j < kw; ++j) from_f32(acc[j], gx[m * K + k0 + j]);
 }
 
 // gw[n][k8] = sum_m gy[m][n] * x[m][k8..]; k-chunk 0 waves also emit
@@ -91,30 +96,31 @@ void linear_bwd_weight_kernel(const T* __restrict__ gy,  // [M][N]
                               float* __restrict__ gb,    // [N] fp32 or null
                               long long M, int N, int K) {
   const int kc8 = (K + 7) / 8;
-  const long long wid =
-      ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int lane = threadIdx.x & 63;
-  if (wid >= (long long)N * kc8) return;
-  const int n = (int)(wid / kc8);
-  const int k0 = (int)(wid % kc8) * 8;
-  const int kw = min(8, K - k0);
-  float acc[8] = {};
-  float bacc = 0.f;
-  for (long long m = lane; m < M; m += 64) {
-    float g = to_f32(gy[m * N + n]);
-    const T* xr = x + m * K + k0;
-    for (int j = 0; j < kw; ++j) acc[j] += g * to_f32(xr[j]);
-    bacc += g;
-  }
+  const long long wstride = ((long long)gridDim.x * blockDim.x) >> 6;
+  for (long long wid = ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+       wid < (long long)N * kc8; wid += wstride) {
+    const int n = (int)(wid / kc8);
+    const int k0 = (int)(wid % kc8) * 8;
+    const int kw = min(8, K - k0);
+    float acc[8] = {};
+    float bacc = 0.f;
+    for (long long m = lane; m < M; m += 64) {
+      float g = to_f32(gy[m * N + n]);
+      const T* xr = x + m * K + k0;
+      for (int j = 0; j < kw; ++j) acc[j] += g * to_f32(xr[j]);
+      bacc += g;
+    }
 #pragma unroll
-  for (int j = 0; j < 8; ++j)
+    for (int j = 0; j < 8; ++j)
+      for (int off = 32; off > 0; off >>= 1)
+        acc[j] += __shfl_xor(acc[j], off, 64);
     for (int off = 32; off > 0; off >>= 1)
-      acc[j] += __shfl_xor(acc[j], off, 64);
-  for (int off = 32; off > 0; off >>= 1)
-    bacc += __shfl_xor(bacc, off, 64);
-  if (lane == 0) {
-    for (int j = 0; j < kw; ++j) gw[(long long)n * K + k0 + j] = acc[j];
-    if (gb && k0 == 0) gb[n] = bacc;
+      bacc += __shfl_xor(bacc, off, 64);
+    if (lane == 0) {
+      for (int j = 0; j < kw; ++j) gw[(long long)n * K + k0 + j] = acc[j];
+      if (gb && k0 == 0) gb[n] = bacc;
+    }
   }
 }
 
