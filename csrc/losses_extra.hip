@@ -330,6 +330,8 @@ std::vector<at::Tensor> fedkit_vaecl_terms_fwd(
     const at::Tensor& mu_q, const at::Tensor& s_q, const at::Tensor& mu_p,
     const at::Tensor& s_p, long B) {
   TORCH_CHECK(mu_th.is_contiguous() && s_th.is_contiguous() &&
+              mu_q.is_contiguous() && s_q.is_contiguous() &&
+              mu_p.is_contiguous() && s_p.is_contiguous() &&
               x.is_contiguous(), "vaecl_terms expects contiguous tensors");
   long long KcB = mu_th.size(0);
   long long D1 = mu_th.numel() / KcB;
