@@ -275,3 +275,21 @@ def test_round_checkpoint_resume_vae_rng(tmp_path):
     sd_c = job_c.nets[0].state_dict()
     for k in sd_a:
         assert torch.allclose(sd_a[k], sd_c[k], atol=1e-7), k
+
+
+@pytest.mark.slow
+def test_cpc_driver_cpu_end_to_end(tmp_path):
+    """federated_cpc.py end to end on CPU: synthetic LOFAR visibilities,
+    LBFGS closures, InfoNCE, per-sub-model FedAvg, reference print
+    format (the driver the reference cannot even start — its
+    unfreeze_one_block TypeError — runs here)."""
+    import subprocess
+    import sys
+    r = subprocess.run(
+        [sys.executable, "federated_cpc.py", "--K", "2", "--Niter", "1",
+         "--Nloop", "1", "--Nadmm", "1", "--save_model", "0",
+         "--load_model", "0", "--use_cuda", "0"],
+        capture_output=True, text=True, timeout=540,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 0, r.stderr[-800:]
+    assert "dual (N=" in r.stdout     # reference-format FedAvg residual
